@@ -1,0 +1,275 @@
+"""Per-epoch dataset construction — seeded, rank-aware.
+
+Replicates the *distribution* of the reference's per-epoch resampling
+(reference model/dataset_builder.py:112-210): every epoch, each method's
+path-context bag is re-shuffled and truncated to ``max_path_length``
+contexts (implicit data augmentation that is load-bearing for F1), the
+``@method_0`` terminal is replaced by ``@question``, and the three index
+sequences are zero-padded.
+
+Differences from the reference, by design:
+- the reference's ``random`` module is never seeded (its 80/20 split and
+  resampling differ run-to-run); here all randomness derives from
+  ``(seed, epoch, rank)`` so DP replicas agree and runs reproduce,
+- items are sharded across DP ranks per epoch,
+- outputs are int32 numpy arrays (half the H2D bytes of the reference's
+  int64 tensors); the model casts as needed.
+"""
+
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass
+from typing import List, Optional, Sequence
+
+import numpy as np
+
+from .reader import CodeItem, CorpusReader
+
+logger = logging.getLogger(__name__)
+
+
+@dataclass
+class EpochData:
+    """One epoch's tensors for one rank (cf. reference CodeDataset rows)."""
+
+    ids: List[int]
+    starts: np.ndarray  # [N, C] int32
+    paths: np.ndarray   # [N, C] int32
+    ends: np.ndarray    # [N, C] int32
+    labels: np.ndarray  # [N] int64
+
+    def __len__(self) -> int:
+        return self.starts.shape[0]
+
+
+def _filter_variable_aliases(aliases: dict) -> List[str]:
+    return [a for a in aliases if a.startswith("@var_")]
+
+
+class DatasetBuilder:
+    """Train/test split + per-epoch tensor construction.
+
+    Mirrors reference model/dataset_builder.py:19-62 with seeded RNG.
+    """
+
+    def __init__(
+        self,
+        reader: CorpusReader,
+        option,
+        split_ratio: float = 0.2,
+        seed: int = 123,
+        rank: int = 0,
+        world_size: int = 1,
+    ) -> None:
+        self.reader = reader
+        self.option = option
+        self.seed = seed
+        self.rank = rank
+        self.world_size = world_size
+
+        rng = np.random.default_rng([seed, 0xC0DE])
+        order = rng.permutation(len(reader.items))
+        test_count = int(len(reader.items) * split_ratio)
+        self.test_items = [reader.items[i] for i in order[:test_count]]
+        self.train_items = [reader.items[i] for i in order[test_count:]]
+        logger.info("train item size: %d", len(self.train_items))
+        logger.info("test item size: %d", len(self.test_items))
+
+        train_sz, test_sz = self._dataset_sizes()
+        logger.info("train dataset size: %d", train_sz)
+        logger.info("test dataset size: %d", test_sz)
+
+        self.train_dataset: Optional[EpochData] = None
+        self.test_dataset: Optional[EpochData] = None
+
+        logger.info("OOV rate: %s", self.out_of_vocabulary_rate())
+
+    # ------------------------------------------------------------------
+    def _dataset_sizes(self):
+        train_sz = test_sz = 0
+        if self.reader.infer_method:
+            train_sz += len(self.train_items)
+            test_sz += len(self.test_items)
+        if self.reader.infer_variable:
+            train_sz += sum(
+                len(_filter_variable_aliases(it.aliases)) for it in self.train_items
+            )
+            test_sz += sum(
+                len(_filter_variable_aliases(it.aliases)) for it in self.test_items
+            )
+        return train_sz, test_sz
+
+    def _get_labels(self, normalized_label: str) -> Sequence[str]:
+        """Label subtokens for OOV accounting (reference :64-71)."""
+        if self.option.eval_method == "exact":
+            return [normalized_label]
+        idx = self.reader.label_vocab.stoi[normalized_label]
+        return self.reader.label_vocab.itosubtokens[idx]
+
+    def out_of_vocabulary_rate(self) -> float:
+        """Fraction of test label subtokens unseen in train labels
+        (reference model/dataset_builder.py:72-110)."""
+        train_vocab = set()
+        match = 0
+        count = 0
+        if self.reader.infer_method:
+            for it in self.train_items:
+                train_vocab.update(self._get_labels(it.normalized_label))
+        if self.reader.infer_variable:
+            for it in self.train_items:
+                for alias in _filter_variable_aliases(it.aliases):
+                    train_vocab.update(self._get_labels(it.aliases[alias]))
+        if self.reader.infer_method:
+            for it in self.test_items:
+                tokens = self._get_labels(it.normalized_label)
+                match += sum(1 for t in tokens if t in train_vocab)
+                count += len(tokens)
+        if self.reader.infer_variable:
+            for it in self.test_items:
+                for alias in _filter_variable_aliases(it.aliases):
+                    tokens = self._get_labels(it.aliases[alias])
+                    match += sum(1 for t in tokens if t in train_vocab)
+                    count += len(tokens)
+        if count == 0:
+            return 0.0
+        return 1.0 - match / count
+
+    # ------------------------------------------------------------------
+    def refresh_train_dataset(self, epoch: int = 0) -> EpochData:
+        """Rebuild the train tensors for ``epoch`` on this rank's shard."""
+        items = self._shard(self.train_items, epoch)
+        self.train_dataset = self.build_data(
+            items, self.option.max_path_length, epoch=epoch, stream=1
+        )
+        return self.train_dataset
+
+    def refresh_test_dataset(self, epoch: int = 0) -> EpochData:
+        items = self._shard(self.test_items, epoch)
+        self.test_dataset = self.build_data(
+            items, self.option.max_path_length, epoch=epoch, stream=2
+        )
+        return self.test_dataset
+
+    def _shard(self, items: List[CodeItem], epoch: int) -> List[CodeItem]:
+        if self.world_size <= 1:
+            return items
+        rng = np.random.default_rng([self.seed, epoch, 0x5A5A])
+        order = rng.permutation(len(items))
+        return [items[i] for i in order[self.rank :: self.world_size]]
+
+    # ------------------------------------------------------------------
+    def build_data(
+        self,
+        items: List[CodeItem],
+        max_path_length: int,
+        epoch: int = 0,
+        stream: int = 1,
+    ) -> EpochData:
+        """Resample + pad one epoch's tensors.
+
+        Method-name task (reference :122-150): shuffle the item's bag, keep
+        the first ``max_path_length`` contexts, replace @method_0 with
+        @question in start/end, pad with 0.
+        Variable-name task (reference :152-204): one sample per @var alias,
+        contexts filtered to those touching the variable.
+        """
+        C = max_path_length
+        reader = self.reader
+        rng = np.random.default_rng([self.seed, epoch, stream, self.rank])
+        q = reader.QUESTION_TOKEN_INDEX
+
+        ids: List[int] = []
+        labels: List[int] = []
+        starts_rows: List[np.ndarray] = []
+        paths_rows: List[np.ndarray] = []
+        ends_rows: List[np.ndarray] = []
+
+        if reader.infer_method:
+            method_token_index = reader.terminal_vocab.stoi["@method_0"]
+            label_stoi = reader.label_vocab.stoi
+            for item in items:
+                pcs = item.path_contexts
+                n = pcs.shape[0]
+                if n > C:
+                    sel = rng.permutation(n)[:C]
+                    chosen = pcs[sel]
+                else:
+                    chosen = pcs[rng.permutation(n)] if n > 1 else pcs
+                s = chosen[:, 0].copy()
+                p = chosen[:, 1]
+                e = chosen[:, 2].copy()
+                s[s == method_token_index] = q
+                e[e == method_token_index] = q
+                ids.append(item.id)
+                labels.append(label_stoi[item.normalized_label])
+                starts_rows.append(s)
+                paths_rows.append(np.asarray(p))
+                ends_rows.append(e)
+
+        if reader.infer_variable:
+            self._build_variable_rows(
+                items, C, rng, ids, labels, starts_rows, paths_rows, ends_rows
+            )
+
+        N = len(starts_rows)
+        starts = np.zeros((N, C), dtype=np.int32)
+        paths = np.zeros((N, C), dtype=np.int32)
+        ends = np.zeros((N, C), dtype=np.int32)
+        for i, (s, p, e) in enumerate(zip(starts_rows, paths_rows, ends_rows)):
+            k = min(len(s), C)
+            starts[i, :k] = s[:k]
+            paths[i, :k] = p[:k]
+            ends[i, :k] = e[:k]
+        return EpochData(
+            ids=ids,
+            starts=starts,
+            paths=paths,
+            ends=ends,
+            labels=np.asarray(labels, dtype=np.int64),
+        )
+
+    def _build_variable_rows(
+        self, items, C, rng, ids, labels, starts_rows, paths_rows, ends_rows
+    ) -> None:
+        reader = self.reader
+        q = reader.QUESTION_TOKEN_INDEX
+        term_stoi = reader.terminal_vocab.stoi
+        label_stoi = reader.label_vocab.stoi
+        variable_indexes = np.asarray(reader.variable_indexes, dtype=np.int32)
+        # identity remap unless shuffling is on (reference :157-164)
+        remap = {int(v): int(v) for v in variable_indexes}
+        for item in items:
+            alias_names = _filter_variable_aliases(item.aliases)
+            if not alias_names:
+                continue
+            alias_indexes = [term_stoi[a] for a in alias_names]
+            if reader.shuffle_variable_indexes:
+                perm = rng.permutation(len(variable_indexes))
+                remap = {
+                    int(k): int(variable_indexes[perm[i]])
+                    for i, k in enumerate(variable_indexes)
+                }
+            pcs = item.path_contexts
+            alias_set = np.isin(pcs[:, 0], alias_indexes) | np.isin(
+                pcs[:, 2], alias_indexes
+            )
+            var_pcs = pcs[alias_set]
+            var_pcs = var_pcs[rng.permutation(var_pcs.shape[0])]
+            for alias_name, var_idx in zip(alias_names, alias_indexes):
+                mask = (var_pcs[:, 0] == var_idx) | (var_pcs[:, 2] == var_idx)
+                sel = var_pcs[mask]
+                s = sel[:, 0].copy()
+                p = sel[:, 1]
+                e = sel[:, 2].copy()
+                for arr in (s, e):
+                    other = arr != var_idx
+                    arr[~other] = q
+                    if reader.shuffle_variable_indexes:
+                        for j in np.nonzero(other)[0]:
+                            arr[j] = remap.get(int(arr[j]), int(arr[j]))
+                ids.append(item.id)
+                labels.append(label_stoi[item.aliases[alias_name]])
+                starts_rows.append(s[:C])
+                paths_rows.append(np.asarray(p[:C]))
+                ends_rows.append(e[:C])

@@ -1,0 +1,199 @@
+"""Autograd wrappers around the HIP kernels.
+
+Each ``torch.autograd.Function`` here pairs a hand-written CDNA4 forward
+kernel with its backward kernels; plain dgrad/wgrad GEMMs go through
+rocBLAS (``torch.matmul``) by design — hand-written MFMA kernels cover the
+*fused* ops that no library provides.
+
+Numerical contracts are defined by ops/reference.py; tests compare against
+it in fp32.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import ext
+
+_rng_state = {"seed": None, "offset": 0}
+
+
+def _next_philox(n: int):
+    """Per-call (seed, offset) for the counter-based dropout RNG."""
+    if _rng_state["seed"] is None:
+        _rng_state["seed"] = torch.initial_seed() & 0x7FFFFFFFFFFFFFFF
+    seed = _rng_state["seed"]
+    offset = _rng_state["offset"]
+    _rng_state["offset"] += n
+    return seed, offset
+
+
+def reseed_dropout_rng(seed: int) -> None:
+    _rng_state["seed"] = seed & 0x7FFFFFFFFFFFFFFF
+    _rng_state["offset"] = 0
+
+
+class GatherConcat(torch.autograd.Function):
+    """K1/K2 + K13: fused triple embedding gather+concat; scatter-add bwd.
+
+    out[b*C+c] = [term[starts] (TS cols) | path[paths] (PS) | term[ends] (TS)]
+    """
+
+    @staticmethod
+    def forward(ctx, starts, paths, ends, term_w, path_w):
+        B, C = starts.shape
+        TS = term_w.shape[1]
+        PS = path_w.shape[1]
+        out = torch.empty(
+            B * C, 2 * TS + PS, dtype=torch.bfloat16, device=starts.device
+        )
+        ext().gather_concat_fwd(starts, paths, ends, term_w, path_w, out)
+        ctx.save_for_backward(starts, paths, ends)
+        ctx.shapes = (term_w.shape, path_w.shape)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        starts, paths, ends = ctx.saved_tensors
+        term_shape, path_shape = ctx.shapes
+        dterm32 = torch.zeros(term_shape, dtype=torch.float32, device=grad_out.device)
+        dpath32 = torch.zeros(path_shape, dtype=torch.float32, device=grad_out.device)
+        ext().gather_concat_bwd(
+            starts, paths, ends, grad_out.contiguous(), dterm32, dpath32
+        )
+        return None, None, None, dterm32.to(torch.bfloat16), dpath32.to(torch.bfloat16)
+
+
+class CombinerLNTanh(torch.autograd.Function):
+    """K3-K6: MFMA GEMM (x @ w) -> LayerNorm(E) -> tanh -> dropout, fused.
+
+    x: bf16 [M, KP]; w: bf16 [KP, EP]; gamma/beta: f32 [EP]; E = valid cols.
+    Saves z (pre-LN GEMM output, bf16) + per-row mean/rstd for backward;
+    the dropout mask is recomputed from the counter RNG, never stored.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, gamma, beta, E: int, p: float, training: bool):
+        M = x.shape[0]
+        EP = w.shape[1]
+        out = torch.empty(M, EP, dtype=torch.bfloat16, device=x.device)
+        z = torch.empty(M, EP, dtype=torch.bfloat16, device=x.device)
+        mean = torch.empty(M, dtype=torch.float32, device=x.device)
+        rstd = torch.empty(M, dtype=torch.float32, device=x.device)
+        p_eff = float(p) if training else 0.0
+        seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
+        ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff, seed, offset)
+        ctx.save_for_backward(x, w, gamma, beta, z, mean, rstd)
+        ctx.meta = (E, p_eff, seed, offset)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, w, gamma, beta, z, mean, rstd = ctx.saved_tensors
+        E, p, seed, offset = ctx.meta
+        M, EP = z.shape
+        dz = torch.empty(M, EP, dtype=torch.bfloat16, device=z.device)
+        dgamma = torch.zeros(EP, dtype=torch.float32, device=z.device)
+        dbeta = torch.zeros(EP, dtype=torch.float32, device=z.device)
+        ext().combiner_bwd(
+            dout.contiguous(), z, mean, rstd, gamma, beta, dz, dgamma, dbeta,
+            E, p, seed, offset,
+        )
+        # plain GEMMs -> rocBLAS
+        dx = dz @ w.t()
+        dw = x.t() @ dz
+        return dx, dw, dgamma, dbeta, None, None, None
+
+
+class AttentionPool(torch.autograd.Function):
+    """K7-K9 fused: masked single-query attention + weighted pool.
+
+    ccv: bf16 [B, C, EP]; a: f32 [EP]; starts: i32 [B, C] (mask = starts>0).
+    Returns cv f32 [B, EP] and attn f32 [B, C].
+    """
+
+    @staticmethod
+    def forward(ctx, ccv, a, starts, E: int):
+        B, C, EP = ccv.shape
+        cv = torch.empty(B, EP, dtype=torch.float32, device=ccv.device)
+        attn = torch.empty(B, C, dtype=torch.float32, device=ccv.device)
+        ext().attention_fwd(ccv, a, starts, cv, attn, E)
+        ctx.save_for_backward(ccv, a, starts, attn)
+        ctx.E = E
+        ctx.set_materialize_grads(False)
+        return cv, attn
+
+    @staticmethod
+    def backward(ctx, dcv, dattn):
+        ccv, a, starts, attn = ctx.saved_tensors
+        B, C, EP = ccv.shape
+        if dcv is None:
+            dcv = torch.zeros(B, EP, dtype=torch.float32, device=ccv.device)
+        dccv = torch.empty_like(ccv)
+        da = torch.zeros(EP, dtype=torch.float32, device=ccv.device)
+        has_dattn = dattn is not None
+        if not has_dattn:
+            dattn = torch.empty(0, dtype=torch.float32, device=ccv.device)
+        ext().attention_bwd(
+            dcv.contiguous(), dattn.contiguous() if has_dattn else dattn,
+            ccv, a, starts, attn, dccv, da, ctx.E, has_dattn,
+        )
+        return dccv, da, None, None
+
+
+class FusedLogSoftmaxNLL(torch.autograd.Function):
+    """K12: full-vocab log-softmax + weighted NLL, fused fwd and bwd.
+
+    logits: bf16 [B, L]; label: i64 [B]; weight: f32 [L] (1/freq weights,
+    reference main.py:129-130).  loss = sum(w_y*(lse - logit_y)) / sum(w_y).
+    """
+
+    @staticmethod
+    def forward(ctx, logits, label, weight):
+        B, L = logits.shape
+        lse = torch.empty(B, dtype=torch.float32, device=logits.device)
+        # acc[0] = sum(w_y * nll), acc[1] = sum(w_y)
+        acc = torch.zeros(2, dtype=torch.float32, device=logits.device)
+        ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc)
+        loss = acc[0] / acc[1]
+        ctx.save_for_backward(logits, label, weight, lse, acc)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, label, weight, lse, acc = ctx.saved_tensors
+        dlogits = torch.empty_like(logits)
+        ext().logsoftmax_nll_bwd(
+            logits, label, weight, lse, acc, dloss.contiguous().float(), dlogits
+        )
+        return dlogits, None, None
+
+
+def adam_step(
+    param: torch.Tensor,
+    grad: torch.Tensor,
+    master: Optional[torch.Tensor],
+    m: torch.Tensor,
+    v: torch.Tensor,
+    step: int,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+) -> None:
+    """K16: fused Adam.  bf16 params carry an f32 master (updated in f32,
+    rounded to bf16 param); f32 params update in place (master is None)."""
+    if param.dtype == torch.bfloat16:
+        assert master is not None
+        ext().adam_step_bf16(
+            param.view(-1), grad.view(-1), master, m, v,
+            step, lr, beta1, beta2, eps, weight_decay,
+        )
+    else:
+        ext().adam_step_f32(
+            param.view(-1), grad.view(-1).float(), m, v,
+            step, lr, beta1, beta2, eps, weight_decay,
+        )

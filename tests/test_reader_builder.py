@@ -1,0 +1,131 @@
+"""Corpus parse + per-epoch builder semantics (reference
+model/dataset_reader.py:72-128, model/dataset_builder.py:112-210)."""
+
+import numpy as np
+import pytest
+
+from code2vec_amd.data.builder import DatasetBuilder
+from code2vec_amd.data.reader import CorpusReader
+from code2vec_amd.utils.options import Option
+
+
+def make_reader(tiny_corpus, **kw):
+    return CorpusReader(
+        tiny_corpus["corpus_path"],
+        tiny_corpus["path_idx_path"],
+        tiny_corpus["terminal_idx_path"],
+        **kw,
+    )
+
+
+def make_option(reader, **kw):
+    defaults = dict(max_path_length=16, terminal_embed_size=8,
+                    path_embed_size=8, encode_size=12)
+    defaults.update(kw)
+    return Option(
+        terminal_count=len(reader.terminal_vocab),
+        path_count=len(reader.path_vocab),
+        label_count=len(reader.label_vocab),
+        **defaults,
+    )
+
+
+def test_reader_question_offset(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    # terminal vocab has @question at 1; start/end corpus indexes shifted +1
+    assert r.terminal_vocab.stoi["@question"] == 1
+    assert r.terminal_vocab.stoi["<PAD/>"] == 0
+    # raw file terminal index i appears as i+1 in path_contexts start/end
+    for item in r.items[:5]:
+        assert item.path_contexts[:, 0].min() >= 2  # raw >= 1, +1 shift
+        assert item.path_contexts[:, 1].min() >= 1  # paths unshifted
+
+
+def test_reader_items_and_labels(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    assert len(r.items) == 48
+    for item in r.items:
+        assert item.normalized_label in r.label_vocab.stoi
+        idx = r.label_vocab.stoi[item.normalized_label]
+        assert idx in r.label_vocab.itosubtokens
+        assert item.id is not None
+        assert item.aliases  # vars: section parsed
+
+
+def test_reader_variable_indexes(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    for idx in r.variable_indexes:
+        assert r.terminal_vocab.itos[idx].startswith("@var_")
+
+
+def test_builder_split_and_shapes(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    opt = make_option(r)
+    b = DatasetBuilder(r, opt, seed=11)
+    assert len(b.train_items) + len(b.test_items) == len(r.items)
+    assert len(b.test_items) == int(len(r.items) * 0.2)
+
+    data = b.refresh_train_dataset(epoch=0)
+    C = opt.max_path_length
+    assert data.starts.shape == (len(b.train_items), C)
+    assert data.starts.dtype == np.int32
+    assert data.labels.dtype == np.int64
+    # padding is zero; non-pad values positive
+    assert (data.starts >= 0).all()
+
+
+def test_builder_resamples_per_epoch(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    opt = make_option(r, max_path_length=4)  # force truncation
+    b = DatasetBuilder(r, opt, seed=11)
+    d0 = b.refresh_train_dataset(epoch=0)
+    d1 = b.refresh_train_dataset(epoch=1)
+    # same items, different resampled contexts
+    assert d0.labels.tolist() == d1.labels.tolist()
+    assert not np.array_equal(d0.paths, d1.paths)
+    # deterministic per (seed, epoch)
+    d0b = b.refresh_train_dataset(epoch=0)
+    assert np.array_equal(d0.paths, d0b.paths)
+
+
+def test_builder_question_replacement(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    opt = make_option(r)
+    b = DatasetBuilder(r, opt, seed=3)
+    method_idx = r.terminal_vocab.stoi["@method_0"]
+    data = b.refresh_train_dataset(epoch=0)
+    assert not np.any(data.starts == method_idx)
+    assert not np.any(data.ends == method_idx)
+
+
+def test_builder_rank_sharding(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    opt = make_option(r)
+    shards = []
+    for rank in range(2):
+        b = DatasetBuilder(r, opt, seed=5, rank=rank, world_size=2)
+        d = b.refresh_train_dataset(epoch=0)
+        shards.append(set(d.ids))
+    assert shards[0].isdisjoint(shards[1])
+    b_all = DatasetBuilder(r, opt, seed=5)
+    d_all = b_all.refresh_train_dataset(epoch=0)
+    assert shards[0] | shards[1] == set(d_all.ids)
+
+
+def test_builder_variable_task(tiny_corpus):
+    r = make_reader(tiny_corpus, infer_method=False, infer_variable=True)
+    opt = make_option(r)
+    b = DatasetBuilder(r, opt, seed=3)
+    data = b.refresh_train_dataset(epoch=0)
+    # one row per @var alias per item that has matching contexts
+    assert len(data) > 0
+    # all labels exist in the variable label vocab
+    assert all(0 <= l < len(r.label_vocab) for l in data.labels)
+
+
+def test_oov_rate_bounds(tiny_corpus):
+    r = make_reader(tiny_corpus)
+    opt = make_option(r)
+    b = DatasetBuilder(r, opt, seed=11)
+    rate = b.out_of_vocabulary_rate()
+    assert 0.0 <= rate <= 1.0

@@ -1,0 +1,128 @@
+"""End-to-end CPU training on a tiny synthetic corpus (BASELINE.json
+config 1 shape) + artifact format round-trips."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+import main as cli
+from code2vec_amd.data.builder import DatasetBuilder
+from code2vec_amd.data.reader import CorpusReader
+from code2vec_amd.engine.trainer import Trainer, TrainerConfig
+from code2vec_amd.models.code2vec import build_model, init_logical_params
+from code2vec_amd.parallel.dist import DistContext
+from code2vec_amd.utils.options import Option
+
+
+def run_cli(tmp_path, tiny_corpus, extra=None):
+    out_dir = tmp_path / "out"
+    argv = [
+        "--corpus_path", tiny_corpus["corpus_path"],
+        "--path_idx_path", tiny_corpus["path_idx_path"],
+        "--terminal_idx_path", tiny_corpus["terminal_idx_path"],
+        "--model_path", str(out_dir),
+        "--vectors_path", str(out_dir / "code.vec"),
+        "--test_result_path", str(out_dir / "test_results.tsv"),
+        "--max_epoch", "2",
+        "--batch_size", "16",
+        "--terminal_embed_size", "12",
+        "--path_embed_size", "12",
+        "--encode_size", "16",
+        "--max_path_length", "12",
+        "--no_cuda",
+        "--print_sample_cycle", "0",
+    ]
+    if extra:
+        argv.extend(extra)
+    cli.main(argv)
+    return out_dir
+
+
+def test_cli_end_to_end(tmp_path, tiny_corpus):
+    out_dir = run_cli(tmp_path, tiny_corpus)
+    vec_file = out_dir / "code.vec"
+    assert vec_file.exists()
+    lines = vec_file.read_text().splitlines()
+    # header: {len(reader.items)}\t{encode_size} (the reference's quirk:
+    # count is items, rows are train+test dataset sizes)
+    count, esz = lines[0].split("\t")
+    assert int(count) == 48
+    assert int(esz) == 16
+    assert len(lines) == 1 + 48  # method task: rows == items
+    for row in lines[1:]:
+        label, vec = row.split("\t")
+        values = vec.split(" ")
+        assert len(values) == 16
+        float(values[0])
+
+    # checkpoint with reference-format keys
+    sd = torch.load(out_dir / "code2vec.model", weights_only=True)
+    assert "terminal_embedding.weight" in sd
+    assert "input_linear.weight" in sd
+    assert "attention_parameter" in sd
+    assert sd["input_linear.weight"].shape == (16, 36)
+
+    # test-result TSV: id \t correct \t expected \t predicted \t prob
+    tsv = (out_dir / "test_results.tsv").read_text().splitlines()
+    assert len(tsv) == int(48 * 0.2)
+    parts = tsv[0].split("\t")
+    assert len(parts) == 5
+    int(parts[0]); float(parts[4])
+    assert parts[1] in ("True", "False")
+
+
+def test_cli_variable_task(tmp_path, tiny_corpus):
+    out_dir = run_cli(
+        tmp_path, tiny_corpus,
+        extra=["--infer_method_name", "false", "--infer_variable_name", "true"],
+    )
+    assert (out_dir / "code.vec").exists()
+
+
+def test_training_reduces_loss(tiny_corpus):
+    """Loss-curve sanity: a few epochs on the tiny corpus reduce train loss."""
+    reader = CorpusReader(
+        tiny_corpus["corpus_path"], tiny_corpus["path_idx_path"],
+        tiny_corpus["terminal_idx_path"],
+    )
+    opt = Option(
+        terminal_count=len(reader.terminal_vocab),
+        path_count=len(reader.path_vocab),
+        label_count=len(reader.label_vocab),
+        max_path_length=12, terminal_embed_size=12, path_embed_size=12,
+        encode_size=16, dropout_prob=0.0, batch_size=16,
+        device=torch.device("cpu"),
+    )
+    builder = DatasetBuilder(reader, opt, seed=1)
+    g = torch.Generator().manual_seed(0)
+    model = build_model(opt, backend="torch", logical=init_logical_params(opt, g))
+    ctx = DistContext(0, 1, 0, torch.device("cpu"))
+
+    class A:  # minimal args carrier
+        max_epoch = 1; lr = 0.01; beta_min = 0.9; beta_max = 0.999
+        weight_decay = 0.0; model_path = "/tmp/c2v_test_out"
+        vectors_path = "/tmp/c2v_test_out/code.vec"; test_result_path = None
+        env = None; print_sample_cycle = 0; eval_method = "subtoken"
+        random_seed = 1; batch_size = 16
+
+    trainer = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
+    losses = []
+    for epoch in range(4):
+        loss, _ = trainer._train_epoch(epoch)
+        losses.append(loss)
+    assert losses[-1] < losses[0]
+
+
+def test_code_vec_visualizer_parse_roundtrip(tmp_path, tiny_corpus):
+    """The exported code.vec parses with the downstream consumer's logic
+    (reference visualize_code_vec.py:8-23: skip header, tab-split label,
+    space-split floats)."""
+    out_dir = run_cli(tmp_path, tiny_corpus)
+    from tools.visualize_code_vec import parse_code_vec
+
+    labels, mat = parse_code_vec(str(out_dir / "code.vec"))
+    assert len(labels) == mat.shape[0] == 48
+    assert mat.shape[1] == 16
+    assert np.isfinite(mat).all()
