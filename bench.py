@@ -1,0 +1,169 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — the driver contract.
+
+Measures the headline metric of BASELINE.json: whole-node
+path-contexts/sec of code2vec training on the top11-shaped config
+(terminal vocab 360,631(+question), path vocab 342,845(+pad), batch 1024,
+200 contexts/method, embed/encode 100, bf16) with synthetic dense data and
+random-init weights.  One process per GPU (torchrun), RCCL data parallelism,
+weak scaling (per-GPU batch fixed at 1024).
+
+A timed step = forward + fused loss + backward + bucketed gradient
+all-reduce + fused Adam — the full training step.  label_count for top11 is
+not published by the reference; 30,000 is assumed and stated in config.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--config top11]
+For N>1 the driver launches via torch.distributed.run.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from code2vec_amd.data.synthetic import synthetic_batch
+from code2vec_amd.engine.optim import FusedAdam
+from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+from code2vec_amd.parallel.ddp import BucketedAllReduce
+from code2vec_amd.parallel.dist import init_distributed
+from code2vec_amd.utils.options import Option
+
+CONFIGS = {
+    # BASELINE.json config 2/3: top11-shaped (top11_dataset/params.txt:7-9)
+    "top11": dict(terminal_count=360632, path_count=342846, label_count=30000,
+                  embed=100, encode=100, batch=1024, contexts=200),
+    # BASELINE.json config 4: java-large-scale synthetic
+    "java-large": dict(terminal_count=1_100_000, path_count=1_300_000,
+                       label_count=261_000, embed=128, encode=128,
+                       batch=1024, contexts=200),
+    # tiny CPU-shaped plumbing config
+    "tiny": dict(terminal_count=12000, path_count=10000, label_count=5000,
+                 embed=100, encode=100, batch=32, contexts=200),
+}
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--config", type=str, default="top11", choices=CONFIGS)
+    ap.add_argument("--pool", type=int, default=4, help="synthetic batch pool size")
+    args = ap.parse_args()
+
+    cfg = CONFIGS[args.config]
+    ctx = init_distributed("cuda" if torch.cuda.is_available() else "cpu")
+    world = ctx.world_size
+    assert world == args.gpus or world == 1, (
+        f"WORLD_SIZE={world} but --gpus={args.gpus}"
+    )
+    device = ctx.device
+    on_gpu = device.type == "cuda"
+
+    opt = Option(
+        terminal_count=cfg["terminal_count"], path_count=cfg["path_count"],
+        label_count=cfg["label_count"], max_path_length=cfg["contexts"],
+        terminal_embed_size=cfg["embed"], path_embed_size=cfg["embed"],
+        encode_size=cfg["encode"], dropout_prob=0.25,
+        batch_size=cfg["batch"], device=device,
+    )
+    g = torch.Generator().manual_seed(123)
+    if on_gpu:
+        model = Code2VecHIP(opt, init_logical_params(opt, g), device=device)
+        optim = FusedAdam(model.parameters(), lr=0.01, betas=(0.9, 0.999))
+    else:
+        from code2vec_amd.models.code2vec import Code2VecTorch
+
+        model = Code2VecTorch(opt, init_logical_params(opt, g))
+        optim = torch.optim.Adam(model.parameters(), lr=0.01)
+
+    ddp = BucketedAllReduce(list(model.parameters()), world)
+    ddp.broadcast_parameters()
+
+    B, C = cfg["batch"], cfg["contexts"]
+    rng = np.random.default_rng(1000 + ctx.rank)
+    pool = []
+    for _ in range(args.pool):
+        s, p, e, y = synthetic_batch(rng, B, C, cfg["terminal_count"],
+                                     cfg["path_count"], cfg["label_count"])
+        pool.append(tuple(
+            torch.from_numpy(a).to(device) for a in (s, p, e, y)
+        ))
+    class_weight = torch.ones(cfg["label_count"], device=device)
+
+    model.train()
+
+    def step(i: int) -> None:
+        s, p, e, y = pool[i % len(pool)]
+        ddp.zero_grad()
+        outputs, _, _ = model(s, p, e, y)
+        loss = model.loss(outputs, y, class_weight)
+        loss.backward()
+        ddp.finish()
+        optim.step()
+
+    for i in range(args.warmup):
+        step(i)
+
+    if ctx.initialized:
+        torch.distributed.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if ctx.initialized:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if ctx.initialized:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    contexts_per_step = world * B * C
+    value = contexts_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if ctx.is_rank0:
+        print(json.dumps({
+            "metric": "path_contexts_per_sec",
+            "value": value,
+            "unit": "path-contexts/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic (dense 200 contexts/method, random indices, random-init weights)",
+            "config": {
+                "model": f"code2vec-{args.config}",
+                "global_batch": world * B,
+                "seq_len": C,
+                "parallelism": f"dp{world}",
+                "terminal_vocab": cfg["terminal_count"],
+                "path_vocab": cfg["path_count"],
+                "label_vocab": cfg["label_count"],
+                "embed": cfg["embed"],
+                "encode": cfg["encode"],
+                "note": "step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam; top11 label vocab size unpublished, 30k assumed",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

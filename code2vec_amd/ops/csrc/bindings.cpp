@@ -1,0 +1,207 @@
+// Python bindings for the code2vec_amd HIP kernel library (_c2v_hip).
+// Pure C++ TU: declares the extern "C" launchers implemented in the .hip
+// files and validates tensor dtype/contiguity/shape before launching on the
+// current torch HIP stream.
+
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#define CHK(x) TORCH_CHECK(x, #x)
+#define CHK_CONTIG(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+#define CHK_CUDA(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+#define CHK_DT(t, dt) TORCH_CHECK((t).scalar_type() == dt, #t " dtype mismatch")
+
+extern "C" {
+void launch_gather_concat_fwd(const int*, const int*, const int*, const void*,
+                              const void*, void*, long, int, int, hipStream_t);
+void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
+                              float*, float*, long, int, int, hipStream_t);
+void launch_combiner_fwd(const void*, const void*, const float*, const float*,
+                         void*, void*, float*, float*, long, int, int, int,
+                         float, unsigned long long, unsigned long long,
+                         hipStream_t);
+void launch_combiner_bwd(const void*, const void*, const float*, const float*,
+                         const float*, const float*, void*, float*, float*,
+                         long, int, int, float, unsigned long long,
+                         unsigned long long, hipStream_t);
+void launch_attention_fwd(const void*, const float*, const int*, float*,
+                          float*, int, int, int, int, hipStream_t);
+void launch_attention_bwd(const float*, const float*, const void*,
+                          const float*, const int*, const float*, void*,
+                          float*, int, int, int, int, int, hipStream_t);
+void launch_lsm_nll_fwd(const void*, const long*, const float*, float*,
+                        float*, int, long, hipStream_t);
+void launch_lsm_nll_bwd(const void*, const long*, const float*, const float*,
+                        const float*, const float*, void*, int, long,
+                        hipStream_t);
+void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
+                      float, float, float, float, float, hipStream_t);
+void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
+                     float, float, float, float, hipStream_t);
+}
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void gather_concat_fwd(torch::Tensor starts, torch::Tensor paths,
+                       torch::Tensor ends, torch::Tensor term,
+                       torch::Tensor path, torch::Tensor out) {
+  CHK_CUDA(starts); CHK_CONTIG(starts); CHK_DT(starts, torch::kInt32);
+  CHK_CONTIG(paths); CHK_CONTIG(ends);
+  CHK_DT(term, torch::kBFloat16); CHK_CONTIG(term);
+  CHK_DT(path, torch::kBFloat16); CHK_CONTIG(path);
+  CHK_DT(out, torch::kBFloat16); CHK_CONTIG(out);
+  const long M = starts.numel();
+  const int TS = term.size(1), PS = path.size(1);
+  TORCH_CHECK(out.size(0) == M && out.size(1) == 2 * TS + PS, "out shape");
+  TORCH_CHECK(TS % 32 == 0 && PS % 32 == 0, "padded strides must be 32-mult");
+  launch_gather_concat_fwd(starts.data_ptr<int>(), paths.data_ptr<int>(),
+                           ends.data_ptr<int>(), term.data_ptr(),
+                           path.data_ptr(), out.data_ptr(), M, TS, PS,
+                           cur_stream());
+}
+
+void gather_concat_bwd(torch::Tensor starts, torch::Tensor paths,
+                       torch::Tensor ends, torch::Tensor gout,
+                       torch::Tensor dterm, torch::Tensor dpath) {
+  CHK_CUDA(gout); CHK_CONTIG(gout); CHK_DT(gout, torch::kBFloat16);
+  CHK_DT(dterm, torch::kFloat32); CHK_CONTIG(dterm);
+  CHK_DT(dpath, torch::kFloat32); CHK_CONTIG(dpath);
+  const long M = starts.numel();
+  const int TS = dterm.size(1), PS = dpath.size(1);
+  launch_gather_concat_bwd(starts.data_ptr<int>(), paths.data_ptr<int>(),
+                           ends.data_ptr<int>(), gout.data_ptr(),
+                           dterm.data_ptr<float>(), dpath.data_ptr<float>(),
+                           M, TS, PS, cur_stream());
+}
+
+void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
+                  torch::Tensor beta, torch::Tensor out, torch::Tensor z,
+                  torch::Tensor mean, torch::Tensor rstd, int64_t E,
+                  double p, int64_t seed, int64_t offset) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
+  CHK_DT(gamma, torch::kFloat32); CHK_DT(beta, torch::kFloat32);
+  CHK_DT(out, torch::kBFloat16); CHK_DT(z, torch::kBFloat16);
+  CHK_DT(mean, torch::kFloat32); CHK_DT(rstd, torch::kFloat32);
+  const long M = x.size(0);
+  const int KP = x.size(1), EP = w.size(1);
+  TORCH_CHECK(w.size(0) == KP && KP % 32 == 0 && EP % 32 == 0, "w shape");
+  launch_combiner_fwd(x.data_ptr(), w.data_ptr(), gamma.data_ptr<float>(),
+                      beta.data_ptr<float>(), out.data_ptr(), z.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(), M, KP,
+                      EP, (int)E, (float)p, (unsigned long long)seed,
+                      (unsigned long long)offset, cur_stream());
+}
+
+void combiner_bwd(torch::Tensor dout, torch::Tensor z, torch::Tensor mean,
+                  torch::Tensor rstd, torch::Tensor gamma, torch::Tensor beta,
+                  torch::Tensor dz, torch::Tensor dgamma, torch::Tensor dbeta,
+                  int64_t E, double p, int64_t seed, int64_t offset) {
+  CHK_CUDA(dout); CHK_CONTIG(dout); CHK_DT(dout, torch::kBFloat16);
+  CHK_CONTIG(z); CHK_DT(dz, torch::kBFloat16);
+  const long M = z.size(0);
+  const int EP = z.size(1);
+  launch_combiner_bwd(dout.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
+                      rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                      beta.data_ptr<float>(), dz.data_ptr(),
+                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M,
+                      EP, (int)E, (float)p, (unsigned long long)seed,
+                      (unsigned long long)offset, cur_stream());
+}
+
+void attention_fwd(torch::Tensor ccv, torch::Tensor a, torch::Tensor starts,
+                   torch::Tensor cv, torch::Tensor attn, int64_t E) {
+  CHK_CUDA(ccv); CHK_CONTIG(ccv); CHK_DT(ccv, torch::kBFloat16);
+  CHK_DT(a, torch::kFloat32); CHK_DT(cv, torch::kFloat32);
+  CHK_DT(attn, torch::kFloat32); CHK_DT(starts, torch::kInt32);
+  const int B = ccv.size(0), C = ccv.size(1), EP = ccv.size(2);
+  launch_attention_fwd(ccv.data_ptr(), a.data_ptr<float>(),
+                       starts.data_ptr<int>(), cv.data_ptr<float>(),
+                       attn.data_ptr<float>(), B, C, EP, (int)E,
+                       cur_stream());
+}
+
+void attention_bwd(torch::Tensor dcv, torch::Tensor dattn, torch::Tensor ccv,
+                   torch::Tensor a, torch::Tensor starts, torch::Tensor attn,
+                   torch::Tensor dccv, torch::Tensor da, int64_t E,
+                   bool has_dattn) {
+  CHK_CUDA(dcv); CHK_CONTIG(dcv); CHK_DT(dcv, torch::kFloat32);
+  CHK_DT(dccv, torch::kBFloat16); CHK_DT(da, torch::kFloat32);
+  const int B = ccv.size(0), C = ccv.size(1), EP = ccv.size(2);
+  launch_attention_bwd(dcv.data_ptr<float>(),
+                       has_dattn ? dattn.data_ptr<float>() : nullptr,
+                       ccv.data_ptr(), a.data_ptr<float>(),
+                       starts.data_ptr<int>(), attn.data_ptr<float>(),
+                       dccv.data_ptr(), da.data_ptr<float>(), B, C, EP,
+                       (int)E, has_dattn ? 1 : 0, cur_stream());
+}
+
+void logsoftmax_nll_fwd(torch::Tensor logits, torch::Tensor label,
+                        torch::Tensor weight, torch::Tensor lse,
+                        torch::Tensor acc) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_DT(label, torch::kInt64); CHK_DT(lse, torch::kFloat32);
+  const int B = logits.size(0);
+  const long L = logits.size(1);
+  const float* wp = weight.defined() && weight.numel() > 0
+                        ? weight.data_ptr<float>()
+                        : nullptr;
+  launch_lsm_nll_fwd(logits.data_ptr(), label.data_ptr<long>(), wp,
+                     lse.data_ptr<float>(), acc.data_ptr<float>(), B, L,
+                     cur_stream());
+}
+
+void logsoftmax_nll_bwd(torch::Tensor logits, torch::Tensor label,
+                        torch::Tensor weight, torch::Tensor lse,
+                        torch::Tensor acc, torch::Tensor gscale,
+                        torch::Tensor dlogits) {
+  CHK_CUDA(logits); CHK_CONTIG(dlogits); CHK_DT(dlogits, torch::kBFloat16);
+  const int B = logits.size(0);
+  const long L = logits.size(1);
+  const float* wp = weight.defined() && weight.numel() > 0
+                        ? weight.data_ptr<float>()
+                        : nullptr;
+  launch_lsm_nll_bwd(logits.data_ptr(), label.data_ptr<long>(), wp,
+                     lse.data_ptr<float>(), acc.data_ptr<float>(),
+                     gscale.data_ptr<float>(), dlogits.data_ptr(), B, L,
+                     cur_stream());
+}
+
+void adam_step_bf16(torch::Tensor p, torch::Tensor g, torch::Tensor master,
+                    torch::Tensor m, torch::Tensor v, int64_t step, double lr,
+                    double b1, double b2, double eps, double wd) {
+  CHK_CUDA(p); CHK_DT(p, torch::kBFloat16); CHK_DT(g, torch::kBFloat16);
+  CHK_DT(master, torch::kFloat32);
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == master.numel(), "sizes");
+  launch_adam_bf16(p.data_ptr(), g.data_ptr(), master.data_ptr<float>(),
+                   m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+                   (int)step, (float)lr, (float)b1, (float)b2, (float)eps,
+                   (float)wd, cur_stream());
+}
+
+void adam_step_f32(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                   torch::Tensor v, int64_t step, double lr, double b1,
+                   double b2, double eps, double wd) {
+  CHK_CUDA(p); CHK_DT(p, torch::kFloat32); CHK_DT(g, torch::kFloat32);
+  launch_adam_f32(p.data_ptr<float>(), g.data_ptr<float>(),
+                  m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+                  (int)step, (float)lr, (float)b1, (float)b2, (float)eps,
+                  (float)wd, cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gather_concat_fwd", &gather_concat_fwd);
+  m.def("gather_concat_bwd", &gather_concat_bwd);
+  m.def("combiner_fwd", &combiner_fwd);
+  m.def("combiner_bwd", &combiner_bwd);
+  m.def("attention_fwd", &attention_fwd);
+  m.def("attention_bwd", &attention_bwd);
+  m.def("logsoftmax_nll_fwd", &logsoftmax_nll_fwd);
+  m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
+  m.def("adam_step_bf16", &adam_step_bf16);
+  m.def("adam_step_f32", &adam_step_f32);
+}

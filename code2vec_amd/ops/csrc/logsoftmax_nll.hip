@@ -1,0 +1,122 @@
+// K12: fused full-vocab log-softmax + weighted NLL, forward and backward.
+// Reference math: main.py:251-264 with criterion nn.NLLLoss(weight=1/freq)
+// (main.py:129-130): loss = sum_i w[y_i]*(lse_i - logit_i[y_i]) / sum_i w[y_i].
+//
+// Single pass over L per row (online max+sum), fp32 accumulation over bf16
+// logits; built for L up to ~261k columns (java-large config).
+
+#include "common.h"
+
+// one block per row; online (max, sum) over the row's L columns
+__global__ __launch_bounds__(256) void lsm_nll_fwd_kernel(
+    const bf16* __restrict__ logits, const long* __restrict__ label,
+    const float* __restrict__ weight, float* __restrict__ lse,
+    float* __restrict__ acc, int B, long L) {
+  const int b = blockIdx.x;
+  const bf16* row = logits + (long)b * L;
+  float m = -3.0e38f, s = 0.f;
+  const long L8 = L & ~7L;
+  for (long j = (long)threadIdx.x * 8; j < L8; j += (long)blockDim.x * 8) {
+    bf16 v[8];
+    *(uint4*)v = *(const uint4*)(row + j);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const float x = bf2f(v[k]);
+      if (x > m) {
+        s *= __expf(m - x);
+        m = x;
+      }
+      s += __expf(x - m);
+    }
+  }
+  for (long j = L8 + threadIdx.x; j < L; j += blockDim.x) {
+    const float x = bf2f(row[j]);
+    if (x > m) {
+      s *= __expf(m - x);
+      m = x;
+    }
+    s += __expf(x - m);
+  }
+  // block-combine (m, s) pairs
+  __shared__ float red_m[64], red_s[64];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float om = __shfl_xor(m, off);
+    const float os = __shfl_xor(s, off);
+    const float nm = fmaxf(m, om);
+    s = s * __expf(m - nm) + os * __expf(om - nm);
+    m = nm;
+  }
+  if (lane == 0) { red_m[wave] = m; red_s[wave] = s; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int nwaves = blockDim.x / WAVE;
+    float M0 = red_m[0], S0 = red_s[0];
+    for (int w = 1; w < nwaves; ++w) {
+      const float nm = fmaxf(M0, red_m[w]);
+      S0 = S0 * __expf(M0 - nm) + red_s[w] * __expf(red_m[w] - nm);
+      M0 = nm;
+    }
+    const float l = M0 + __logf(S0);
+    lse[b] = l;
+    const long y = label[b];
+    const float wy = weight ? weight[y] : 1.0f;
+    atomic_add_f32(&acc[0], wy * (l - bf2f(row[y])));
+    atomic_add_f32(&acc[1], wy);
+  }
+}
+
+// dlogits[b,j] = gscale * (w_y/wsum) * (exp(x - lse_b) - [j == y])
+__global__ __launch_bounds__(256) void lsm_nll_bwd_kernel(
+    const bf16* __restrict__ logits, const long* __restrict__ label,
+    const float* __restrict__ weight, const float* __restrict__ lse,
+    const float* __restrict__ acc, const float* __restrict__ gscale,
+    bf16* __restrict__ dlogits, int B, long L) {
+  const int b = blockIdx.x;
+  const bf16* row = logits + (long)b * L;
+  bf16* drow = dlogits + (long)b * L;
+  const long y = label[b];
+  const float wy = weight ? weight[y] : 1.0f;
+  const float coef = gscale[0] * wy / acc[1];
+  const float l = lse[b];
+  const long L8 = L & ~7L;
+  for (long j = (long)threadIdx.x * 8; j < L8; j += (long)blockDim.x * 8) {
+    bf16 v[8];
+    *(uint4*)v = *(const uint4*)(row + j);
+    bf16 d[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float g = coef * __expf(bf2f(v[k]) - l);
+      if (j + k == y) g -= coef;
+      d[k] = f2bf(g);
+    }
+    *(uint4*)(drow + j) = *(uint4*)d;
+  }
+  for (long j = L8 + threadIdx.x; j < L; j += blockDim.x) {
+    float g = coef * __expf(bf2f(row[j]) - l);
+    if (j == y) g -= coef;
+    drow[j] = f2bf(g);
+  }
+}
+
+extern "C" {
+
+void launch_lsm_nll_fwd(const void* logits, const long* label,
+                        const float* weight, float* lse, float* acc, int B,
+                        long L, hipStream_t stream) {
+  lsm_nll_fwd_kernel<<<B, 256, 0, stream>>>((const bf16*)logits, label,
+                                            weight, lse, acc, B, L);
+}
+
+void launch_lsm_nll_bwd(const void* logits, const long* label,
+                        const float* weight, const float* lse,
+                        const float* acc, const float* gscale, void* dlogits,
+                        int B, long L, hipStream_t stream) {
+  lsm_nll_bwd_kernel<<<B, 256, 0, stream>>>((const bf16*)logits, label,
+                                            weight, lse, acc, gscale,
+                                            (bf16*)dlogits, B, L);
+}
+
+}  // extern "C"

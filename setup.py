@@ -1,0 +1,38 @@
+"""Build the in-tree HIP extension for gfx950 (MI355X).
+
+  PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+hipcc cross-compiles without a GPU; the built .so lives in
+code2vec_amd/ops/ and travels with the repo snapshot to GPU boxes.
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+SRC = [
+    "code2vec_amd/ops/csrc/bindings.cpp",
+    "code2vec_amd/ops/csrc/gather_concat.hip",
+    "code2vec_amd/ops/csrc/combiner.hip",
+    "code2vec_amd/ops/csrc/attention.hip",
+    "code2vec_amd/ops/csrc/logsoftmax_nll.hip",
+    "code2vec_amd/ops/csrc/adam.hip",
+]
+
+setup(
+    name="code2vec_amd_ext",
+    ext_modules=[
+        CUDAExtension(
+            name="code2vec_amd.ops._c2v_hip",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

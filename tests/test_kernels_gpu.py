@@ -1,0 +1,316 @@
+"""HIP kernel numerics vs the fp32 PyTorch oracle (ops/reference.py).
+
+Every kernel is compared against the exact reference math computed in fp32;
+tolerances account for bf16 storage/compute (rel. Frobenius < ~2%).
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from code2vec_amd.ops import reference as R
+from code2vec_amd.ops import round_up
+
+
+def relerr(a, b):
+    a = a.float()
+    b = b.float()
+    d = (a - b).norm()
+    n = b.norm()
+    return float(d / n) if float(n) > 0 else float(d)
+
+
+@pytest.fixture(scope="module")
+def dev():
+    return torch.device("cuda:0")
+
+
+def make_tables(T, P, dt, dp, dev, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    TS, PS = round_up(dt), round_up(dp)
+    term = torch.zeros(T, TS)
+    path = torch.zeros(P, PS)
+    term[:, :dt] = torch.randn(T, dt, generator=g)
+    path[:, :dp] = torch.randn(P, dp, generator=g)
+    return term.to(dev, torch.bfloat16), path.to(dev, torch.bfloat16), TS, PS
+
+
+def make_batch(B, C, T, P, dev, seed=1, with_pad=True):
+    g = torch.Generator().manual_seed(seed)
+    starts = torch.randint(1, T, (B, C), generator=g, dtype=torch.int32)
+    paths = torch.randint(1, P, (B, C), generator=g, dtype=torch.int32)
+    ends = torch.randint(1, T, (B, C), generator=g, dtype=torch.int32)
+    if with_pad:
+        lens = torch.randint(1, C + 1, (B,), generator=g)
+        mask = torch.arange(C)[None, :] >= lens[:, None]
+        starts[mask] = 0
+        paths[mask] = 0
+        ends[mask] = 0
+    return starts.to(dev), paths.to(dev), ends.to(dev)
+
+
+# ---------------------------------------------------------------------------
+class TestGatherConcat:
+    def test_fwd(self, dev):
+        from code2vec_amd.ops.functional import GatherConcat
+
+        B, C, T, P, dt, dp = 8, 24, 500, 400, 100, 100
+        term, path, TS, PS = make_tables(T, P, dt, dp, dev)
+        starts, paths, ends = make_batch(B, C, T, P, dev)
+        out = GatherConcat.apply(starts, paths, ends, term, path)
+        assert out.shape == (B * C, 2 * TS + PS)
+        # oracle on the padded fp32 tables
+        ref = R.gather_concat(
+            starts, paths, ends, term.float(), path.float()
+        ).view(B * C, -1)
+        assert torch.equal(out.float(), ref)  # pure gather: bitwise
+
+    def test_bwd_matches_autograd(self, dev):
+        from code2vec_amd.ops.functional import GatherConcat
+
+        B, C, T, P, dt, dp = 4, 16, 300, 200, 36, 36
+        term, path, TS, PS = make_tables(T, P, dt, dp, dev)
+        starts, paths, ends = make_batch(B, C, T, P, dev)
+        term_l = term.detach().clone().requires_grad_(True)
+        path_l = path.detach().clone().requires_grad_(True)
+        out = GatherConcat.apply(starts, paths, ends, term_l, path_l)
+        gout = torch.randn_like(out, dtype=torch.float32).to(torch.bfloat16)
+        # zero grads for pad contexts (mirrors real training: attention math
+        # makes them exactly zero; the kernel skips pad rows entirely)
+        gout = gout.view(B, C, -1)
+        gout[(starts == 0)] = 0
+        gout = gout.view(B * C, -1)
+        out.backward(gout)
+
+        term_r = term.float().requires_grad_(True)
+        path_r = path.float().requires_grad_(True)
+        ref = R.gather_concat(starts, paths, ends, term_r, path_r).view(B * C, -1)
+        ref.backward(gout.float())
+        assert relerr(term_l.grad, term_r.grad.to(torch.bfloat16)) < 2e-2
+        assert relerr(path_l.grad, path_r.grad.to(torch.bfloat16)) < 2e-2
+
+
+# ---------------------------------------------------------------------------
+class TestCombiner:
+    def _setup(self, dev, M=512, dt=100, dp=100, E=100, seed=3):
+        g = torch.Generator().manual_seed(seed)
+        TS, PS, EP = round_up(dt), round_up(dp), round_up(E)
+        KP = 2 * TS + PS
+        x = torch.zeros(M, KP)
+        x[:, :dt] = torch.randn(M, dt, generator=g)
+        x[:, TS : TS + dp] = torch.randn(M, dp, generator=g)
+        x[:, TS + PS : TS + PS + dt] = torch.randn(M, dt, generator=g)
+        w = torch.zeros(KP, EP)
+        wl = torch.randn(E, 2 * dt + dp, generator=g) * 0.05
+        w[:dt, :E] = wl[:, :dt].t()
+        w[TS : TS + dp, :E] = wl[:, dt : dt + dp].t()
+        w[TS + PS :TS + PS + dt, :E] = wl[:, dt + dp :].t()
+        gamma = torch.zeros(EP); gamma[:E] = torch.rand(E, generator=g) + 0.5
+        beta = torch.zeros(EP); beta[:E] = torch.randn(E, generator=g) * 0.1
+        return (x.to(dev, torch.bfloat16), w.to(dev, torch.bfloat16),
+                gamma.to(dev), beta.to(dev), wl, E, EP, KP, TS, PS, dt, dp)
+
+    def test_fwd_no_dropout(self, dev):
+        from code2vec_amd.ops.functional import CombinerLNTanh
+
+        x, w, gamma, beta, wl, E, EP, KP, TS, PS, dt, dp = self._setup(dev)
+        out = CombinerLNTanh.apply(x, w, gamma, beta, E, 0.0, False)
+        # fp32 oracle on the same padded operands
+        ref = (x.float() @ w.float())
+        mu = ref[:, :E].mean(dim=1, keepdim=True)
+        var = ref[:, :E].var(dim=1, unbiased=False, keepdim=True)
+        y = torch.tanh((ref[:, :E] - mu) / torch.sqrt(var + 1e-5)
+                       * gamma[:E] + beta[:E])
+        assert relerr(out[:, :E], y) < 2e-2
+        assert torch.all(out[:, E:].float() == 0)
+
+    def test_fwd_matches_reference_op(self, dev):
+        """End math parity: padded kernel output == R.combiner on logical."""
+        from code2vec_amd.ops.functional import CombinerLNTanh
+
+        x, w, gamma, beta, wl, E, EP, KP, TS, PS, dt, dp = self._setup(dev)
+        out = CombinerLNTanh.apply(x, w, gamma, beta, E, 0.0, False)
+        xl = torch.cat(
+            [x.float()[:, :dt], x.float()[:, TS:TS + dp],
+             x.float()[:, TS + PS:TS + PS + dt]], dim=1)
+        ref = R.combiner(xl.unsqueeze(0), wl.to(dev), gamma[:E], beta[:E])[0]
+        assert relerr(out[:, :E], ref) < 2e-2
+
+    def test_dropout_stats_and_determinism(self, dev):
+        from code2vec_amd.ops import functional as Fn
+
+        x, w, gamma, beta, wl, E, EP, *_ = self._setup(dev, M=2048)
+        Fn.reseed_dropout_rng(42)
+        o1 = Fn.CombinerLNTanh.apply(x, w, gamma, beta, E, 0.5, True)
+        Fn.reseed_dropout_rng(42)
+        o2 = Fn.CombinerLNTanh.apply(x, w, gamma, beta, E, 0.5, True)
+        assert torch.equal(o1, o2)
+        keep_rate = (o1[:, :E] != 0).float().mean().item()
+        assert abs(keep_rate - 0.5) < 0.02
+        # kept values scaled by 1/(1-p)
+        o_nodrop = Fn.CombinerLNTanh.apply(x, w, gamma, beta, E, 0.0, False)
+        kept = o1[:, :E] != 0
+        assert relerr(o1[:, :E][kept], (o_nodrop[:, :E] * 2.0)[kept]) < 2e-2
+
+    def test_autograd_full(self, dev):
+        from code2vec_amd.ops.functional import CombinerLNTanh
+
+        x, w, gamma, beta, wl, E, EP, KP, TS, PS, dt, dp = self._setup(dev, M=256)
+        xh = x.detach().clone().requires_grad_(True)
+        wh = w.detach().clone().requires_grad_(True)
+        gh = gamma.detach().clone().requires_grad_(True)
+        bh = beta.detach().clone().requires_grad_(True)
+        out = CombinerLNTanh.apply(xh, wh, gh, bh, E, 0.0, False)
+        gout = (torch.randn_like(out, dtype=torch.float32) * 0.1)
+        gout[:, E:] = 0
+        out.backward(gout.to(torch.bfloat16))
+
+        xr = x.float().detach().requires_grad_(True)
+        wr = w.float().detach().requires_grad_(True)
+        gr = gamma.detach().clone().requires_grad_(True)
+        br = beta.detach().clone().requires_grad_(True)
+        z = xr @ wr
+        mu = z[:, :E].mean(dim=1, keepdim=True)
+        var = z[:, :E].var(dim=1, unbiased=False, keepdim=True)
+        y = torch.tanh((z[:, :E] - mu) / torch.sqrt(var + 1e-5) * gr[:E] + br[:E])
+        y.backward(gout[:, :E].to(torch.bfloat16).float())
+
+        assert relerr(xh.grad, xr.grad) < 4e-2
+        assert relerr(wh.grad[:, :E], wr.grad[:, :E]) < 4e-2
+        assert relerr(gh.grad[:E], gr.grad[:E]) < 4e-2
+        assert relerr(bh.grad[:E], br.grad[:E]) < 4e-2
+        # pad regions of grads are exactly zero
+        assert torch.all(wh.grad.float()[:, E:] == 0)
+        assert torch.all(gh.grad[E:] == 0)
+
+
+# ---------------------------------------------------------------------------
+class TestAttention:
+    def _setup(self, dev, B=16, C=50, E=100, seed=5):
+        g = torch.Generator().manual_seed(seed)
+        EP = round_up(E)
+        ccv = torch.zeros(B, C, EP)
+        ccv[:, :, :E] = torch.tanh(torch.randn(B, C, E, generator=g))
+        a = torch.zeros(EP)
+        a[:E] = torch.randn(E, generator=g) * 0.2
+        starts = torch.randint(1, 100, (B, C), generator=g, dtype=torch.int32)
+        lens = torch.randint(1, C + 1, (B,), generator=g)
+        starts[torch.arange(C)[None, :] >= lens[:, None]] = 0
+        return (ccv.to(dev, torch.bfloat16), a.to(dev), starts.to(dev), E, EP)
+
+    def test_fwd(self, dev):
+        from code2vec_amd.ops.functional import AttentionPool
+
+        ccv, a, starts, E, EP = self._setup(dev)
+        cv, attn = AttentionPool.apply(ccv, a, starts, E)
+        mask = (starts > 0).float()
+        cv_ref, attn_ref = R.attention_code_vector(ccv.float(), a, mask)
+        assert relerr(attn, attn_ref) < 2e-2
+        assert relerr(cv[:, :E], cv_ref[:, :E]) < 2e-2
+        assert torch.all(attn.sum(dim=1).sub(1).abs() < 1e-3)
+
+    def test_fwd_all_pad_row_uniform(self, dev):
+        from code2vec_amd.ops.functional import AttentionPool
+
+        ccv, a, starts, E, EP = self._setup(dev, B=4, C=13)
+        starts = torch.zeros_like(starts)
+        cv, attn = AttentionPool.apply(ccv, a, starts, E)
+        assert torch.allclose(attn, torch.full_like(attn, 1.0 / attn.shape[1]),
+                              atol=1e-5)
+
+    def test_bwd(self, dev):
+        from code2vec_amd.ops.functional import AttentionPool
+
+        ccv, a, starts, E, EP = self._setup(dev, B=8, C=40)
+        ch = ccv.detach().clone().requires_grad_(True)
+        ah = a.detach().clone().requires_grad_(True)
+        cv, attn = AttentionPool.apply(ch, ah, starts, E)
+        dcv = torch.randn_like(cv) * 0.1
+        dcv[:, E:] = 0
+        cv.backward(dcv)
+
+        cr = ccv.float().detach().requires_grad_(True)
+        ar = a.detach().clone().requires_grad_(True)
+        mask = (starts > 0).float()
+        cv_ref, _ = R.attention_code_vector(cr, ar, mask)
+        cv_ref.backward(dcv)
+        assert relerr(ch.grad.float(), cr.grad.to(torch.bfloat16).float()) < 4e-2
+        assert relerr(ah.grad[:E], ar.grad[:E]) < 4e-2
+
+
+# ---------------------------------------------------------------------------
+class TestLogSoftmaxNLL:
+    def test_fwd_bwd(self, dev):
+        from code2vec_amd.ops.functional import FusedLogSoftmaxNLL
+
+        B, L = 64, 7321
+        g = torch.Generator().manual_seed(9)
+        logits32 = torch.randn(B, L, generator=g) * 3
+        label = torch.randint(0, L, (B,), generator=g).to(dev)
+        weight = (torch.rand(L, generator=g) + 0.5).to(dev)
+        lh = logits32.to(dev, torch.bfloat16).requires_grad_(True)
+        loss = FusedLogSoftmaxNLL.apply(lh, label, weight)
+        loss.backward()
+
+        lr = lh.detach().float().requires_grad_(True)
+        ref = R.logsoftmax_nll(lr, label, weight)
+        ref.backward()
+        assert abs(float(loss) - float(ref)) / abs(float(ref)) < 1e-2
+        assert relerr(lh.grad.float(), lr.grad) < 3e-2
+
+    def test_large_vocab(self, dev):
+        from code2vec_amd.ops.functional import FusedLogSoftmaxNLL
+
+        B, L = 16, 261_000
+        logits = (torch.randn(B, L) * 2).to(dev, torch.bfloat16)
+        label = torch.randint(0, L, (B,)).to(dev)
+        weight = torch.ones(L, device=dev)
+        lh = logits.requires_grad_(True)
+        loss = FusedLogSoftmaxNLL.apply(lh, label, weight)
+        loss.backward()
+        ref = R.logsoftmax_nll(logits.float(), label, weight)
+        assert abs(float(loss) - float(ref)) / abs(float(ref)) < 1e-2
+        assert torch.isfinite(lh.grad.float()).all()
+
+
+# ---------------------------------------------------------------------------
+class TestAdam:
+    def test_bf16_master_matches_torch_adam(self, dev):
+        from code2vec_amd.engine.optim import FusedAdam
+
+        g = torch.Generator().manual_seed(11)
+        p32 = torch.randn(4097, generator=g)
+        param = torch.nn.Parameter(p32.to(dev, torch.bfloat16))
+        opt = FusedAdam([param], lr=0.01, betas=(0.9, 0.999), weight_decay=0.01)
+
+        ref_param = torch.nn.Parameter(param.detach().float().clone())
+        ref_opt = torch.optim.Adam([ref_param], lr=0.01, betas=(0.9, 0.999),
+                                   weight_decay=0.01)
+        for step in range(5):
+            grad32 = torch.randn(4097, generator=g).to(dev)
+            param.grad = grad32.to(torch.bfloat16)
+            ref_param.grad = param.grad.float()  # same bf16-rounded grads
+            opt.step()
+            ref_opt.step()
+        master = opt.state[param]["master"]
+        assert relerr(master, ref_param.detach()) < 1e-4
+        assert relerr(param.detach().float(), ref_param.detach()) < 1e-2
+
+    def test_f32(self, dev):
+        from code2vec_amd.engine.optim import FusedAdam
+
+        param = torch.nn.Parameter(torch.randn(1000, device=dev))
+        ref_param = torch.nn.Parameter(param.detach().clone())
+        opt = FusedAdam([param], lr=0.05)
+        ref_opt = torch.optim.Adam([ref_param], lr=0.05)
+        for _ in range(3):
+            grad = torch.randn(1000, device=dev)
+            param.grad = grad.clone()
+            ref_param.grad = grad.clone()
+            opt.step()
+            ref_opt.step()
+        assert relerr(param.detach(), ref_param.detach()) < 1e-5
