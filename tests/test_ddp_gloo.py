@@ -1,0 +1,174 @@
+"""Multi-process CPU (gloo, world_size=2) tests of the DP layer:
+bucketed all-reduce equivalence vs single-process large-batch training."""
+
+import os
+import sys
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _worker(rank, world, port, result_file):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    from code2vec_amd.models.code2vec import Code2VecTorch, init_logical_params
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+    from code2vec_amd.utils.options import Option
+    from code2vec_amd.data.synthetic import synthetic_batch
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    opt = Option(terminal_count=200, path_count=150, label_count=40,
+                 max_path_length=12, terminal_embed_size=16,
+                 path_embed_size=16, encode_size=20, dropout_prob=0.0)
+    g = torch.Generator().manual_seed(7)
+    model = Code2VecTorch(opt, init_logical_params(opt, g))
+    ddp = BucketedAllReduce(list(model.parameters()), world,
+                            bucket_bytes=4096)  # force multiple buckets
+    ddp.broadcast_parameters()
+    optim = torch.optim.Adam(model.parameters(), lr=0.01)
+    w = torch.ones(opt.label_count)
+
+    # each rank gets its own half of a fixed global batch
+    rng = np.random.default_rng(99)
+    s, p, e, y = synthetic_batch(rng, 8 * world, opt.max_path_length,
+                                 opt.terminal_count, opt.path_count,
+                                 opt.label_count)
+    sl = slice(rank * 8, (rank + 1) * 8)
+    s = torch.from_numpy(s[sl]); p = torch.from_numpy(p[sl])
+    e = torch.from_numpy(e[sl]); y = torch.from_numpy(y[sl])
+
+    for step in range(3):
+        ddp.zero_grad()
+        out, _, _ = model(s.long(), p.long(), e.long(), y)
+        loss = model.loss(out, y, w)
+        loss.backward()
+        ddp.finish()
+        optim.step()
+
+    if rank == 0:
+        torch.save({n: p.detach().clone() for n, p in model.named_parameters()},
+                   result_file)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp2_matches_single_process_large_batch(tmp_path):
+    """DP=2 with half batches + grad averaging == single process on the
+    full batch (same init, fp32 => near-bitwise)."""
+    ctx = mp.get_context("spawn")
+    port = 29531
+    result_file = str(tmp_path / "params.pt")
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, result_file))
+             for r in range(2)]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    params = torch.load(result_file, weights_only=True)
+
+    # single-process reference on the concatenated batch
+    from code2vec_amd.models.code2vec import Code2VecTorch, init_logical_params
+    from code2vec_amd.utils.options import Option
+    from code2vec_amd.data.synthetic import synthetic_batch
+
+    opt = Option(terminal_count=200, path_count=150, label_count=40,
+                 max_path_length=12, terminal_embed_size=16,
+                 path_embed_size=16, encode_size=20, dropout_prob=0.0)
+    g = torch.Generator().manual_seed(7)
+    model = Code2VecTorch(opt, init_logical_params(opt, g))
+    optim = torch.optim.Adam(model.parameters(), lr=0.01)
+    w = torch.ones(opt.label_count)
+    rng = np.random.default_rng(99)
+    s, p, e, y = synthetic_batch(rng, 16, opt.max_path_length,
+                                 opt.terminal_count, opt.path_count,
+                                 opt.label_count)
+    s = torch.from_numpy(s); p = torch.from_numpy(p)
+    e = torch.from_numpy(e); y = torch.from_numpy(y)
+    for step in range(3):
+        optim.zero_grad()
+        out, _, _ = model(s.long(), p.long(), e.long(), y)
+        # average of the two half-batch means == full-batch mean (equal sizes)
+        loss = model.loss(out, y, w)
+        loss.backward()
+        optim.step()
+
+    for name, p_ref in model.named_parameters():
+        p_ddp = params[name]
+        assert torch.allclose(p_ddp, p_ref, atol=1e-5), name
+
+
+def _worker_trainer(rank, world, port, tmpdir, result_file):
+    """Full Trainer epoch under gloo world=2 (sharded data, reduced metrics)."""
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    from code2vec_amd.data.builder import DatasetBuilder
+    from code2vec_amd.data.reader import CorpusReader
+    from code2vec_amd.data.synthetic import SyntheticSpec, write_synthetic_corpus
+    from code2vec_amd.engine.trainer import Trainer, TrainerConfig
+    from code2vec_amd.models.code2vec import build_model, init_logical_params
+    from code2vec_amd.parallel.dist import DistContext
+    from code2vec_amd.utils.options import Option
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    files = write_synthetic_corpus(
+        os.path.join(tmpdir, "data"),
+        SyntheticSpec(n_methods=40, n_terminals=60, n_paths=50,
+                      max_contexts=16, seed=5),
+    )
+    reader = CorpusReader(files["corpus_path"], files["path_idx_path"],
+                          files["terminal_idx_path"])
+    opt = Option(terminal_count=len(reader.terminal_vocab),
+                 path_count=len(reader.path_vocab),
+                 label_count=len(reader.label_vocab),
+                 max_path_length=10, terminal_embed_size=12,
+                 path_embed_size=12, encode_size=16, dropout_prob=0.0,
+                 batch_size=8, device=torch.device("cpu"))
+    builder = DatasetBuilder(reader, opt, seed=3, rank=rank, world_size=world)
+    g = torch.Generator().manual_seed(1)
+    model = build_model(opt, backend="torch", logical=init_logical_params(opt, g))
+    ctx = DistContext(rank, world, rank, torch.device("cpu"))
+
+    class A:
+        max_epoch = 2; lr = 0.01; beta_min = 0.9; beta_max = 0.999
+        weight_decay = 0.0; model_path = os.path.join(tmpdir, f"out{rank}")
+        vectors_path = os.path.join(tmpdir, f"out{rank}", "code.vec")
+        test_result_path = None; env = None; print_sample_cycle = 0
+        eval_method = "subtoken"; random_seed = 3; batch_size = 8
+
+    trainer = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
+    obj = trainer.train()
+    if rank == 0:
+        with open(result_file, "w") as f:
+            f.write(str(obj))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_trainer_world2_runs(tmp_path):
+    ctx = mp.get_context("spawn")
+    result_file = str(tmp_path / "obj.txt")
+    procs = [
+        ctx.Process(target=_worker_trainer,
+                    args=(r, 2, 29532, str(tmp_path), result_file))
+        for r in range(2)
+    ]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    obj = float(open(result_file).read())
+    assert 0.0 <= obj <= 1.0
