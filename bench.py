@@ -24,6 +24,13 @@ import os
 import sys
 import time
 
+# hipBLASLt/rocBLAS GEMM autotuning for the plain dgrad/wgrad GEMMs — the
+# tuning happens on first use of each shape, inside the untimed warmup steps.
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                      os.path.join(os.environ.get("TMPDIR", "/tmp"),
+                                   "c2v_tunableop_%d.csv"))
+
 import numpy as np
 import torch
 

@@ -15,34 +15,27 @@ __global__ void adam_bf16_kernel(bf16* __restrict__ p,
                                  float inv_bc2) {
   const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   const long stride = (long)gridDim.x * blockDim.x * 4;
+  typedef float f32x4v __attribute__((ext_vector_type(4)));
+  typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
   for (long i = i0; i < n; i += stride) {
     if (i + 4 <= n) {
-      float4 mm = *(float4*)(m + i);
-      float4 vv = *(float4*)(v + i);
-      float4 pp = *(float4*)(master + i);
-      bf16 gg[4];
-      *(ushort2*)gg = *(const ushort2*)(g + i);
-      *(((ushort2*)gg) + 1) = *(((const ushort2*)(g + i)) + 1);
-      bf16 pout[4];
+      f32x4v mm = *(const f32x4v*)(m + i);
+      f32x4v vv = *(const f32x4v*)(v + i);
+      f32x4v pp = *(const f32x4v*)(master + i);
+      const bf16x4 gg = *(const bf16x4*)(g + i);
+      bf16x4 pout;
 #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        float grad = bf2f(gg[k]);
-        float* mk = k == 0 ? &mm.x : k == 1 ? &mm.y : k == 2 ? &mm.z : &mm.w;
-        float* vk = k == 0 ? &vv.x : k == 1 ? &vv.y : k == 2 ? &vv.z : &vv.w;
-        float* pk = k == 0 ? &pp.x : k == 1 ? &pp.y : k == 2 ? &pp.z : &pp.w;
-        grad += wd * (*pk);
-        *mk = b1 * (*mk) + (1.f - b1) * grad;
-        *vk = b2 * (*vk) + (1.f - b2) * grad * grad;
-        const float mhat = (*mk) * inv_bc1;
-        const float vhat = (*vk) * inv_bc2;
-        *pk -= lr * mhat / (sqrtf(vhat) + eps);
-        pout[k] = f2bf(*pk);
+        float grad = bf2f(gg[k]) + wd * pp[k];
+        mm[k] = b1 * mm[k] + (1.f - b1) * grad;
+        vv[k] = b2 * vv[k] + (1.f - b2) * grad * grad;
+        pp[k] -= lr * (mm[k] * inv_bc1) / (sqrtf(vv[k] * inv_bc2) + eps);
+        pout[k] = f2bf(pp[k]);
       }
-      *(float4*)(m + i) = mm;
-      *(float4*)(v + i) = vv;
-      *(float4*)(master + i) = pp;
-      *(ushort2*)(p + i) = *(ushort2*)pout;
-      *(((ushort2*)(p + i)) + 1) = *(((ushort2*)pout) + 1);
+      *(f32x4v*)(m + i) = mm;
+      *(f32x4v*)(v + i) = vv;
+      *(f32x4v*)(master + i) = pp;
+      *(bf16x4*)(p + i) = pout;
     } else {
       for (long j = i; j < n; ++j) {
         float grad = bf2f(g[j]) + wd * master[j];

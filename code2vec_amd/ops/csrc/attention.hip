@@ -179,7 +179,8 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
       dbase[(long)c * EP] = f2bf(d);
       da_acc += ds[c] * bf2f(base[(long)c * EP]);
     }
-    if (e < E) atomic_add_f32(da + e, da_acc);
+    // per-block partials row (summed host-side; avoids same-address atomics)
+    da[(long)b * EP + e] = e < E ? da_acc : 0.f;
   }
 }
 

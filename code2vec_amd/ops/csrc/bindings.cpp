@@ -18,14 +18,17 @@ void launch_gather_concat_fwd(const int*, const int*, const int*, const void*,
                               const void*, void*, long, int, int, hipStream_t);
 void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
                               float*, float*, long, int, int, hipStream_t);
+void launch_embed_scatter_sorted(const int*, const long*, const void*, float*,
+                                 long, long, int, int, int, int, hipStream_t);
+void launch_count_indices(const int*, int*, long, hipStream_t);
+void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
                          float, unsigned long long, unsigned long long,
                          hipStream_t);
-void launch_combiner_bwd(const void*, const void*, const float*, const float*,
-                         const float*, const float*, void*, float*, float*,
-                         long, int, int, float, unsigned long long,
-                         unsigned long long, hipStream_t);
+void launch_combiner_bwd(const void*, const void*, const void*, const float*,
+                         const float*, const float*, const float*, void*,
+                         float*, float*, long, int, int, float, hipStream_t);
 void launch_attention_fwd(const void*, const float*, const int*, float*,
                           float*, int, int, int, int, hipStream_t);
 void launch_attention_bwd(const float*, const float*, const void*,
@@ -78,6 +81,21 @@ void gather_concat_bwd(torch::Tensor starts, torch::Tensor paths,
                            M, TS, PS, cur_stream());
 }
 
+void embed_scatter_sorted(torch::Tensor sorted_idx, torch::Tensor perm,
+                          torch::Tensor gout, torch::Tensor dtable,
+                          int64_t M, int64_t KP, int64_t off0, int64_t off1) {
+  CHK_CUDA(sorted_idx); CHK_CONTIG(sorted_idx);
+  CHK_DT(sorted_idx, torch::kInt32); CHK_DT(perm, torch::kInt64);
+  CHK_DT(gout, torch::kBFloat16); CHK_CONTIG(gout);
+  CHK_DT(dtable, torch::kFloat32); CHK_CONTIG(dtable);
+  const long N = sorted_idx.numel();
+  const int S = dtable.size(1);
+  launch_embed_scatter_sorted(sorted_idx.data_ptr<int>(),
+                              perm.data_ptr<long>(), gout.data_ptr(),
+                              dtable.data_ptr<float>(), N, M, (int)KP, S,
+                              (int)off0, (int)off1, cur_stream());
+}
+
 void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                   torch::Tensor beta, torch::Tensor out, torch::Tensor z,
                   torch::Tensor mean, torch::Tensor rstd, int64_t E,
@@ -97,20 +115,36 @@ void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                       (unsigned long long)offset, cur_stream());
 }
 
-void combiner_bwd(torch::Tensor dout, torch::Tensor z, torch::Tensor mean,
-                  torch::Tensor rstd, torch::Tensor gamma, torch::Tensor beta,
-                  torch::Tensor dz, torch::Tensor dgamma, torch::Tensor dbeta,
-                  int64_t E, double p, int64_t seed, int64_t offset) {
+void combiner_bwd(torch::Tensor dout, torch::Tensor z, torch::Tensor out,
+                  torch::Tensor mean, torch::Tensor rstd, torch::Tensor gamma,
+                  torch::Tensor beta, torch::Tensor dz, torch::Tensor dgamma,
+                  torch::Tensor dbeta, int64_t E, double p) {
   CHK_CUDA(dout); CHK_CONTIG(dout); CHK_DT(dout, torch::kBFloat16);
-  CHK_CONTIG(z); CHK_DT(dz, torch::kBFloat16);
+  CHK_CONTIG(z); CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  CHK_DT(dz, torch::kBFloat16);
   const long M = z.size(0);
   const int EP = z.size(1);
-  launch_combiner_bwd(dout.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
-                      rstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                      beta.data_ptr<float>(), dz.data_ptr(),
-                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M,
-                      EP, (int)E, (float)p, (unsigned long long)seed,
-                      (unsigned long long)offset, cur_stream());
+  launch_combiner_bwd(dout.data_ptr(), z.data_ptr(), out.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                      dz.data_ptr(), dgamma.data_ptr<float>(),
+                      dbeta.data_ptr<float>(), M, EP, (int)E, (float)p,
+                      cur_stream());
+}
+
+void group_by_index(torch::Tensor idx, torch::Tensor counts,
+                    torch::Tensor cursor, torch::Tensor sorted_idx,
+                    torch::Tensor perm, bool count_only) {
+  CHK_CUDA(idx); CHK_CONTIG(idx); CHK_DT(idx, torch::kInt32);
+  const long N = idx.numel();
+  if (count_only) {
+    launch_count_indices(idx.data_ptr<int>(), counts.data_ptr<int>(), N,
+                         cur_stream());
+  } else {
+    launch_scatter_group(idx.data_ptr<int>(), cursor.data_ptr<int>(),
+                         sorted_idx.data_ptr<int>(), perm.data_ptr<long>(), N,
+                         cur_stream());
+  }
 }
 
 void attention_fwd(torch::Tensor ccv, torch::Tensor a, torch::Tensor starts,
@@ -196,6 +230,8 @@ void adam_step_f32(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_concat_fwd", &gather_concat_fwd);
   m.def("gather_concat_bwd", &gather_concat_bwd);
+  m.def("embed_scatter_sorted", &embed_scatter_sorted);
+  m.def("group_by_index", &group_by_index);
   m.def("combiner_fwd", &combiner_fwd);
   m.def("combiner_bwd", &combiner_bwd);
   m.def("attention_fwd", &attention_fwd);

@@ -16,6 +16,7 @@
 #include "common.h"
 
 #define LN_EPS 1e-5f
+#define COMBINER_BWD_MAX_GRID 1024
 
 // ---------------------------------------------------------------------------
 // Forward.  Block = 256 threads = 4 waves; BM = 128 rows (wave w owns rows
@@ -167,25 +168,43 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
 // ---------------------------------------------------------------------------
 // Backward (elementwise + row-reduce part): given dOut, produce
 // dz = LN/tanh/dropout chain; accumulate dgamma/dbeta.
-// One wave per row; two passes over the row's columns re-reading from cache
-// (rows are 256-640 B -> L1-resident between passes).
-__global__ void combiner_bwd_kernel(
+//
+// Key trick: the forward OUTPUT (post-dropout tanh) is already kept alive by
+// the attention autograd node, so backward recovers both the dropout mask
+// (out != 0 => kept) and the tanh value (y = out*(1-p)) from it — no tanh
+// and no RNG recompute.  (A kept element whose tanh rounds to bf16 zero is
+// treated as dropped; |u| would have to be < ~1e-38 — measure-zero.)
+//
+// One wave per row, single pass: each lane owns column pairs (2 bf16 = 4 B
+// vector loads); dgamma/dbeta accumulate per-lane across the grid-stride
+// loop and flush with one atomic per lane at kernel end.
+template <int NPAIR>  // ceil(EP/128) column-pair iterations per lane
+__global__ __launch_bounds__(256) void combiner_bwd_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ z,
-    const float* __restrict__ mean, const float* __restrict__ rstd,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    bf16* __restrict__ dz, float* __restrict__ dgamma_part,
-    float* __restrict__ dbeta_part, long M, int EP, int E, float p,
-    float inv1mp, unsigned long long seed, unsigned long long offset) {
+    const bf16* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ rstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, bf16* __restrict__ dz,
+    float* __restrict__ dgamma_part, float* __restrict__ dbeta_part, long M,
+    int EP, int E, float p, float inv1mp) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int waves_per_block = blockDim.x / WAVE;
   const float invE = 1.0f / (float)E;
+  const float one_mp = 1.0f - p;  // out -> y scale (1 when p == 0)
 
-  // per-lane dgamma/dbeta accumulators for this lane's columns
-  float acc_dg[8];  // EP/64 <= 8 (EP <= 512)
-  float acc_db[8];
-  const int epc = EP / WAVE >= 1 ? (EP + WAVE - 1) / WAVE : 1;
-  for (int i = 0; i < 8; ++i) { acc_dg[i] = 0.f; acc_db[i] = 0.f; }
+  float g_c[NPAIR][2];
+  float acc_dg[NPAIR][2] = {};
+  float acc_db[NPAIR][2] = {};
+  int cols[NPAIR];
+#pragma unroll
+  for (int i = 0; i < NPAIR; ++i) {
+    cols[i] = i * 128 + lane * 2;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int c = cols[i] + j;
+      g_c[i][j] = c < EP ? gamma[c] : 0.f;
+    }
+  }
 
   long row = (long)blockIdx.x * waves_per_block + wave;
   const long stride = (long)gridDim.x * waves_per_block;
@@ -194,63 +213,83 @@ __global__ void combiner_bwd_kernel(
     const float rs = rstd[row];
     const bf16* drow = dout + row * EP;
     const bf16* zrow = z + row * EP;
+    const bf16* orow = out + row * EP;
     bf16* dzrow = dz + row * EP;
+    float xhat[NPAIR][2], du[NPAIR][2];
     float s1 = 0.f, s2 = 0.f;
-    // pass 1: h = du * gamma sums
-    for (int i = 0; i < epc; ++i) {
-      const int col = lane + i * WAVE;
-      if (col >= E) continue;
-      const float g = gamma[col];
-      const float zv = bf2f(zrow[col]);
-      const float xhat = (zv - mu) * rs;
-      const float u = xhat * g + beta[col];
-      const float y = tanhf(u);
-      float dy = bf2f(drow[col]);
-      if (p > 0.0f) {
-        const float u01 =
-            rng_uniform(seed, offset + (unsigned long long)row * EP + col);
-        dy = (u01 >= p) ? dy * inv1mp : 0.0f;
+#pragma unroll
+    for (int i = 0; i < NPAIR; ++i) {
+      const int c0 = cols[i];
+      bf16x2 zv2 = {bf16(0.f), bf16(0.f)}, dy2 = {bf16(0.f), bf16(0.f)};
+      bf16x2 ov2 = {bf16(0.f), bf16(0.f)};
+      if (c0 < EP) {
+        zv2 = *(const bf16x2*)(zrow + c0);
+        dy2 = *(const bf16x2*)(drow + c0);
+        ov2 = *(const bf16x2*)(orow + c0);
       }
-      const float du = dy * (1.0f - y * y);
-      acc_dg[i] += du * xhat;
-      acc_db[i] += du;
-      const float h = du * g;
-      s1 += h;
-      s2 += h * xhat;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const float xh = (bf2f(zv2[j]) - mu) * rs;
+        const float ov = bf2f(ov2[j]);
+        // kept => y = out*(1-p), dy_eff = dout/(1-p); dropped/pad => 0
+        const float y = ov * one_mp;
+        const float dy = (ov != 0.f) ? bf2f(dy2[j]) * inv1mp : 0.f;
+        const float d = dy * (1.0f - y * y);
+        xhat[i][j] = xh;
+        du[i][j] = d;
+        acc_dg[i][j] += d * xh;
+        acc_db[i][j] += d;
+        const float h = d * g_c[i][j];
+        s1 += h;
+        s2 += h * xh;
+      }
     }
     s1 = wave_reduce_sum(s1) * invE;
     s2 = wave_reduce_sum(s2) * invE;
-    // pass 2: dz
-    for (int i = 0; i < epc; ++i) {
-      const int col = lane + i * WAVE;
-      if (col >= EP) continue;
-      float out_v = 0.f;
-      if (col < E) {
-        const float g = gamma[col];
-        const float zv = bf2f(zrow[col]);
-        const float xhat = (zv - mu) * rs;
-        const float u = xhat * g + beta[col];
-        const float y = tanhf(u);
-        float dy = bf2f(drow[col]);
-        if (p > 0.0f) {
-          const float u01 =
-              rng_uniform(seed, offset + (unsigned long long)row * EP + col);
-          dy = (u01 >= p) ? dy * inv1mp : 0.0f;
-        }
-        const float du = dy * (1.0f - y * y);
-        const float h = du * g;
-        out_v = rs * (h - s1 - xhat * s2);
+#pragma unroll
+    for (int i = 0; i < NPAIR; ++i) {
+      const int c0 = cols[i];
+      if (c0 >= EP) continue;
+      bf16x2 o;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const bool valid = (c0 + j) < E;
+        const float h = du[i][j] * g_c[i][j];
+        o[j] = f2bf(valid ? rs * (h - s1 - xhat[i][j] * s2) : 0.f);
       }
-      dzrow[col] = f2bf(out_v);
+      *(bf16x2*)(dzrow + c0) = o;
     }
   }
-  // flush per-lane dgamma/dbeta partials (one atomic per lane per column)
-  for (int i = 0; i < epc; ++i) {
-    const int col = lane + i * WAVE;
-    if (col < E) {
-      atomic_add_f32(dgamma_part + col, acc_dg[i]);
-      atomic_add_f32(dbeta_part + col, acc_db[i]);
+  // block-combine the per-lane dgamma/dbeta partials in LDS, then one
+  // partials-row store per block (summed host-side; avoids the
+  // all-waves-hit-128-addresses atomic serialization).
+  __shared__ float red_g[256 * 2 * NPAIR];
+  float* red_b = red_g + 128 * NPAIR;
+#pragma unroll
+  for (int i = 0; i < NPAIR; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int c = cols[i] + j;
+      if (wave == 0 && c < 128 * NPAIR) {
+        red_g[c] = 0.f;
+        red_b[c] = 0.f;
+      }
     }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < NPAIR; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int c = cols[i] + j;
+      if (c < EP) {
+        atomicAdd(&red_g[c], acc_dg[i][j]);  // LDS atomics: 4 waves only
+        atomicAdd(&red_b[c], acc_db[i][j]);
+      }
+    }
+  __syncthreads();
+  for (int c = threadIdx.x; c < EP; c += blockDim.x) {
+    dgamma_part[(long)blockIdx.x * EP + c] = red_g[c];
+    dbeta_part[(long)blockIdx.x * EP + c] = red_b[c];
   }
 }
 
@@ -280,19 +319,28 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
 #undef CASE
 }
 
-void launch_combiner_bwd(const void* dout, const void* z, const float* mean,
-                         const float* rstd, const float* gamma,
-                         const float* beta, void* dz, float* dgamma,
-                         float* dbeta, long M, int EP, int E, float p,
-                         unsigned long long seed, unsigned long long offset,
-                         hipStream_t stream) {
+void launch_combiner_bwd(const void* dout, const void* z, const void* out,
+                         const float* mean, const float* rstd,
+                         const float* gamma, const float* beta, void* dz,
+                         float* dgamma, float* dbeta, long M, int EP, int E,
+                         float p, hipStream_t stream) {
   const int block = 256;
   const int wpb = block / WAVE;
-  const int grid = (int)min((M + wpb - 1) / wpb, (long)8192);
+  const int grid = (int)min((M + wpb - 1) / wpb, (long)COMBINER_BWD_MAX_GRID);
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
-  combiner_bwd_kernel<<<grid, block, 0, stream>>>(
-      (const bf16*)dout, (const bf16*)z, mean, rstd, gamma, beta, (bf16*)dz,
-      dgamma, dbeta, M, EP, E, p, inv1mp, seed, offset);
+  const int npair = (EP + 127) / 128;
+#define BCASE(np)                                                             \
+  case np:                                                                    \
+    combiner_bwd_kernel<np><<<grid, block, 0, stream>>>(                      \
+        (const bf16*)dout, (const bf16*)z, (const bf16*)out, mean, rstd,      \
+        gamma, beta, (bf16*)dz, dgamma, dbeta, M, EP, E, p, inv1mp);          \
+    break;
+  switch (npair) {
+    BCASE(1) BCASE(2) BCASE(3) BCASE(4)
+    default:
+      printf("combiner_bwd: unsupported EP=%d\n", EP);
+  }
+#undef BCASE
 }
 
 }  // extern "C"

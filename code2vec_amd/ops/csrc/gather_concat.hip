@@ -83,7 +83,152 @@ __global__ void gather_concat_bwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// K13 v2: sort-based segmented scatter-add (atomic-free on the common path).
+// The caller sorts the flattened index list (torch radix sort) and passes the
+// permutation; each wave owns the runs that start inside its chunk and writes
+// each touched table row with plain f32 stores.  Runs that cross a chunk
+// boundary (heavy-hitter indexes like @question) are split across waves and
+// combined with fp32 atomics — precision identical, order nondeterminism
+// limited to boundary partials.
+//
+// Entry o in [0, N): o < M -> gout row o at column offset off0;
+// o >= M -> gout row (o - M) at offset off1 (start/end entries of the
+// terminal table share one sort; the path table passes N == M).
+template <int NP>  // ceil(S/128) column-pair iterations per lane
+__global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
+    const int* __restrict__ sorted_idx, const long* __restrict__ perm,
+    const bf16* __restrict__ gout, float* __restrict__ dtable, long N, long M,
+    int KP, int S, int off0, int off1, int R) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave_in_block = threadIdx.x / WAVE;
+  const long wid = (long)blockIdx.x * (blockDim.x / WAVE) + wave_in_block;
+  const long c0 = wid * R;
+  if (c0 >= N) return;
+  const long c1 = min(c0 + R, N);
+
+  long t = c0;
+  // skip the head run if it started in an earlier chunk... no: process it,
+  // flagged as boundary-crossing (atomic flush).  Each wave handles exactly
+  // its chunk's entries, so every entry is accumulated exactly once.
+  while (t < c1) {
+    const int idx = sorted_idx[t];
+    const bool head_outside = (t == c0) && (c0 > 0) && (sorted_idx[c0 - 1] == idx);
+    float acc[NP][2];
+#pragma unroll
+    for (int i = 0; i < NP; ++i) acc[i][0] = acc[i][1] = 0.f;
+    long u = t;
+    while (u < c1 && sorted_idx[u] == idx) {
+      const long o = perm[u];
+      const bf16* grow =
+          o < M ? gout + o * KP + off0 : gout + (o - M) * KP + off1;
+#pragma unroll
+      for (int i = 0; i < NP; ++i) {
+        const int col = i * 128 + lane * 2;
+        if (col < S) {
+          const bf16x2 g = *(const bf16x2*)(grow + col);
+          acc[i][0] += bf2f(g[0]);
+          acc[i][1] += bf2f(g[1]);
+        }
+      }
+      ++u;
+    }
+    const bool tail_outside = (u == c1) && (c1 < N) && (sorted_idx[c1] == idx);
+    if (idx != 0) {  // index 0 is <PAD/>: pad contexts carry zero grads
+      float* drow = dtable + (long)idx * S;
+      if (head_outside || tail_outside) {
+#pragma unroll
+        for (int i = 0; i < NP; ++i) {
+          const int col = i * 128 + lane * 2;
+          if (col < S) {
+            atomic_add_f32(drow + col, acc[i][0]);
+            atomic_add_f32(drow + col + 1, acc[i][1]);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < NP; ++i) {
+          const int col = i * 128 + lane * 2;
+          if (col < S) {
+            float2 v = {acc[i][0], acc[i][1]};
+            *(float2*)(drow + col) = v;
+          }
+        }
+      }
+    }
+    t = u;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Counting-sort pair (replaces torch.sort's merge path for grouping index
+// lists by value):
+//  1) histogram: counts[idx] += 1 over all entries,
+//  2) host-side torch.cumsum -> exclusive offsets (cursor),
+//  3) scatter_group: pos = cursor[idx]++ ; sorted_idx[pos] = idx;
+//     perm[pos] = original position.
+// Within-group order is nondeterministic (atomic race), which only permutes
+// fp32 summation order inside a run.
+__global__ void count_indices_kernel(const int* __restrict__ idx,
+                                     int* __restrict__ counts, long N) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < N; t += stride) atomicAdd(&counts[idx[t]], 1);
+}
+
+__global__ void scatter_group_kernel(const int* __restrict__ idx,
+                                     int* __restrict__ cursor,
+                                     int* __restrict__ sorted_idx,
+                                     long* __restrict__ perm, long N) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < N; t += stride) {
+    const int v = idx[t];
+    const int pos = atomicAdd(&cursor[v], 1);
+    sorted_idx[pos] = v;
+    perm[pos] = t;
+  }
+}
+
 extern "C" {
+
+void launch_count_indices(const int* idx, int* counts, long N,
+                          hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((N + block - 1) / block, (long)4096);
+  count_indices_kernel<<<grid, block, 0, stream>>>(idx, counts, N);
+}
+
+void launch_scatter_group(const int* idx, int* cursor, int* sorted_idx,
+                          long* perm, long N, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((N + block - 1) / block, (long)4096);
+  scatter_group_kernel<<<grid, block, 0, stream>>>(idx, cursor, sorted_idx,
+                                                   perm, N);
+}
+
+void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
+                                 const void* gout, float* dtable, long N,
+                                 long M, int KP, int S, int off0, int off1,
+                                 hipStream_t stream) {
+  const int R = 16;  // entries per wave-chunk
+  const long waves = (N + R - 1) / R;
+  const int wpb = 4;
+  const int grid = (int)((waves + wpb - 1) / wpb);
+  const int np = (S + 127) / 128;
+#define SCASE(n)                                                              \
+  case n:                                                                     \
+    embed_scatter_sorted_kernel<n><<<grid, 256, 0, stream>>>(                 \
+        sorted_idx, perm, (const bf16*)gout, dtable, N, M, KP, S, off0,       \
+        off1, R);                                                             \
+    break;
+  switch (np) {
+    SCASE(1) SCASE(2) SCASE(3) SCASE(4)
+    default:
+      printf("embed_scatter_sorted: unsupported S=%d\n", S);
+  }
+#undef SCASE
+}
 
 void launch_gather_concat_fwd(const int* starts, const int* paths,
                               const int* ends, const void* term,

@@ -58,12 +58,40 @@ class GatherConcat(torch.autograd.Function):
     def backward(ctx, grad_out):
         starts, paths, ends = ctx.saved_tensors
         term_shape, path_shape = ctx.shapes
-        dterm32 = torch.zeros(term_shape, dtype=torch.float32, device=grad_out.device)
-        dpath32 = torch.zeros(path_shape, dtype=torch.float32, device=grad_out.device)
-        ext().gather_concat_bwd(
-            starts, paths, ends, grad_out.contiguous(), dterm32, dpath32
-        )
+        dev = grad_out.device
+        M = starts.numel()
+        TS, PS = term_shape[1], path_shape[1]
+        KP = 2 * TS + PS
+        gout = grad_out.contiguous()
+        dterm32 = torch.zeros(term_shape, dtype=torch.float32, device=dev)
+        dpath32 = torch.zeros(path_shape, dtype=torch.float32, device=dev)
+        # sort-based segmented scatter (K13 v2): counting-sort groups the
+        # index lists by value, then run-owner waves write each touched row
+        # once (fp32 atomics only at chunk boundaries of heavy-hitter runs).
+        idx_se = torch.cat([starts.view(-1), ends.view(-1)])
+        sorted_se, perm_se = _group_by_index(idx_se, term_shape[0])
+        ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32,
+                                   M, KP, 0, TS + PS)
+        sorted_p, perm_p = _group_by_index(paths.view(-1), path_shape[0])
+        ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32,
+                                   M, KP, TS, TS)
         return None, None, None, dterm32.to(torch.bfloat16), dpath32.to(torch.bfloat16)
+
+
+def _group_by_index(idx: torch.Tensor, table_rows: int):
+    """Counting sort: returns (sorted_idx i32, perm i64) grouping equal
+    indexes contiguously (ascending)."""
+    N = idx.numel()
+    counts = torch.zeros(table_rows + 1, dtype=torch.int32, device=idx.device)
+    empty_i = torch.empty(0, dtype=torch.int32, device=idx.device)
+    empty_l = torch.empty(0, dtype=torch.int64, device=idx.device)
+    ext().group_by_index(idx, counts, empty_i, empty_i, empty_l, True)
+    cursor = torch.zeros_like(counts)
+    cursor[1:] = torch.cumsum(counts[:-1], 0)
+    sorted_idx = torch.empty(N, dtype=torch.int32, device=idx.device)
+    perm = torch.empty(N, dtype=torch.int64, device=idx.device)
+    ext().group_by_index(idx, counts, cursor, sorted_idx, perm, False)
+    return sorted_idx, perm
 
 
 class CombinerLNTanh(torch.autograd.Function):
@@ -85,22 +113,27 @@ class CombinerLNTanh(torch.autograd.Function):
         p_eff = float(p) if training else 0.0
         seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
         ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff, seed, offset)
-        ctx.save_for_backward(x, w, gamma, beta, z, mean, rstd)
-        ctx.meta = (E, p_eff, seed, offset)
+        ctx.save_for_backward(x, w, gamma, beta, z, mean, rstd, out)
+        ctx.meta = (E, p_eff)
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        x, w, gamma, beta, z, mean, rstd = ctx.saved_tensors
-        E, p, seed, offset = ctx.meta
+        x, w, gamma, beta, z, mean, rstd, out = ctx.saved_tensors
+        E, p = ctx.meta
         M, EP = z.shape
         dz = torch.empty(M, EP, dtype=torch.bfloat16, device=z.device)
-        dgamma = torch.zeros(EP, dtype=torch.float32, device=z.device)
-        dbeta = torch.zeros(EP, dtype=torch.float32, device=z.device)
+        # per-block partials (kernel grid cap = 1024): summed here, which is
+        # deterministic and avoids same-address atomic serialization
+        nblocks = min((M + 4 - 1) // 4, 1024)
+        dgamma_p = torch.empty(nblocks, EP, dtype=torch.float32, device=z.device)
+        dbeta_p = torch.empty(nblocks, EP, dtype=torch.float32, device=z.device)
         ext().combiner_bwd(
-            dout.contiguous(), z, mean, rstd, gamma, beta, dz, dgamma, dbeta,
-            E, p, seed, offset,
+            dout.contiguous(), z, out, mean, rstd, gamma, beta, dz, dgamma_p,
+            dbeta_p, E, p,
         )
+        dgamma = dgamma_p.sum(dim=0)
+        dbeta = dbeta_p.sum(dim=0)
         # plain GEMMs -> rocBLAS
         dx = dz @ w.t()
         dw = x.t() @ dz
@@ -132,15 +165,16 @@ class AttentionPool(torch.autograd.Function):
         if dcv is None:
             dcv = torch.zeros(B, EP, dtype=torch.float32, device=ccv.device)
         dccv = torch.empty_like(ccv)
-        da = torch.zeros(EP, dtype=torch.float32, device=ccv.device)
+        # per-block partials [B, EP], summed here (deterministic, no atomics)
+        da_p = torch.empty(B, EP, dtype=torch.float32, device=ccv.device)
         has_dattn = dattn is not None
         if not has_dattn:
             dattn = torch.empty(0, dtype=torch.float32, device=ccv.device)
         ext().attention_bwd(
             dcv.contiguous(), dattn.contiguous() if has_dattn else dattn,
-            ccv, a, starts, attn, dccv, da, ctx.E, has_dattn,
+            ccv, a, starts, attn, dccv, da_p, ctx.E, has_dattn,
         )
-        return dccv, da, None, None
+        return dccv, da_p.sum(dim=0), None, None
 
 
 class FusedLogSoftmaxNLL(torch.autograd.Function):

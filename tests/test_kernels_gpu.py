@@ -314,3 +314,35 @@ class TestAdam:
             opt.step()
             ref_opt.step()
         assert relerr(param.detach(), ref_param.detach()) < 1e-5
+
+
+class TestCombinerDropoutBwd:
+    def test_autograd_with_dropout(self, dev):
+        """Backward under dropout: the kernel recovers the mask from the
+        forward output; compare against torch autograd using that mask."""
+        from code2vec_amd.ops import functional as Fn
+
+        x, w, gamma, beta, wl, E, EP, KP, TS, PS, dt, dp = \
+            TestCombiner()._setup(dev, M=512)
+        p = 0.5
+        xh = x.detach().clone().requires_grad_(True)
+        wh = w.detach().clone().requires_grad_(True)
+        Fn.reseed_dropout_rng(7)
+        out = Fn.CombinerLNTanh.apply(xh, wh, gamma, beta, E, p, True)
+        gout = (torch.randn_like(out, dtype=torch.float32) * 0.1)
+        gout[:, E:] = 0
+        out.backward(gout.to(torch.bfloat16))
+
+        mask = (out.detach()[:, :E] != 0).float()
+        xr = x.float().detach().requires_grad_(True)
+        wr = w.float().detach().requires_grad_(True)
+        z = xr @ wr
+        mu = z[:, :E].mean(dim=1, keepdim=True)
+        var = z[:, :E].var(dim=1, unbiased=False, keepdim=True)
+        y = torch.tanh((z[:, :E] - mu) / torch.sqrt(var + 1e-5)
+                       * gamma[:E] + beta[:E])
+        out_ref = y * mask / (1 - p)
+        assert relerr(out[:, :E], out_ref) < 2e-2
+        out_ref.backward(gout[:, :E].to(torch.bfloat16).float())
+        assert relerr(xh.grad, xr.grad) < 5e-2
+        assert relerr(wh.grad[:, :E], wr.grad[:, :E]) < 5e-2
