@@ -39,6 +39,8 @@ void launch_lsm_nll_fwd(const void*, const long*, const float*, float*,
 void launch_lsm_nll_bwd(const void*, const long*, const float*, const float*,
                         const float*, const float*, void*, int, long,
                         hipStream_t);
+void launch_wgrad(const void*, const void*, float*, long, int, int, int,
+                  hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
 void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
@@ -205,6 +207,17 @@ void logsoftmax_nll_bwd(torch::Tensor logits, torch::Tensor label,
                      cur_stream());
 }
 
+void wgrad(torch::Tensor x, torch::Tensor dz, torch::Tensor partials) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
+  CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
+  const long M = x.size(0);
+  const int KP = x.size(1), EP = dz.size(1);
+  TORCH_CHECK(partials.size(1) == KP && partials.size(2) == EP, "partials");
+  launch_wgrad(x.data_ptr(), dz.data_ptr(), partials.data_ptr<float>(), M,
+               KP, EP, partials.size(0), cur_stream());
+}
+
 void adam_step_bf16(torch::Tensor p, torch::Tensor g, torch::Tensor master,
                     torch::Tensor m, torch::Tensor v, int64_t step, double lr,
                     double b1, double b2, double eps, double wd) {
@@ -238,6 +251,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_bwd", &attention_bwd);
   m.def("logsoftmax_nll_fwd", &logsoftmax_nll_fwd);
   m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
+  m.def("wgrad", &wgrad);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);
 }

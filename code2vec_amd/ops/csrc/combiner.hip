@@ -71,44 +71,71 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   }
   const int kj = (lane >> 4) * 8;  // this lane's k sub-offset within a K-step
 
-  // stage B K-step kk into buffer buf
-  auto stage_b = [&](int kk, int buf) {
-    const int chunks = 32 * EP / 8;  // 16-B chunks in a [32, EP] W tile
-    for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
-      const int krow = c / (EP / 8);
-      const int col8 = (c % (EP / 8)) * 8;
-      const bf16* src = W + (long)(kk * 32 + krow) * EP + col8;
-      bf16 vals[8];
-      *(uint4*)vals = *(const uint4*)src;
-      const int n = col8 / 16;
-      const int base_l = (col8 & 15) + (krow >> 3) * 16;
-      const int jslot = krow & 7;
-      bf16* dst = lds_b + (((long)buf * NT + n) * 64) * 8;
+  // B-tile staging, T14-split (issue-early / write-late): global loads for
+  // K-step kk+1 are issued before the MFMAs on kk (HBM/L2 latency hides
+  // under compute); the LDS repack writes land after, behind one barrier.
+  constexpr int CPT = (NT + 3) / 4;  // 16-B chunks per thread per K-step
+  uint4 stage_regs[CPT];
+  auto stage_load = [&](int kk) {
+    const int chunks = 32 * EP / 8;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
+    for (int i = 0; i < CPT; ++i) {
+      const int c = threadIdx.x + i * 256;
+      if (c < chunks) {
+        const int krow = c / (EP / 8);
+        const int col8 = (c % (EP / 8)) * 8;
+        stage_regs[i] = *(const uint4*)(W + (long)(kk * 32 + krow) * EP + col8);
+      }
     }
   };
+  auto stage_write = [&](int buf) {
+    const int chunks = 32 * EP / 8;
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      const int c = threadIdx.x + i * 256;
+      if (c < chunks) {
+        const int krow = c / (EP / 8);
+        const int col8 = (c % (EP / 8)) * 8;
+        bf16 vals[8];
+        *(uint4*)vals = stage_regs[i];
+        const int n = col8 / 16;
+        const int base_l = (col8 & 15) + (krow >> 3) * 16;
+        const int jslot = krow & 7;
+        bf16* dst = lds_b + (((long)buf * NT + n) * 64) * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
+      }
+    }
+  };
+  auto load_a = [&](int kk, bf16x8* a) {
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+      a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kk * 32 + kj);
+  };
 
-  stage_b(0, 0);
+  stage_load(0);
+  stage_write(0);
+  bf16x8 a_cur[2], a_next[2];
+  load_a(0, a_cur);
   __syncthreads();
 
   for (int kk = 0; kk < NK; ++kk) {
     const int buf = kk & 1;
-    bf16x8 a[2];
-#pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
-      const bf16* ap = X + arow[mi] * KP + kk * 32 + kj;
-      a[mi] = *(const bf16x8*)ap;
+    if (kk + 1 < NK) {
+      stage_load(kk + 1);
+      load_a(kk + 1, a_next);
     }
 #pragma unroll
     for (int n = 0; n < NT; ++n) {
       const bf16x8 b =
           *(const bf16x8*)(lds_b + (((long)buf * NT + n) * 64 + lane) * 8);
-      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b, acc[0][n], 0, 0, 0);
-      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b, acc[1][n], 0, 0, 0);
+      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur[0], b, acc[0][n], 0, 0, 0);
+      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur[1], b, acc[1][n], 0, 0, 0);
     }
     if (kk + 1 < NK) {
-      stage_b(kk + 1, buf ^ 1);
+      stage_write(buf ^ 1);
+      a_cur[0] = a_next[0];
+      a_cur[1] = a_next[1];
     }
     __syncthreads();
   }

@@ -1,0 +1,146 @@
+// K14 (wgrad): dW[KP, EP] = X^T[KP, M] @ dZ[M, EP] — the combiner weight
+// gradient.  hipBLASLt leaves ~3.5x on the table for this skinny big-K shape
+// (K = M up to 204800), so it is hand-written: split-K over blocks, each
+// block MFMA-accumulates a full [KP, EP] fp32 partial over its row chunk and
+// writes one partials slab (no atomics); host sums slabs (deterministic).
+//
+// Geometry: 512 threads = 8 waves; wave w owns KP-tiles [IT*w, IT*(w+1))
+// where IT = (KP/16)/8; LDS stages per 32-row K-step: the X tile [32, KP]
+// and dZ tile [32, EP], both repacked into MFMA-fragment images.
+//
+// MFMA 16x16x32_bf16 fragment maps (same as combiner.hip):
+//   A: lane l holds A[i = l&15][k = (l>>4)*8 + j]  -> A = X^T: X[m][kp]
+//   B: lane l holds B[k = (l>>4)*8 + j][col = l&15] -> dZ[m][ep]
+//   C/D: lane l, reg r holds C[i = (l>>4)*4 + r][col = l&15]
+
+#include "common.h"
+
+template <int NT, int IT>  // NT = EP/16 col tiles; IT = KP/16/8 row tiles/wave
+__global__ __launch_bounds__(512) void wgrad_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ dZ,
+    float* __restrict__ partials, long M, int KP, int EP, long rows_per_block) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const long m0 = (long)blockIdx.x * rows_per_block;
+  const long m1 = min(m0 + rows_per_block, M);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // fragment images: x_img [KP/16 tiles][64 lanes][8], dz_img [NT][64][8]
+  bf16* x_img = (bf16*)smem;                    // (KP/16)*1024 B
+  bf16* dz_img = x_img + (size_t)(KP / 16) * 64 * 8;
+
+  f32x4 acc[IT][NT];
+#pragma unroll
+  for (int i = 0; i < IT; ++i)
+#pragma unroll
+    for (int n = 0; n < NT; ++n) acc[i][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (long mb = m0; mb < m1; mb += 32) {
+    const int rows = (int)min((long)32, m1 - mb);
+    // stage X rows [mb, mb+32) -> x_img (fragment layout, zero-padded tail)
+    {
+      const int chunks = 32 * KP / 8;
+      for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
+        const int krow = c / (KP / 8);  // m offset within the 32-row step
+        const int col8 = (c % (KP / 8)) * 8;
+        bf16 vals[8];
+        if (krow < rows) {
+          *(uint4*)vals = *(const uint4*)(X + (mb + krow) * KP + col8);
+        } else {
+          uint4 zz = {0, 0, 0, 0};
+          *(uint4*)vals = zz;
+        }
+        const int n = col8 / 16;
+        const int base_l = (col8 & 15) + (krow >> 3) * 16;
+        const int jslot = krow & 7;
+        bf16* dst = x_img + ((size_t)n * 64) * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
+      }
+    }
+    // stage dZ rows -> dz_img
+    {
+      const int chunks = 32 * EP / 8;
+      for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
+        const int krow = c / (EP / 8);
+        const int col8 = (c % (EP / 8)) * 8;
+        bf16 vals[8];
+        if (krow < rows) {
+          *(uint4*)vals = *(const uint4*)(dZ + (mb + krow) * EP + col8);
+        } else {
+          uint4 zz = {0, 0, 0, 0};
+          *(uint4*)vals = zz;
+        }
+        const int n = col8 / 16;
+        const int base_l = (col8 & 15) + (krow >> 3) * 16;
+        const int jslot = krow & 7;
+        bf16* dst = dz_img + ((size_t)n * 64) * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
+      }
+    }
+    __syncthreads();
+    // MFMA: wave w covers KP-tiles [IT*w .. IT*w+IT)
+    bf16x8 bfrag[NT];
+#pragma unroll
+    for (int n = 0; n < NT; ++n)
+      bfrag[n] = *(const bf16x8*)(dz_img + ((size_t)n * 64 + lane) * 8);
+#pragma unroll
+    for (int i = 0; i < IT; ++i) {
+      const int itile = IT * wave + i;
+      if (itile >= KP / 16) break;
+      const bf16x8 a =
+          *(const bf16x8*)(x_img + ((size_t)itile * 64 + lane) * 8);
+#pragma unroll
+      for (int n = 0; n < NT; ++n)
+        acc[i][n] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfrag[n], acc[i][n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // write this block's fp32 partial slab [KP, EP]
+  float* slab = partials + (size_t)blockIdx.x * KP * EP;
+#pragma unroll
+  for (int i = 0; i < IT; ++i) {
+    const int itile = IT * wave + i;
+    if (itile >= KP / 16) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = itile * 16 + (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int n = 0; n < NT; ++n) {
+        const int col = n * 16 + (lane & 15);
+        slab[(size_t)row * EP + col] = acc[i][n][r];
+      }
+    }
+  }
+}
+
+extern "C" {
+
+void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
+                  int KP, int EP, int nblocks, hipStream_t stream) {
+  const long rows_per_block = ((M + nblocks - 1) / nblocks + 31) / 32 * 32;
+  const int smem = (KP / 16) * 1024 + (EP / 16) * 1024;
+  const int NT = EP / 16;
+  const int IT = (KP / 16 + 7) / 8;
+  // supported shapes: KP/16 divisible into 8 waves; dispatch common cases
+#define WCASE(nt, it)                                                        \
+  if (NT == nt && IT == it) {                                                \
+    wgrad_kernel<nt, it><<<nblocks, 512, smem, stream>>>(                    \
+        (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,                \
+        rows_per_block);                                                     \
+    return;                                                                  \
+  }
+  // NT > 8 (encode > 128) overflows the register budget -> callers fall
+  // back to rocBLAS for those shapes (ops/functional.py gates on EP <= 128).
+  WCASE(8, 1) WCASE(8, 2) WCASE(8, 3) WCASE(8, 4)
+  WCASE(6, 1) WCASE(6, 2) WCASE(6, 3) WCASE(6, 4)
+  WCASE(4, 1) WCASE(4, 2) WCASE(4, 3) WCASE(4, 4)
+  WCASE(2, 1) WCASE(2, 2) WCASE(2, 3) WCASE(2, 4)
+#undef WCASE
+  printf("wgrad: unsupported KP=%d EP=%d\n", KP, EP);
+}
+
+}  // extern "C"

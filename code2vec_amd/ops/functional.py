@@ -134,9 +134,19 @@ class CombinerLNTanh(torch.autograd.Function):
         )
         dgamma = dgamma_p.sum(dim=0)
         dbeta = dbeta_p.sum(dim=0)
-        # plain GEMMs -> rocBLAS
+        # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned)
         dx = dz @ w.t()
-        dw = x.t() @ dz
+        # wgrad: custom split-K MFMA kernel for the skinny big-K shape
+        # (hipBLASLt is ~3.5x off there); partial slabs summed here.
+        KP = x.shape[1]
+        if EP <= 128 and KP <= 512:
+            nsplit = 256
+            partials = torch.empty(nsplit, KP, EP, dtype=torch.float32,
+                                   device=x.device)
+            ext().wgrad(x, dz, partials)
+            dw = partials.sum(dim=0).to(torch.bfloat16)
+        else:
+            dw = x.t() @ dz
         return dx, dw, dgamma, dbeta, None, None, None
 
 

@@ -20,6 +20,7 @@ SRC = [
     "code2vec_amd/ops/csrc/attention.hip",
     "code2vec_amd/ops/csrc/logsoftmax_nll.hip",
     "code2vec_amd/ops/csrc/adam.hip",
+    "code2vec_amd/ops/csrc/wgrad.hip",
 ]
 
 setup(
