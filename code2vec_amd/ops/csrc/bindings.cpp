@@ -21,6 +21,8 @@ void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
 void launch_embed_scatter_sorted(const int*, const long*, const void*, float*,
                                  long, long, int, int, int, int, hipStream_t);
 void launch_count_indices(const int*, int*, long, hipStream_t);
+void launch_cast_clear_rows(float*, const int*, void*, long, int,
+                            hipStream_t);
 void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
@@ -96,6 +98,18 @@ void embed_scatter_sorted(torch::Tensor sorted_idx, torch::Tensor perm,
                               perm.data_ptr<long>(), gout.data_ptr(),
                               dtable.data_ptr<float>(), N, M, (int)KP, S,
                               (int)off0, (int)off1, cur_stream());
+}
+
+void cast_clear_rows(torch::Tensor dtable, torch::Tensor counts,
+                     torch::Tensor out) {
+  CHK_CUDA(dtable); CHK_CONTIG(dtable); CHK_DT(dtable, torch::kFloat32);
+  CHK_DT(counts, torch::kInt32); CHK_DT(out, torch::kBFloat16);
+  CHK_CONTIG(out);
+  const long T = dtable.size(0);
+  const int S = dtable.size(1);
+  TORCH_CHECK(counts.numel() >= T, "counts too small");
+  launch_cast_clear_rows(dtable.data_ptr<float>(), counts.data_ptr<int>(),
+                         out.data_ptr(), T, S, cur_stream());
 }
 
 void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
@@ -245,6 +259,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_concat_bwd", &gather_concat_bwd);
   m.def("embed_scatter_sorted", &embed_scatter_sorted);
   m.def("group_by_index", &group_by_index);
+  m.def("cast_clear_rows", &cast_clear_rows);
   m.def("combiner_fwd", &combiner_fwd);
   m.def("combiner_bwd", &combiner_bwd);
   m.def("attention_fwd", &attention_fwd);

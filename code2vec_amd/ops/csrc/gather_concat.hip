@@ -190,7 +190,47 @@ __global__ void scatter_group_kernel(const int* __restrict__ idx,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused cast-and-clear: convert the fp32 scatter scratch to the bf16 grad
+// tensor AND re-zero the touched fp32 rows (so the scratch is reusable next
+// step without a full-buffer memset).  Rows with counts == 0 were never
+// written: emit bf16 zeros without touching the fp32 buffer.
+__global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
+                                       const int* __restrict__ counts,
+                                       bf16* __restrict__ out, long T, int S) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int wpb = blockDim.x / WAVE;
+  long row = (long)blockIdx.x * wpb + wave;
+  const long stride = (long)gridDim.x * wpb;
+  for (; row < T; row += stride) {
+    const bool touched = counts[row] > 0;
+    float* frow = dtable + row * S;
+    bf16* orow = out + row * S;
+    for (int c0 = lane * 2; c0 < S; c0 += WAVE * 2) {
+      bf16x2 o = {bf16(0.f), bf16(0.f)};
+      if (touched) {
+        const float2 v = *(const float2*)(frow + c0);
+        o[0] = f2bf(v.x);
+        o[1] = f2bf(v.y);
+        float2 zz = {0.f, 0.f};
+        *(float2*)(frow + c0) = zz;
+      }
+      *(bf16x2*)(orow + c0) = o;
+    }
+  }
+}
+
 extern "C" {
+
+void launch_cast_clear_rows(float* dtable, const int* counts, void* out,
+                            long T, int S, hipStream_t stream) {
+  const int block = 256;
+  const int wpb = block / WAVE;
+  const int grid = (int)min((T + wpb - 1) / wpb, (long)8192);
+  cast_clear_rows_kernel<<<grid, block, 0, stream>>>(dtable, counts,
+                                                     (bf16*)out, T, S);
+}
 
 void launch_count_indices(const int* idx, int* counts, long N,
                           hipStream_t stream) {

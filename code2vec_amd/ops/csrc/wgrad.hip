@@ -25,9 +25,11 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
   const long m1 = min(m0 + rows_per_block, M);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // fragment images: x_img [KP/16 tiles][64 lanes][8], dz_img [NT][64][8]
-  bf16* x_img = (bf16*)smem;                    // (KP/16)*1024 B
-  bf16* dz_img = x_img + (size_t)(KP / 16) * 64 * 8;
+  // double-buffered fragment images:
+  //   x_img [2][KP/16 tiles][64 lanes][8], dz_img [2][NT][64][8]
+  const size_t xsz = (size_t)(KP / 16) * 64 * 8;
+  bf16* x_img = (bf16*)smem;
+  bf16* dz_img = x_img + 2 * xsz;
 
   f32x4 acc[IT][NT];
 #pragma unroll
@@ -35,67 +37,98 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
 #pragma unroll
     for (int n = 0; n < NT; ++n) acc[i][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (long mb = m0; mb < m1; mb += 32) {
+  // T14 split staging: issue the next K-step's global loads into registers
+  // before this K-step's MFMAs, write them to the other LDS buffer after.
+  constexpr int CX = 4;  // max 16-B x chunks/thread (KP <= 512, 512 thr)
+  uint4 xr[CX];
+  uint4 dzr;  // EP <= 128 -> exactly one dz chunk/thread
+  auto stage_load = [&](long mb) {
     const int rows = (int)min((long)32, m1 - mb);
-    // stage X rows [mb, mb+32) -> x_img (fragment layout, zero-padded tail)
+    const int chunks_x = 32 * KP / 8;
+#pragma unroll
+    for (int i = 0; i < CX; ++i) {
+      const int c = threadIdx.x + i * 512;
+      uint4 v = {0, 0, 0, 0};
+      if (c < chunks_x) {
+        const int krow = c / (KP / 8);
+        if (krow < rows)
+          v = *(const uint4*)(X + (mb + krow) * KP + (c % (KP / 8)) * 8);
+      }
+      xr[i] = v;
+    }
     {
-      const int chunks = 32 * KP / 8;
-      for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
-        const int krow = c / (KP / 8);  // m offset within the 32-row step
+      const int c = threadIdx.x;
+      uint4 v = {0, 0, 0, 0};
+      if (c < 32 * EP / 8) {
+        const int krow = c / (EP / 8);
+        if (krow < rows)
+          v = *(const uint4*)(dZ + (mb + krow) * EP + (c % (EP / 8)) * 8);
+      }
+      dzr = v;
+    }
+  };
+  auto stage_write = [&](int buf) {
+    const int chunks_x = 32 * KP / 8;
+#pragma unroll
+    for (int i = 0; i < CX; ++i) {
+      const int c = threadIdx.x + i * 512;
+      if (c < chunks_x) {
+        const int krow = c / (KP / 8);
         const int col8 = (c % (KP / 8)) * 8;
         bf16 vals[8];
-        if (krow < rows) {
-          *(uint4*)vals = *(const uint4*)(X + (mb + krow) * KP + col8);
-        } else {
-          uint4 zz = {0, 0, 0, 0};
-          *(uint4*)vals = zz;
-        }
+        *(uint4*)vals = xr[i];
         const int n = col8 / 16;
         const int base_l = (col8 & 15) + (krow >> 3) * 16;
         const int jslot = krow & 7;
-        bf16* dst = x_img + ((size_t)n * 64) * 8;
+        bf16* dst = x_img + buf * xsz + ((size_t)n * 64) * 8;
 #pragma unroll
         for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
       }
     }
-    // stage dZ rows -> dz_img
     {
-      const int chunks = 32 * EP / 8;
-      for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
+      const int c = threadIdx.x;
+      if (c < 32 * EP / 8) {
         const int krow = c / (EP / 8);
         const int col8 = (c % (EP / 8)) * 8;
         bf16 vals[8];
-        if (krow < rows) {
-          *(uint4*)vals = *(const uint4*)(dZ + (mb + krow) * EP + col8);
-        } else {
-          uint4 zz = {0, 0, 0, 0};
-          *(uint4*)vals = zz;
-        }
+        *(uint4*)vals = dzr;
         const int n = col8 / 16;
         const int base_l = (col8 & 15) + (krow >> 3) * 16;
         const int jslot = krow & 7;
-        bf16* dst = dz_img + ((size_t)n * 64) * 8;
+        bf16* dst = dz_img + (size_t)buf * NT * 64 * 8 + ((size_t)n * 64) * 8;
 #pragma unroll
         for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
       }
     }
+  };
+
+  if (m0 < m1) {
+    stage_load(m0);
+    stage_write(0);
     __syncthreads();
+  }
+  int buf = 0;
+  for (long mb = m0; mb < m1; mb += 32) {
+    if (mb + 32 < m1) stage_load(mb + 32);
     // MFMA: wave w covers KP-tiles [IT*w .. IT*w+IT)
     bf16x8 bfrag[NT];
 #pragma unroll
     for (int n = 0; n < NT; ++n)
-      bfrag[n] = *(const bf16x8*)(dz_img + ((size_t)n * 64 + lane) * 8);
+      bfrag[n] = *(const bf16x8*)(dz_img + (size_t)buf * NT * 64 * 8 +
+                                  ((size_t)n * 64 + lane) * 8);
 #pragma unroll
     for (int i = 0; i < IT; ++i) {
       const int itile = IT * wave + i;
       if (itile >= KP / 16) break;
-      const bf16x8 a =
-          *(const bf16x8*)(x_img + ((size_t)itile * 64 + lane) * 8);
+      const bf16x8 a = *(const bf16x8*)(x_img + buf * xsz +
+                                        ((size_t)itile * 64 + lane) * 8);
 #pragma unroll
       for (int n = 0; n < NT; ++n)
         acc[i][n] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfrag[n], acc[i][n], 0, 0, 0);
     }
+    if (mb + 32 < m1) stage_write(buf ^ 1);
+    buf ^= 1;
     __syncthreads();
   }
 
@@ -122,7 +155,7 @@ extern "C" {
 void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
                   int KP, int EP, int nblocks, hipStream_t stream) {
   const long rows_per_block = ((M + nblocks - 1) / nblocks + 31) / 32 * 32;
-  const int smem = (KP / 16) * 1024 + (EP / 16) * 1024;
+  const int smem = 2 * ((KP / 16) * 1024 + (EP / 16) * 1024);
   const int NT = EP / 16;
   const int IT = (KP / 16 + 7) / 8;
   // supported shapes: KP/16 divisible into 8 waves; dispatch common cases

@@ -63,24 +63,45 @@ class GatherConcat(torch.autograd.Function):
         TS, PS = term_shape[1], path_shape[1]
         KP = 2 * TS + PS
         gout = grad_out.contiguous()
-        dterm32 = torch.zeros(term_shape, dtype=torch.float32, device=dev)
-        dpath32 = torch.zeros(path_shape, dtype=torch.float32, device=dev)
+        # persistent fp32 scatter scratch — zeroed once; the fused
+        # cast-and-clear kernel re-zeroes exactly the rows it consumed, so
+        # no full-buffer memset per step
+        dterm32 = _scratch_f32(term_shape, dev)
+        dpath32 = _scratch_f32(path_shape, dev)
         # sort-based segmented scatter (K13 v2): counting-sort groups the
         # index lists by value, then run-owner waves write each touched row
         # once (fp32 atomics only at chunk boundaries of heavy-hitter runs).
         idx_se = torch.cat([starts.view(-1), ends.view(-1)])
-        sorted_se, perm_se = _group_by_index(idx_se, term_shape[0])
+        sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
         ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32,
                                    M, KP, 0, TS + PS)
-        sorted_p, perm_p = _group_by_index(paths.view(-1), path_shape[0])
+        sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
         ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32,
                                    M, KP, TS, TS)
-        return None, None, None, dterm32.to(torch.bfloat16), dpath32.to(torch.bfloat16)
+        dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
+        dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
+        ext().cast_clear_rows(dterm32, counts_se, dterm)
+        ext().cast_clear_rows(dpath32, counts_p, dpath)
+        return None, None, None, dterm, dpath
+
+
+_scratch_cache = {}
+
+
+def _scratch_f32(shape, device) -> torch.Tensor:
+    """Persistent fp32 scatter scratch (invariant: all-zero between steps —
+    maintained by cast_clear_rows)."""
+    key = (tuple(shape), str(device))
+    buf = _scratch_cache.get(key)
+    if buf is None:
+        buf = torch.zeros(shape, dtype=torch.float32, device=device)
+        _scratch_cache[key] = buf
+    return buf
 
 
 def _group_by_index(idx: torch.Tensor, table_rows: int):
-    """Counting sort: returns (sorted_idx i32, perm i64) grouping equal
-    indexes contiguously (ascending)."""
+    """Counting sort: returns (sorted_idx i32, perm i64, counts i32)
+    grouping equal indexes contiguously (ascending)."""
     N = idx.numel()
     counts = torch.zeros(table_rows + 1, dtype=torch.int32, device=idx.device)
     empty_i = torch.empty(0, dtype=torch.int32, device=idx.device)
@@ -91,7 +112,7 @@ def _group_by_index(idx: torch.Tensor, table_rows: int):
     sorted_idx = torch.empty(N, dtype=torch.int32, device=idx.device)
     perm = torch.empty(N, dtype=torch.int64, device=idx.device)
     ext().group_by_index(idx, counts, cursor, sorted_idx, perm, False)
-    return sorted_idx, perm
+    return sorted_idx, perm, counts
 
 
 class CombinerLNTanh(torch.autograd.Function):

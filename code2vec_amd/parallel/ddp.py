@@ -1,15 +1,19 @@
-"""Bucketed gradient all-reduce overlapped with backward (K18).
+"""Gradient all-reduce overlapped with backward (K18) — hybrid scheme.
 
-Designed for xGMI's per-link ring bound: gradients live in
-persistent flat bucket buffers (p.grad is a view into its bucket, so
-autograd accumulates in place with zero copies); as backward finishes the
-last grad of a bucket, that bucket's all-reduce launches asynchronously
-(RCCL orders it on its own stream) while earlier layers' backward kernels
-keep running.  ``finish()`` waits for all handles and averages.
+Two modes per parameter, chosen by size:
+- DIRECT (>= direct_threshold bytes — the embedding tables and the output
+  head): the parameter's grad tensor is all-reduced in place as soon as its
+  post-accumulate hook fires.  p.grad stays None between steps
+  (set_to_none), so autograd *assigns* the backward's freshly produced grad
+  tensor with zero copies — no persistent flat buffer, no zeroing pass, no
+  accumulate-add pass over 100M+ elements.
+- BUCKET (small params: LN, attention, bias, combiner weight): grads live
+  as views into a persistent flat buffer; the bucket all-reduces once its
+  last grad lands.
 
-Bucket order follows reverse parameter order (output head first — it is
-both the first grad produced and the largest dense matrix, SURVEY.md §7
-"Hard parts"), one dtype per bucket.
+Designed for xGMI's per-link ring bound: the big tensors
+are each far above RCCL's efficient message size, and the first grads
+produced (output head) overlap with the rest of backward.
 """
 
 from __future__ import annotations
@@ -44,6 +48,7 @@ class BucketedAllReduce:
         params: List[torch.nn.Parameter],
         world_size: int,
         bucket_bytes: int = 32 * 1024 * 1024,
+        direct_threshold: int = 1024 * 1024,
         process_group=None,
         enabled: Optional[bool] = None,
     ) -> None:
@@ -52,7 +57,18 @@ class BucketedAllReduce:
         self.enabled = enabled if enabled is not None else world_size > 1
         self.params = [p for p in params if p.requires_grad]
 
-        # reverse order ~ autograd completion order
+        self.direct_params = [
+            p for p in self.params
+            if p.numel() * p.element_size() >= direct_threshold
+        ]
+        bucket_params = [
+            p for p in self.params
+            if p.numel() * p.element_size() < direct_threshold
+        ]
+        self._direct_set = set(id(p) for p in self.direct_params)
+        self._direct_handles = []
+
+        # reverse order ~ autograd completion order; one dtype per bucket
         self.buckets: List[Bucket] = []
         cur: List[torch.nn.Parameter] = []
         cur_bytes = 0
@@ -64,7 +80,7 @@ class BucketedAllReduce:
                 self.buckets.append(Bucket(cur, cur_dtype, cur[0].device))
             cur, cur_bytes, cur_dtype = [], 0, None
 
-        for p in reversed(self.params):
+        for p in reversed(bucket_params):
             nbytes = p.numel() * p.element_size()
             if cur and (p.dtype != cur_dtype or cur_bytes + nbytes > bucket_bytes):
                 flush()
@@ -77,7 +93,7 @@ class BucketedAllReduce:
         for b in self.buckets:
             for p, view in zip(b.params, b.views):
                 p.grad = view  # autograd accumulates directly into the bucket
-                self._param_bucket[p] = b
+                self._param_bucket[id(p)] = b
 
         self._hooks = []
         if self.enabled:
@@ -90,9 +106,17 @@ class BucketedAllReduce:
         for b in self.buckets:
             b.pending = len(b.params)
             b.handle = None
+        self._direct_handles = []
 
     def _on_grad_ready(self, param) -> None:
-        b = self._param_bucket[param]
+        if id(param) in self._direct_set:
+            h = dist.all_reduce(
+                param.grad, op=dist.ReduceOp.SUM, group=self.group,
+                async_op=True,
+            )
+            self._direct_handles.append((h, param))
+            return
+        b = self._param_bucket[id(param)]
         b.pending -= 1
         if b.pending == 0:
             b.handle = dist.all_reduce(
@@ -103,6 +127,9 @@ class BucketedAllReduce:
         """Wait for outstanding all-reduces and average; call every step."""
         if not self.enabled:
             return
+        for h, p in self._direct_handles:
+            h.wait()
+            p.grad.div_(self.world_size)
         for b in self.buckets:
             if b.handle is not None:
                 b.handle.wait()
@@ -115,6 +142,8 @@ class BucketedAllReduce:
         self._reset_pending()
 
     def zero_grad(self) -> None:
+        for p in self.direct_params:
+            p.grad = None  # next backward assigns the fresh tensor
         for b in self.buckets:
             b.flat.zero_()
         self._reset_pending()

@@ -30,8 +30,10 @@ def _worker(rank, world, port, result_file):
                  path_embed_size=16, encode_size=20, dropout_prob=0.0)
     g = torch.Generator().manual_seed(7)
     model = Code2VecTorch(opt, init_logical_params(opt, g))
+    # small buckets + low direct threshold: exercises BOTH the bucketed
+    # path (LN/attention/bias) and the direct per-tensor path (embeddings)
     ddp = BucketedAllReduce(list(model.parameters()), world,
-                            bucket_bytes=4096)  # force multiple buckets
+                            bucket_bytes=4096, direct_threshold=8192)
     ddp.broadcast_parameters()
     optim = torch.optim.Adam(model.parameters(), lr=0.01)
     w = torch.ones(opt.label_count)
