@@ -153,13 +153,13 @@ class Code2VecHIP(nn.Module):
 
         term = pad2(logical["terminal_embedding"], option.terminal_count, self.TS)
         path = pad2(logical["path_embedding"], option.path_count, self.PS)
-        # combiner weight: logical [E, K=2dt+dp] -> padded B-operand [KP, EP];
-        # logical K segment k lands at padded row seg_base + (k - seg_lo).
+        # combiner weight: logical [E, K=2dt+dp] -> padded TRANSPOSED layout
+        # [EP, KP] (so MFMA B fragments are contiguous; see combiner.hip).
         w_in = logical["input_weight"]  # [E, K]
-        w_pad = torch.zeros(self.KP, self.EP, dtype=torch.float32)
-        w_pad[:dt, :E] = w_in[:, :dt].t()
-        w_pad[self.TS : self.TS + dp, :E] = w_in[:, dt : dt + dp].t()
-        w_pad[self.TS + self.PS : self.TS + self.PS + dt, :E] = w_in[:, dt + dp :].t()
+        w_pad = torch.zeros(self.EP, self.KP, dtype=torch.float32)
+        w_pad[:E, :dt] = w_in[:, :dt]
+        w_pad[:E, self.TS : self.TS + dp] = w_in[:, dt : dt + dp]
+        w_pad[:E, self.TS + self.PS : self.TS + self.PS + dt] = w_in[:, dt + dp :]
 
         w_out = torch.zeros(option.label_count, self.EP, dtype=torch.float32)
         w_out[:, :E] = logical["output_weight"]
@@ -222,12 +222,12 @@ class Code2VecHIP(nn.Module):
         ``code2vec.model`` files are interchangeable."""
         opt = self.option
         dt, dp, E = opt.terminal_embed_size, opt.path_embed_size, opt.encode_size
-        w_pad = self.input_weight.float()
+        w_pad = self.input_weight.float()  # [EP, KP] transposed layout
         w_in = torch.cat(
             [
-                w_pad[:dt, :E].t(),
-                w_pad[self.TS : self.TS + dp, :E].t(),
-                w_pad[self.TS + self.PS : self.TS + self.PS + dt, :E].t(),
+                w_pad[:E, :dt],
+                w_pad[:E, self.TS : self.TS + dp],
+                w_pad[:E, self.TS + self.PS : self.TS + self.PS + dt],
             ],
             dim=1,
         )

@@ -122,8 +122,8 @@ void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
   CHK_DT(out, torch::kBFloat16); CHK_DT(z, torch::kBFloat16);
   CHK_DT(mean, torch::kFloat32); CHK_DT(rstd, torch::kFloat32);
   const long M = x.size(0);
-  const int KP = x.size(1), EP = w.size(1);
-  TORCH_CHECK(w.size(0) == KP && KP % 32 == 0 && EP % 32 == 0, "w shape");
+  const int KP = x.size(1), EP = w.size(0);  // w is TRANSPOSED: [EP, KP]
+  TORCH_CHECK(w.size(1) == KP && KP % 32 == 0 && EP % 32 == 0, "w shape");
   launch_combiner_fwd(x.data_ptr(), w.data_ptr(), gamma.data_ptr<float>(),
                       beta.data_ptr<float>(), out.data_ptr(), z.data_ptr(),
                       mean.data_ptr<float>(), rstd.data_ptr<float>(), M, KP,

@@ -21,8 +21,13 @@
 // ---------------------------------------------------------------------------
 // Forward.  Block = 256 threads = 4 waves; BM = 128 rows (wave w owns rows
 // w*32..w*32+31 as two 16-row MFMA tiles); full EP width per block.
-// A fragments load straight from global (16-B per lane); B tiles are staged
-// per K-step into an LDS image shaped for conflict-free ds_read_b128.
+//
+// Both MFMA operands load straight from memory — no LDS staging, no
+// barriers: A fragments are contiguous 16-B pieces of X rows (HBM, consumed
+// once); B fragments are contiguous 16-B pieces of the TRANSPOSED weight
+// Wt [EP, KP] (<=240 KB, L2-resident, re-read per block).  The transposed
+// parameter layout exists exactly so the B fragment (8 consecutive k at one
+// output column) is memory-contiguous.
 //
 // MFMA fragment maps (gfx950 mfma_f32_16x16x32_bf16):
 //   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
@@ -30,7 +35,7 @@
 //   C/D: lane l, reg r holds C[row = (l>>4)*4 + r][col = l&15]
 template <int NT>  // NT = EP/16 column tiles
 __global__ __launch_bounds__(256) void combiner_fwd_kernel(
-    const bf16* __restrict__ X, const bf16* __restrict__ W,
+    const bf16* __restrict__ X, const bf16* __restrict__ Wt,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     bf16* __restrict__ out, bf16* __restrict__ z_save,
     float* __restrict__ mean_save, float* __restrict__ rstd_save,
@@ -42,16 +47,13 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   const long row0 = (long)blockIdx.x * 128;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered B image: [2][NT][64 lanes][8 bf16]
-  bf16* lds_b = (bf16*)smem;
-  const int bbytes = 2 * NT * 64 * 8 * sizeof(bf16);
-  float* lds_gamma = (float*)(smem + bbytes);
+  float* lds_gamma = (float*)smem;
   float* lds_beta = lds_gamma + EP;
-
   for (int c = threadIdx.x; c < EP; c += blockDim.x) {
     lds_gamma[c] = gamma[c];
     lds_beta[c] = beta[c];
   }
+  __syncthreads();
 
   const int NK = KP / 32;
   f32x4 acc[2][NT];
@@ -60,84 +62,28 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
 #pragma unroll
     for (int n = 0; n < NT; ++n) acc[mi][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // A row this lane reads for tile mi: row0 + wave*32 + mi*16 + (lane&15)
   long arow[2];
-  bool arow_ok[2];
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
     long r = row0 + wave * 32 + mi * 16 + (lane & 15);
-    arow_ok[mi] = r < M;
-    arow[mi] = arow_ok[mi] ? r : (M - 1);
+    arow[mi] = r < M ? r : (M - 1);
   }
   const int kj = (lane >> 4) * 8;  // this lane's k sub-offset within a K-step
+  // B fragment row for tile n: Wt row n*16 + (lane&15)
+  const bf16* wrow = Wt + (size_t)(lane & 15) * KP + kj;
 
-  // B-tile staging, T14-split (issue-early / write-late): global loads for
-  // K-step kk+1 are issued before the MFMAs on kk (HBM/L2 latency hides
-  // under compute); the LDS repack writes land after, behind one barrier.
-  constexpr int CPT = (NT + 3) / 4;  // 16-B chunks per thread per K-step
-  uint4 stage_regs[CPT];
-  auto stage_load = [&](int kk) {
-    const int chunks = 32 * EP / 8;
-#pragma unroll
-    for (int i = 0; i < CPT; ++i) {
-      const int c = threadIdx.x + i * 256;
-      if (c < chunks) {
-        const int krow = c / (EP / 8);
-        const int col8 = (c % (EP / 8)) * 8;
-        stage_regs[i] = *(const uint4*)(W + (long)(kk * 32 + krow) * EP + col8);
-      }
-    }
-  };
-  auto stage_write = [&](int buf) {
-    const int chunks = 32 * EP / 8;
-#pragma unroll
-    for (int i = 0; i < CPT; ++i) {
-      const int c = threadIdx.x + i * 256;
-      if (c < chunks) {
-        const int krow = c / (EP / 8);
-        const int col8 = (c % (EP / 8)) * 8;
-        bf16 vals[8];
-        *(uint4*)vals = stage_regs[i];
-        const int n = col8 / 16;
-        const int base_l = (col8 & 15) + (krow >> 3) * 16;
-        const int jslot = krow & 7;
-        bf16* dst = lds_b + (((long)buf * NT + n) * 64) * 8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
-      }
-    }
-  };
-  auto load_a = [&](int kk, bf16x8* a) {
+  for (int kk = 0; kk < NK; ++kk) {
+    bf16x8 a[2];
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
       a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kk * 32 + kj);
-  };
-
-  stage_load(0);
-  stage_write(0);
-  bf16x8 a_cur[2], a_next[2];
-  load_a(0, a_cur);
-  __syncthreads();
-
-  for (int kk = 0; kk < NK; ++kk) {
-    const int buf = kk & 1;
-    if (kk + 1 < NK) {
-      stage_load(kk + 1);
-      load_a(kk + 1, a_next);
-    }
 #pragma unroll
     for (int n = 0; n < NT; ++n) {
       const bf16x8 b =
-          *(const bf16x8*)(lds_b + (((long)buf * NT + n) * 64 + lane) * 8);
-      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur[0], b, acc[0][n], 0, 0, 0);
-      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur[1], b, acc[1][n], 0, 0, 0);
+          *(const bf16x8*)(wrow + (size_t)n * 16 * KP + kk * 32);
+      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b, acc[0][n], 0, 0, 0);
+      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b, acc[1][n], 0, 0, 0);
     }
-    if (kk + 1 < NK) {
-      stage_write(buf ^ 1);
-      a_cur[0] = a_next[0];
-      a_cur[1] = a_next[1];
-    }
-    __syncthreads();
   }
 
   // ---- epilogue: LayerNorm(E) + tanh + dropout, in-register ----
@@ -330,7 +276,7 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
   const int NT = EP / 16;
   const long grid = (M + 127) / 128;
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
-  const int smem = 2 * NT * 64 * 8 * sizeof(bf16) + 2 * EP * sizeof(float);
+  const int smem = 2 * EP * sizeof(float);
 #define CASE(nt)                                                              \
   case nt:                                                                    \
     combiner_fwd_kernel<nt><<<grid, 256, smem, stream>>>(                     \

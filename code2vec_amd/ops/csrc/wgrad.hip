@@ -27,9 +27,12 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // double-buffered fragment images:
   //   x_img [2][KP/16 tiles][64 lanes][8], dz_img [2][NT][64][8]
-  const size_t xsz = (size_t)(KP / 16) * 64 * 8;
+  // per-n-tile stride padded by 8 elements (16 B) so repack writes spread
+  // across banks (unpadded 1024-B stride puts all 16 writers on one bank)
+  const size_t NSTRIDE = 64 * 8 + 8;
+  const size_t xsz = (size_t)(KP / 16) * NSTRIDE;
   bf16* x_img = (bf16*)smem;
-  bf16* dz_img = x_img + 2 * xsz;
+  bf16* dz_img = x_img + 2 * xsz;  // [2][NT][NSTRIDE]
 
   f32x4 acc[IT][NT];
 #pragma unroll
@@ -80,7 +83,7 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
         const int n = col8 / 16;
         const int base_l = (col8 & 15) + (krow >> 3) * 16;
         const int jslot = krow & 7;
-        bf16* dst = x_img + buf * xsz + ((size_t)n * 64) * 8;
+        bf16* dst = x_img + buf * xsz + (size_t)n * NSTRIDE;
 #pragma unroll
         for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
       }
@@ -95,7 +98,7 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
         const int n = col8 / 16;
         const int base_l = (col8 & 15) + (krow >> 3) * 16;
         const int jslot = krow & 7;
-        bf16* dst = dz_img + (size_t)buf * NT * 64 * 8 + ((size_t)n * 64) * 8;
+        bf16* dst = dz_img + (size_t)buf * NT * NSTRIDE + (size_t)n * NSTRIDE;
 #pragma unroll
         for (int j = 0; j < 8; ++j) dst[(base_l + j) * 8 + jslot] = vals[j];
       }
@@ -114,14 +117,15 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
     bf16x8 bfrag[NT];
 #pragma unroll
     for (int n = 0; n < NT; ++n)
-      bfrag[n] = *(const bf16x8*)(dz_img + (size_t)buf * NT * 64 * 8 +
-                                  ((size_t)n * 64 + lane) * 8);
+      bfrag[n] = *(const bf16x8*)(dz_img + (size_t)buf * NT * NSTRIDE +
+                                  (size_t)n * NSTRIDE + (size_t)lane * 8);
 #pragma unroll
     for (int i = 0; i < IT; ++i) {
       const int itile = IT * wave + i;
       if (itile >= KP / 16) break;
       const bf16x8 a = *(const bf16x8*)(x_img + buf * xsz +
-                                        ((size_t)itile * 64 + lane) * 8);
+                                        (size_t)itile * NSTRIDE +
+                                        (size_t)lane * 8);
 #pragma unroll
       for (int n = 0; n < NT; ++n)
         acc[i][n] =
@@ -155,7 +159,7 @@ extern "C" {
 void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
                   int KP, int EP, int nblocks, hipStream_t stream) {
   const long rows_per_block = ((M + nblocks - 1) / nblocks + 31) / 32 * 32;
-  const int smem = 2 * ((KP / 16) * 1024 + (EP / 16) * 1024);
+  const int smem = 2 * ((KP / 16) + (EP / 16)) * (64 * 8 + 8) * 2;
   const int NT = EP / 16;
   const int IT = (KP / 16 + 7) / 8;
   // supported shapes: KP/16 divisible into 8 waves; dispatch common cases

@@ -118,7 +118,8 @@ def _group_by_index(idx: torch.Tensor, table_rows: int):
 class CombinerLNTanh(torch.autograd.Function):
     """K3-K6: MFMA GEMM (x @ w) -> LayerNorm(E) -> tanh -> dropout, fused.
 
-    x: bf16 [M, KP]; w: bf16 [KP, EP]; gamma/beta: f32 [EP]; E = valid cols.
+    x: bf16 [M, KP]; w: bf16 [EP, KP] (TRANSPOSED so the MFMA B fragment
+    is memory-contiguous); gamma/beta: f32 [EP]; E = valid cols.
     Saves z (pre-LN GEMM output, bf16) + per-row mean/rstd for backward;
     the dropout mask is recomputed from the counter RNG, never stored.
     """
@@ -126,7 +127,7 @@ class CombinerLNTanh(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, gamma, beta, E: int, p: float, training: bool):
         M = x.shape[0]
-        EP = w.shape[1]
+        EP = w.shape[0]
         out = torch.empty(M, EP, dtype=torch.bfloat16, device=x.device)
         z = torch.empty(M, EP, dtype=torch.bfloat16, device=x.device)
         mean = torch.empty(M, dtype=torch.float32, device=x.device)
@@ -155,8 +156,8 @@ class CombinerLNTanh(torch.autograd.Function):
         )
         dgamma = dgamma_p.sum(dim=0)
         dbeta = dbeta_p.sum(dim=0)
-        # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned)
-        dx = dz @ w.t()
+        # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned); w is [EP, KP]
+        dx = dz @ w
         # wgrad: custom split-K MFMA kernel for the skinny big-K shape
         # (hipBLASLt is ~3.5x off there); partial slabs summed here.
         KP = x.shape[1]
@@ -165,9 +166,9 @@ class CombinerLNTanh(torch.autograd.Function):
             partials = torch.empty(nsplit, KP, EP, dtype=torch.float32,
                                    device=x.device)
             ext().wgrad(x, dz, partials)
-            dw = partials.sum(dim=0).to(torch.bfloat16)
+            dw = partials.sum(dim=0).t().contiguous().to(torch.bfloat16)
         else:
-            dw = x.t() @ dz
+            dw = (x.t() @ dz).t().contiguous()
         return dx, dw, dgamma, dbeta, None, None, None
 
 

@@ -104,11 +104,12 @@ class TestCombiner:
         x[:, :dt] = torch.randn(M, dt, generator=g)
         x[:, TS : TS + dp] = torch.randn(M, dp, generator=g)
         x[:, TS + PS : TS + PS + dt] = torch.randn(M, dt, generator=g)
-        w = torch.zeros(KP, EP)
+        # TRANSPOSED weight layout [EP, KP] (see combiner.hip)
+        w = torch.zeros(EP, KP)
         wl = torch.randn(E, 2 * dt + dp, generator=g) * 0.05
-        w[:dt, :E] = wl[:, :dt].t()
-        w[TS : TS + dp, :E] = wl[:, dt : dt + dp].t()
-        w[TS + PS :TS + PS + dt, :E] = wl[:, dt + dp :].t()
+        w[:E, :dt] = wl[:, :dt]
+        w[:E, TS : TS + dp] = wl[:, dt : dt + dp]
+        w[:E, TS + PS : TS + PS + dt] = wl[:, dt + dp :]
         gamma = torch.zeros(EP); gamma[:E] = torch.rand(E, generator=g) + 0.5
         beta = torch.zeros(EP); beta[:E] = torch.randn(E, generator=g) * 0.1
         return (x.to(dev, torch.bfloat16), w.to(dev, torch.bfloat16),
@@ -120,7 +121,7 @@ class TestCombiner:
         x, w, gamma, beta, wl, E, EP, KP, TS, PS, dt, dp = self._setup(dev)
         out = CombinerLNTanh.apply(x, w, gamma, beta, E, 0.0, False)
         # fp32 oracle on the same padded operands
-        ref = (x.float() @ w.float())
+        ref = (x.float() @ w.float().t())
         mu = ref[:, :E].mean(dim=1, keepdim=True)
         var = ref[:, :E].var(dim=1, unbiased=False, keepdim=True)
         y = torch.tanh((ref[:, :E] - mu) / torch.sqrt(var + 1e-5)
@@ -173,18 +174,18 @@ class TestCombiner:
         wr = w.float().detach().requires_grad_(True)
         gr = gamma.detach().clone().requires_grad_(True)
         br = beta.detach().clone().requires_grad_(True)
-        z = xr @ wr
+        z = xr @ wr.t()
         mu = z[:, :E].mean(dim=1, keepdim=True)
         var = z[:, :E].var(dim=1, unbiased=False, keepdim=True)
         y = torch.tanh((z[:, :E] - mu) / torch.sqrt(var + 1e-5) * gr[:E] + br[:E])
         y.backward(gout[:, :E].to(torch.bfloat16).float())
 
         assert relerr(xh.grad, xr.grad) < 4e-2
-        assert relerr(wh.grad[:, :E], wr.grad[:, :E]) < 4e-2
+        assert relerr(wh.grad[:E, :], wr.grad[:E, :]) < 4e-2
         assert relerr(gh.grad[:E], gr.grad[:E]) < 4e-2
         assert relerr(bh.grad[:E], br.grad[:E]) < 4e-2
         # pad regions of grads are exactly zero
-        assert torch.all(wh.grad.float()[:, E:] == 0)
+        assert torch.all(wh.grad.float()[E:, :] == 0)
         assert torch.all(gh.grad[E:] == 0)
 
 
@@ -336,7 +337,7 @@ class TestCombinerDropoutBwd:
         mask = (out.detach()[:, :E] != 0).float()
         xr = x.float().detach().requires_grad_(True)
         wr = w.float().detach().requires_grad_(True)
-        z = xr @ wr
+        z = xr @ wr.t()
         mu = z[:, :E].mean(dim=1, keepdim=True)
         var = z[:, :E].var(dim=1, unbiased=False, keepdim=True)
         y = torch.tanh((z[:, :E] - mu) / torch.sqrt(var + 1e-5)
@@ -345,4 +346,4 @@ class TestCombinerDropoutBwd:
         assert relerr(out[:, :E], out_ref) < 2e-2
         out_ref.backward(gout[:, :E].to(torch.bfloat16).float())
         assert relerr(xh.grad, xr.grad) < 5e-2
-        assert relerr(wh.grad[:, :E], wr.grad[:, :E]) < 5e-2
+        assert relerr(wh.grad[:E, :], wr.grad[:E, :]) < 5e-2

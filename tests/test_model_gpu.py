@@ -103,14 +103,14 @@ def test_pad_regions_stay_zero_after_steps(dev):
         optim.step()
     assert torch.all(hip.terminal_embedding.detach()[:, dt:].float() == 0)
     assert torch.all(hip.path_embedding.detach()[:, dt:].float() == 0)
-    assert torch.all(hip.input_weight.detach()[:, E:].float() == 0)
+    # input_weight is transposed [EP, KP]: pad rows E..EP and pad K columns
+    assert torch.all(hip.input_weight.detach()[E:, :].float() == 0)
     assert torch.all(hip.output_weight.detach()[:, E:].float() == 0)
     assert torch.all(hip.ln_gamma.detach()[E:] == 0)
     assert torch.all(hip.attention_a.detach()[E:] == 0)
-    # pad K rows of the combiner weight (cols dt..TS etc.) stay zero
     TS, PS = hip.TS, hip.PS
-    assert torch.all(hip.input_weight.detach()[dt:TS].float() == 0)
-    assert torch.all(hip.input_weight.detach()[TS + dt:TS + PS].float() == 0)
+    assert torch.all(hip.input_weight.detach()[:, dt:TS].float() == 0)
+    assert torch.all(hip.input_weight.detach()[:, TS + dt:TS + PS].float() == 0)
 
 
 def test_training_reduces_loss_gpu(dev):
