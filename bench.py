@@ -64,6 +64,10 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--config", type=str, default="top11", choices=CONFIGS)
     ap.add_argument("--pool", type=int, default=4, help="synthetic batch pool size")
+    ap.add_argument("--mode", type=str, default="train", choices=["train", "infer"],
+                    help="train: full step (the driver contract); infer: "
+                         "hipGraph-captured batched code-vector export "
+                         "(BASELINE config 5)")
     args = ap.parse_args()
 
     cfg = CONFIGS[args.config]
@@ -105,6 +109,10 @@ def main() -> None:
             torch.from_numpy(a).to(device) for a in (s, p, e, y)
         ))
     class_weight = torch.ones(cfg["label_count"], device=device)
+
+    if args.mode == "infer":
+        run_infer(args, cfg, ctx, model, pool)
+        return
 
     model.train()
 
@@ -168,6 +176,57 @@ def main() -> None:
                 "embed": cfg["embed"],
                 "encode": cfg["encode"],
                 "note": "step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam; top11 label vocab size unpublished, 30k assumed",
+            },
+        }))
+
+
+def run_infer(args, cfg, ctx, model, pool) -> None:
+    """Inference-only: hipGraph-captured forward, code-vector batches/s."""
+    from code2vec_amd.engine.infer import GraphedInference
+
+    device = ctx.device
+    B, C = cfg["batch"], cfg["contexts"]
+    model.eval()
+    graphed = None
+    if device.type == "cuda":
+        graphed = GraphedInference(model, B, device)
+
+    def step(i: int):
+        s, p, e, y = pool[i % len(pool)]
+        if graphed is not None:
+            return graphed.run(s.int(), p.int(), e.int(), y)
+        with torch.no_grad():
+            return model(s, p, e, y)
+
+    for i in range(args.warmup):
+        step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    value = ctx.world_size * B * C * args.steps / elapsed
+    if ctx.is_rank0:
+        print(json.dumps({
+            "metric": "infer_path_contexts_per_sec",
+            "value": value,
+            "unit": "path-contexts/s",
+            "n_gpus": ctx.world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": f"code2vec-{args.config}", "global_batch": B,
+                "seq_len": C, "parallelism": "dp1",
+                "note": "hipGraph-captured batched code-vector export (BASELINE config 5)",
             },
         }))
 

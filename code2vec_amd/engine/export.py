@@ -18,12 +18,15 @@ import torch
 logger = logging.getLogger(__name__)
 
 
-def _forward(model, batch, device):
+def _forward(model, batch, device, graphed=None):
     starts = batch["starts"].to(device)
     paths = batch["paths"].to(device)
     ends = batch["ends"].to(device)
     label = batch["label"].to(device)
-    outputs, code_vector, attn = model(starts, paths, ends, label)
+    if graphed is not None and starts.shape[0] <= graphed.B:
+        outputs, code_vector, attn = graphed.run(starts, paths, ends, label)
+    else:
+        outputs, code_vector, attn = model(starts, paths, ends, label)
     return starts, label, outputs, code_vector, attn
 
 
@@ -36,8 +39,10 @@ def write_code_vectors(
     mode: str,
     test_result_file: Optional[str],
     device,
+    graphed=None,
 ) -> None:
-    """Append one loader's code vectors (reference main.py:393-423)."""
+    """Append one loader's code vectors (reference main.py:393-423).
+    ``graphed``: optional GraphedInference (hipGraph-captured forward)."""
     model.eval()
     itos = reader.label_vocab.itos
     with torch.no_grad():
@@ -46,7 +51,8 @@ def write_code_vectors(
             with open(vector_file, mode) as fv:
                 for batch in loader:
                     ids = batch["id"]
-                    _, label, outputs, code_vector, _ = _forward(model, batch, device)
+                    _, label, outputs, code_vector, _ = _forward(
+                        model, batch, device, graphed)
                     preds_prob, preds_label = torch.max(outputs.float(), dim=1)
                     cv = code_vector.float().cpu()
                     label_cpu = label.cpu()

@@ -304,13 +304,19 @@ class Trainer:
         )
         train_loader = self._make_loader(train=True, epoch=epoch)
         test_loader = self._make_loader(train=False, epoch=epoch)
+        # hipGraph-captured forward for the batched export (config 5)
+        from .infer import graphed_export_forward
+
+        graphed = graphed_export_forward(self.model, self.cfg.batch_size,
+                                         self.device)
         write_code_vectors(
             self.reader, self.model, train_loader, self.option,
-            cfg.vectors_path, "a", None, self.device,
+            cfg.vectors_path, "a", None, self.device, graphed=graphed,
         )
         write_code_vectors(
             self.reader, self.model, test_loader, self.option,
             cfg.vectors_path, "a", cfg.test_result_path, self.device,
+            graphed=graphed,
         )
         if isinstance(self.model, Code2VecHIP):
             sd = self.model.reference_state_dict()

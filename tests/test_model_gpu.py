@@ -139,3 +139,22 @@ def test_native_extension_is_loaded(dev):
     mod = O.ext()
     assert "_c2v_hip" in mod.__file__
     assert "code2vec_amd" in mod.__file__
+
+
+def test_graphed_inference_matches_eager(dev):
+    """hipGraph-captured forward == eager forward (BASELINE config 5)."""
+    from code2vec_amd.engine.infer import GraphedInference
+
+    opt = make_option()
+    hip = Code2VecHIP(opt, device=dev).eval()
+    g = GraphedInference(hip, batch_size=16, device=dev)
+    s, p, e, y = make_inputs(opt, 16, dev, seed=3)
+    out_g, cv_g, attn_g = g.run(s.int(), p.int(), e.int(), y)
+    with torch.no_grad():
+        out_e, cv_e, attn_e = hip(s, p, e, y)
+    assert torch.equal(cv_g, cv_e)
+    assert torch.equal(attn_g, attn_e)
+    assert torch.equal(out_g.float(), out_e.float())
+    # tail batch smaller than capture size
+    out_t, cv_t, attn_t = g.run(s.int()[:5], p.int()[:5], e.int()[:5], y[:5])
+    assert torch.allclose(cv_t, cv_e[:5])
