@@ -28,6 +28,12 @@ from .reader import CodeItem, CorpusReader
 
 logger = logging.getLogger(__name__)
 
+try:  # native (C++/OpenMP) epoch builder — falls back to numpy if absent
+    import torch as _torch
+    from . import _c2v_host as _native
+except Exception:  # noqa: BLE001
+    _native = None
+
 
 @dataclass
 class EpochData:
@@ -82,6 +88,7 @@ class DatasetBuilder:
 
         self.train_dataset: Optional[EpochData] = None
         self.test_dataset: Optional[EpochData] = None
+        self._flat_cache = {}
 
         logger.info("OOV rate: %s", self.out_of_vocabulary_rate())
 
@@ -138,25 +145,84 @@ class DatasetBuilder:
     # ------------------------------------------------------------------
     def refresh_train_dataset(self, epoch: int = 0) -> EpochData:
         """Rebuild the train tensors for ``epoch`` on this rank's shard."""
-        items = self._shard(self.train_items, epoch)
-        self.train_dataset = self.build_data(
-            items, self.option.max_path_length, epoch=epoch, stream=1
-        )
+        self.train_dataset = self._refresh(self.train_items, "train", epoch, 1)
         return self.train_dataset
 
     def refresh_test_dataset(self, epoch: int = 0) -> EpochData:
-        items = self._shard(self.test_items, epoch)
-        self.test_dataset = self.build_data(
-            items, self.option.max_path_length, epoch=epoch, stream=2
-        )
+        self.test_dataset = self._refresh(self.test_items, "test", epoch, 2)
         return self.test_dataset
 
-    def _shard(self, items: List[CodeItem], epoch: int) -> List[CodeItem]:
+    def _refresh(self, base: List[CodeItem], tag: str, epoch: int,
+                 stream: int) -> EpochData:
+        shard_idx = self._shard_indices(len(base), epoch)
+        if (
+            _native is not None
+            and self.reader.infer_method
+            and not self.reader.infer_variable
+        ):
+            return self._build_native(base, tag, shard_idx, epoch, stream)
+        items = [base[i] for i in shard_idx]
+        return self.build_data(items, self.option.max_path_length,
+                               epoch=epoch, stream=stream)
+
+    def _shard_indices(self, n: int, epoch: int) -> np.ndarray:
         if self.world_size <= 1:
-            return items
+            return np.arange(n)
         rng = np.random.default_rng([self.seed, epoch, 0x5A5A])
-        order = rng.permutation(len(items))
-        return [items[i] for i in order[self.rank :: self.world_size]]
+        order = rng.permutation(n)
+        return order[self.rank :: self.world_size]
+
+    # ------------------------------------------------------------------
+    def _get_flat(self, base: List[CodeItem], tag: str):
+        """Flatten a split's contexts once: offsets, [total,3] contexts,
+        labels, ids (feeds the C++ OpenMP epoch builder)."""
+        cached = self._flat_cache.get(tag)
+        if cached is not None:
+            return cached
+        counts = np.fromiter((it.path_contexts.shape[0] for it in base),
+                             count=len(base), dtype=np.int64)
+        offsets = np.zeros(len(base) + 1, dtype=np.int64)
+        np.cumsum(counts, out=offsets[1:])
+        contexts = (
+            np.concatenate([it.path_contexts for it in base], axis=0)
+            if len(base) else np.empty((0, 3), dtype=np.int32)
+        )
+        label_stoi = self.reader.label_vocab.stoi
+        labels = np.fromiter(
+            (label_stoi[it.normalized_label] for it in base),
+            count=len(base), dtype=np.int64,
+        )
+        ids = [it.id for it in base]
+        cached = (
+            _torch.from_numpy(offsets),
+            _torch.from_numpy(np.ascontiguousarray(contexts)),
+            labels,
+            ids,
+        )
+        self._flat_cache[tag] = cached
+        return cached
+
+    def _build_native(self, base, tag, shard_idx, epoch, stream) -> EpochData:
+        offsets_t, contexts_t, labels, ids = self._get_flat(base, tag)
+        C = self.option.max_path_length
+        N = len(shard_idx)
+        item_idx = _torch.from_numpy(np.ascontiguousarray(shard_idx, dtype=np.int64))
+        starts = _torch.empty(N, C, dtype=_torch.int32)
+        paths = _torch.empty(N, C, dtype=_torch.int32)
+        ends = _torch.empty(N, C, dtype=_torch.int32)
+        seed = hash((self.seed, epoch, stream, self.rank)) & 0x7FFFFFFFFFFFFFFF
+        method_token_index = self.reader.terminal_vocab.stoi["@method_0"]
+        _native.build_method_epoch(
+            offsets_t, contexts_t, item_idx, starts, paths, ends,
+            method_token_index, self.reader.QUESTION_TOKEN_INDEX, seed,
+        )
+        return EpochData(
+            ids=[ids[i] for i in shard_idx],
+            starts=starts.numpy(),
+            paths=paths.numpy(),
+            ends=ends.numpy(),
+            labels=labels[shard_idx],
+        )
 
     # ------------------------------------------------------------------
     def build_data(

@@ -11,7 +11,7 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
 
 SRC = [
     "code2vec_amd/ops/csrc/bindings.cpp",
@@ -26,6 +26,12 @@ SRC = [
 setup(
     name="code2vec_amd_ext",
     ext_modules=[
+        CppExtension(
+            name="code2vec_amd.data._c2v_host",
+            sources=["code2vec_amd/data/csrc/epoch_builder.cpp"],
+            extra_compile_args=["-O3", "-std=c++17", "-fopenmp"],
+            extra_link_args=["-fopenmp"],
+        ),
         CUDAExtension(
             name="code2vec_amd.ops._c2v_hip",
             sources=SRC,

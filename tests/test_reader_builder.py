@@ -129,3 +129,49 @@ def test_oov_rate_bounds(tiny_corpus):
     b = DatasetBuilder(r, opt, seed=11)
     rate = b.out_of_vocabulary_rate()
     assert 0.0 <= rate <= 1.0
+
+
+def test_native_builder_matches_python_semantics(tiny_corpus):
+    """The C++ OpenMP epoch builder produces rows that are valid resamples
+    of each item's bag (same labels, contexts drawn w/o replacement, 
+    @method_0 replaced) — structure-equal to the numpy fallback."""
+    from code2vec_amd.data import builder as B
+
+    if B._native is None:
+        pytest.skip("native builder not built")
+    r = make_reader(tiny_corpus)
+    opt = make_option(r, max_path_length=8)
+    b = DatasetBuilder(r, opt, seed=17)
+    native = b.refresh_train_dataset(epoch=0)
+
+    # numpy fallback on the same shard
+    idx = b._shard_indices(len(b.train_items), 0)
+    items = [b.train_items[i] for i in idx]
+    py = b.build_data(items, opt.max_path_length, epoch=0, stream=1)
+
+    assert native.labels.tolist() == py.labels.tolist()
+    assert native.starts.shape == py.starts.shape
+    q = r.QUESTION_TOKEN_INDEX
+    mtok = r.terminal_vocab.stoi["@method_0"]
+    for row, item in zip(range(len(items)), items):
+        pcs = item.path_contexts
+        n = min(pcs.shape[0], opt.max_path_length)
+        # non-pad count matches
+        npad = int((native.paths[row] != 0).sum())
+        assert npad == n or pcs.shape[0] > opt.max_path_length
+        # every sampled triple exists in the item's bag (mod @question swap)
+        allowed = set()
+        for s, p, e in pcs.tolist():
+            s2 = q if s == mtok else s
+            e2 = q if e == mtok else e
+            allowed.add((s2, p, e2))
+        for c in range(n):
+            trip = (int(native.starts[row, c]), int(native.paths[row, c]),
+                    int(native.ends[row, c]))
+            assert trip in allowed
+        # no duplicates (sampling w/o replacement) when bag has no dups
+        trips = [tuple(t) for t in
+                 np.stack([native.starts[row, :n], native.paths[row, :n],
+                           native.ends[row, :n]], axis=1).tolist()]
+        if len(allowed) == pcs.shape[0]:
+            assert len(set(trips)) == len(trips)
