@@ -203,9 +203,8 @@ class Code2VecHIP(nn.Module):
                 self.cos_m, self.sin_m, opt.inverse_temp,
             )
         else:
-            outputs = F.linear(
-                cv.to(torch.bfloat16), self.output_weight,
-                self.output_bias.to(torch.bfloat16),
+            outputs = Fn.OutputHead.apply(
+                cv.to(torch.bfloat16), self.output_weight, self.output_bias
             )
         return outputs, cv[:, : self.E], attn
 

@@ -19,14 +19,15 @@ void launch_gather_concat_fwd(const int*, const int*, const int*, const void*,
 void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
                               float*, float*, long, int, int, hipStream_t);
 void launch_embed_scatter_sorted(const int*, const long*, const void*, float*,
-                                 long, long, int, int, int, int, hipStream_t);
+                                 long, long, int, int, int, int, int,
+                                 hipStream_t);
 void launch_count_indices(const int*, int*, long, hipStream_t);
 void launch_cast_clear_rows(float*, const int*, void*, long, int,
                             hipStream_t);
 void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
-                         float, unsigned long long, unsigned long long,
+                         float, unsigned long long, unsigned long long, int,
                          hipStream_t);
 void launch_combiner_bwd(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*, void*,
@@ -43,6 +44,7 @@ void launch_lsm_nll_bwd(const void*, const long*, const float*, const float*,
                         hipStream_t);
 void launch_wgrad(const void*, const void*, float*, long, int, int, int,
                   hipStream_t);
+void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
 void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
@@ -87,7 +89,8 @@ void gather_concat_bwd(torch::Tensor starts, torch::Tensor paths,
 
 void embed_scatter_sorted(torch::Tensor sorted_idx, torch::Tensor perm,
                           torch::Tensor gout, torch::Tensor dtable,
-                          int64_t M, int64_t KP, int64_t off0, int64_t off1) {
+                          int64_t M, int64_t KP, int64_t off0, int64_t off1,
+                          int64_t R) {
   CHK_CUDA(sorted_idx); CHK_CONTIG(sorted_idx);
   CHK_DT(sorted_idx, torch::kInt32); CHK_DT(perm, torch::kInt64);
   CHK_DT(gout, torch::kBFloat16); CHK_CONTIG(gout);
@@ -97,7 +100,7 @@ void embed_scatter_sorted(torch::Tensor sorted_idx, torch::Tensor perm,
   launch_embed_scatter_sorted(sorted_idx.data_ptr<int>(),
                               perm.data_ptr<long>(), gout.data_ptr(),
                               dtable.data_ptr<float>(), N, M, (int)KP, S,
-                              (int)off0, (int)off1, cur_stream());
+                              (int)off0, (int)off1, (int)R, cur_stream());
 }
 
 void cast_clear_rows(torch::Tensor dtable, torch::Tensor counts,
@@ -115,7 +118,8 @@ void cast_clear_rows(torch::Tensor dtable, torch::Tensor counts,
 void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                   torch::Tensor beta, torch::Tensor out, torch::Tensor z,
                   torch::Tensor mean, torch::Tensor rstd, int64_t E,
-                  double p, int64_t seed, int64_t offset) {
+                  double p, int64_t seed, int64_t offset,
+                  int64_t epilogue_mode) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
   CHK_DT(gamma, torch::kFloat32); CHK_DT(beta, torch::kFloat32);
@@ -128,7 +132,8 @@ void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                       beta.data_ptr<float>(), out.data_ptr(), z.data_ptr(),
                       mean.data_ptr<float>(), rstd.data_ptr<float>(), M, KP,
                       EP, (int)E, (float)p, (unsigned long long)seed,
-                      (unsigned long long)offset, cur_stream());
+                      (unsigned long long)offset, (int)epilogue_mode,
+                      cur_stream());
 }
 
 void combiner_bwd(torch::Tensor dout, torch::Tensor z, torch::Tensor out,
@@ -221,6 +226,13 @@ void logsoftmax_nll_bwd(torch::Tensor logits, torch::Tensor label,
                      cur_stream());
 }
 
+void colsum_bf16(torch::Tensor x, torch::Tensor out) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_DT(out, torch::kFloat32);
+  launch_colsum_bf16(x.data_ptr(), out.data_ptr<float>(), x.size(0),
+                     x.size(1), cur_stream());
+}
+
 void wgrad(torch::Tensor x, torch::Tensor dz, torch::Tensor partials) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
@@ -267,6 +279,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_fwd", &logsoftmax_nll_fwd);
   m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
   m.def("wgrad", &wgrad);
+  m.def("colsum_bf16", &colsum_bf16);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);
 }

@@ -33,7 +33,7 @@
 //   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
 //   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
 //   C/D: lane l, reg r holds C[row = (l>>4)*4 + r][col = l&15]
-template <int NT>  // NT = EP/16 column tiles
+template <int NT, int EPI>  // NT = EP/16 col tiles; EPI: 0 scalar stores, 1 LDS-bounce
 __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ Wt,
     const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -86,8 +86,17 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     }
   }
 
-  // ---- epilogue: LayerNorm(E) + tanh + dropout, in-register ----
+  // ---- epilogue: LayerNorm(E) + tanh + dropout ----
+  // EPI==1: values staged per-wave in LDS (padded row stride) and flushed
+  // as contiguous 16-B chunks — the wave's 32 output rows are contiguous
+  // in memory, so the flush is one coalesced 8-KB span per tensor.
   const float invE = 1.0f / (float)E;
+  bf16* lds_t = nullptr;
+  if (EPI == 1) {
+    lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) +
+            (size_t)wave * 32 * (EP + 8);
+  }
+  float mean_r[2][4], rstd_r[2][4];
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
@@ -105,34 +114,69 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
       const float mean = s1 * invE;
       const float var = fmaxf(s2 * invE - mean * mean, 0.0f);
       const float rstd = rsqrtf(var + LN_EPS);
-      const bool row_ok = row < M;
-      if (row_ok && (lane & 15) == 0) {
+      mean_r[mi][r] = mean;
+      rstd_r[mi][r] = rstd;
+      if (row < M && (lane & 15) == 0) {
         mean_save[row] = mean;
         rstd_save[row] = rstd;
       }
-      if (!row_ok) continue;
-      bf16* zrow = z_save + row * EP;
-      bf16* orow = out + row * EP;
+    }
+  }
+
+  // two value passes: tensor 0 = z (raw GEMM out), tensor 1 = out
+  // (LN+tanh+dropout); each pass stages (EPI==1) or stores scalar (EPI==0).
 #pragma unroll
-      for (int n = 0; n < NT; ++n) {
-        const int col = n * 16 + (lane & 15);
-        const float zv = acc[mi][n][r];
-        float ov = 0.f;
-        float zw = 0.f;
-        if (col < E) {
-          zw = zv;
-          const float xhat = (zv - mean) * rstd;
-          const float u = xhat * lds_gamma[col] + lds_beta[col];
-          float y = tanhf(u);
-          if (p > 0.0f) {
-            const float u01 =
-                rng_uniform(seed, offset + (unsigned long long)row * EP + col);
-            y = (u01 >= p) ? y * inv1mp : 0.0f;
+  for (int pass = 0; pass < 2; ++pass) {
+    bf16* dst_base = pass == 0 ? z_save : out;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row_l = mi * 16 + (lane >> 4) * 4 + r;
+        const long row = row0 + wave * 32 + row_l;
+        const float mean = mean_r[mi][r];
+        const float rstd = rstd_r[mi][r];
+        if (EPI == 0 && row >= M) continue;
+#pragma unroll
+        for (int n = 0; n < NT; ++n) {
+          const int col = n * 16 + (lane & 15);
+          const float zv = acc[mi][n][r];
+          float val = 0.f;
+          if (col < E) {
+            if (pass == 0) {
+              val = zv;
+            } else {
+              const float xhat = (zv - mean) * rstd;
+              const float u = xhat * lds_gamma[col] + lds_beta[col];
+              float y = tanhf(u);
+              if (p > 0.0f) {
+                const float u01 = rng_uniform(
+                    seed, offset + (unsigned long long)row * EP + col);
+                y = (u01 >= p) ? y * inv1mp : 0.0f;
+              }
+              val = y;
+            }
           }
-          ov = y;
+          if (EPI == 1) {
+            lds_t[(size_t)row_l * (EP + 8) + col] = f2bf(val);
+          } else {
+            dst_base[row * EP + col] = f2bf(val);
+          }
         }
-        zrow[col] = f2bf(zw);
-        orow[col] = f2bf(ov);
+      }
+    }
+    if (EPI == 1) {
+      // flush the wave's [32, EP] tile as 16-B chunks (coalesced)
+      const int chunks = 32 * EP / 8;
+      for (int c = lane; c < chunks; c += WAVE) {
+        const int row_l = c / (EP / 8);
+        const int col8 = (c % (EP / 8)) * 8;
+        const long row = row0 + wave * 32 + row_l;
+        if (row < M) {
+          const uint4 v =
+              *(const uint4*)(lds_t + (size_t)row_l * (EP + 8) + col8);
+          *(uint4*)(dst_base + row * EP + col8) = v;
+        }
       }
     }
   }
@@ -272,16 +316,27 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
                          const float* beta, void* out, void* z, float* mean,
                          float* rstd, long M, int KP, int EP, int E, float p,
                          unsigned long long seed, unsigned long long offset,
-                         hipStream_t stream) {
+                         int epilogue_mode, hipStream_t stream) {
   const int NT = EP / 16;
   const long grid = (M + 127) / 128;
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
-  const int smem = 2 * EP * sizeof(float);
+  int epi = epilogue_mode;
+  int smem = 2 * EP * sizeof(float);
+  if (epi == 1) {
+    const int bounce = 4 * 32 * (EP + 8) * (int)sizeof(bf16);
+    if (smem + bounce <= 160 * 1024 - 2048) smem += bounce;
+    else epi = 0;
+  }
 #define CASE(nt)                                                              \
   case nt:                                                                    \
-    combiner_fwd_kernel<nt><<<grid, 256, smem, stream>>>(                     \
-        (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,    \
-        mean, rstd, M, KP, E, p, inv1mp, seed, offset);                       \
+    if (epi == 1)                                                             \
+      combiner_fwd_kernel<nt, 1><<<grid, 256, smem, stream>>>(                \
+          (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset);                     \
+    else                                                                      \
+      combiner_fwd_kernel<nt, 0><<<grid, 256, smem, stream>>>(                \
+          (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset);                     \
     break;
   switch (NT) {
     CASE(2) CASE(4) CASE(6) CASE(8) CASE(10)

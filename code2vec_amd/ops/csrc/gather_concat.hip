@@ -250,8 +250,8 @@ void launch_scatter_group(const int* idx, int* cursor, int* sorted_idx,
 void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
                                  const void* gout, float* dtable, long N,
                                  long M, int KP, int S, int off0, int off1,
-                                 hipStream_t stream) {
-  const int R = 16;  // entries per wave-chunk
+                                 int R, hipStream_t stream) {
+  if (R <= 0) R = 16;  // entries per wave-chunk
   const long waves = (N + R - 1) / R;
   const int wpb = 4;
   const int grid = (int)((waves + wpb - 1) / wpb);

@@ -11,9 +11,15 @@ it in fp32.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
+
+# epilogue store scheme for combiner_fwd: 0 = per-element stores,
+# 1 = per-wave LDS bounce + coalesced 16-B flush (kbench-selected)
+_FWD_EPI = int(os.environ.get("C2V_FWD_EPI", "1"))
+_SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 
 from . import ext
 
@@ -74,10 +80,10 @@ class GatherConcat(torch.autograd.Function):
         idx_se = torch.cat([starts.view(-1), ends.view(-1)])
         sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
         ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32,
-                                   M, KP, 0, TS + PS)
+                                   M, KP, 0, TS + PS, _SCATTER_R)
         sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
         ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32,
-                                   M, KP, TS, TS)
+                                   M, KP, TS, TS, _SCATTER_R)
         dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
         dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
         ext().cast_clear_rows(dterm32, counts_se, dterm)
@@ -134,7 +140,7 @@ class CombinerLNTanh(torch.autograd.Function):
         rstd = torch.empty(M, dtype=torch.float32, device=x.device)
         p_eff = float(p) if training else 0.0
         seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
-        ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff, seed, offset)
+        ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff, seed, offset, _FWD_EPI)
         ctx.save_for_backward(x, w, gamma, beta, z, mean, rstd, out)
         ctx.meta = (E, p_eff)
         return out
@@ -207,6 +213,30 @@ class AttentionPool(torch.autograd.Function):
             ccv, a, starts, attn, dccv, da_p, ctx.E, has_dattn,
         )
         return dccv, da_p.sum(dim=0), None, None
+
+
+class OutputHead(torch.autograd.Function):
+    """K10: label projection logits = cv @ W^T + b with a custom backward:
+    dW/dcv via rocBLAS, dbias via the colsum kernel (torch's reduce is ~4x
+    off HBM bandwidth on [B, 30k+] bf16)."""
+
+    @staticmethod
+    def forward(ctx, cv_bf16, w, bias):
+        logits = torch.nn.functional.linear(
+            cv_bf16, w, bias.to(torch.bfloat16))
+        ctx.save_for_backward(cv_bf16, w)
+        return logits
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        cv, w = ctx.saved_tensors
+        dlogits = dlogits.contiguous()
+        dcv = dlogits @ w                      # [B, EP] bf16
+        dw = dlogits.t() @ cv                  # [L, EP] bf16
+        dbias = torch.zeros(w.shape[0], dtype=torch.float32,
+                            device=w.device)
+        ext().colsum_bf16(dlogits, dbias)
+        return dcv, dw, dbias
 
 
 class FusedLogSoftmaxNLL(torch.autograd.Function):

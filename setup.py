@@ -21,6 +21,7 @@ SRC = [
     "code2vec_amd/ops/csrc/logsoftmax_nll.hip",
     "code2vec_amd/ops/csrc/adam.hip",
     "code2vec_amd/ops/csrc/wgrad.hip",
+    "code2vec_amd/ops/csrc/colsum.hip",
 ]
 
 setup(

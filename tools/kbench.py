@@ -1,0 +1,124 @@
+#!/usr/bin/env python3
+"""Per-op GPU micro-benchmarks (within-process A/B, CUDA events).
+
+Times each framework op at the top11 flagship shape; variants switch via
+env vars read by ops/functional.py.  Usage (on a GPU box):
+    python tools/kbench.py [op ...]
+"""
+
+from __future__ import annotations
+
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from code2vec_amd.ops import functional as Fn
+from code2vec_amd.ops import round_up
+
+
+def timeit(fn, iters=30, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    start.record()
+    for _ in range(iters):
+        fn()
+    end.record()
+    torch.cuda.synchronize()
+    return start.elapsed_time(end) / iters * 1000  # us
+
+
+def main():
+    dev = torch.device("cuda:0")
+    B, C = 1024, 200
+    T, P, L = 360632, 342846, 30000
+    dt = dp = E = 100
+    TS, PS, EP = round_up(dt), round_up(dp), round_up(E)
+    KP = 2 * TS + PS
+    M = B * C
+    g = torch.Generator(device=dev).manual_seed(0)
+
+    term = torch.randn(T, TS, generator=g, device=dev).to(torch.bfloat16)
+    path = torch.randn(P, PS, generator=g, device=dev).to(torch.bfloat16)
+    starts = torch.randint(1, T, (B, C), generator=g, device=dev, dtype=torch.int32)
+    pth = torch.randint(1, P, (B, C), generator=g, device=dev, dtype=torch.int32)
+    ends = torch.randint(1, T, (B, C), generator=g, device=dev, dtype=torch.int32)
+    x = Fn.GatherConcat.apply(starts, pth, ends, term, path)
+    w = (torch.randn(EP, KP, generator=g, device=dev) * 0.05).to(torch.bfloat16)
+    gamma = torch.rand(EP, generator=g, device=dev) + 0.5
+    beta = torch.randn(EP, generator=g, device=dev) * 0.1
+    a = torch.randn(EP, generator=g, device=dev) * 0.2
+    wout = (torch.randn(L, EP, generator=g, device=dev) * 0.05).to(torch.bfloat16)
+    label = torch.randint(0, L, (B,), generator=g, device=dev)
+    weight = torch.ones(L, device=dev)
+
+    ops = sys.argv[1:] or ["gather_fwd", "gather_bwd", "combiner_fwd",
+                           "combiner_bwd", "attention", "lsm", "adam",
+                           "wgrad"]
+    results = {}
+
+    if "gather_fwd" in ops:
+        results["gather_fwd"] = timeit(
+            lambda: Fn.GatherConcat.apply(starts, pth, ends, term, path))
+
+    if "gather_bwd" in ops:
+        xg = None
+        def run():
+            nonlocal xg
+            t2 = term.detach().requires_grad_(True)
+            p2 = path.detach().requires_grad_(True)
+            o = Fn.GatherConcat.apply(starts, pth, ends, t2, p2)
+            o.backward(x)
+        results["gather_bwd(all)"] = timeit(run, iters=15)
+
+    if "combiner_fwd" in ops:
+        results["combiner_fwd"] = timeit(
+            lambda: Fn.CombinerLNTanh.apply(x, w, gamma, beta, E, 0.25, True))
+
+    if "combiner_bwd" in ops:
+        def run():
+            x2 = x.detach().requires_grad_(True)
+            w2 = w.detach().requires_grad_(True)
+            o = Fn.CombinerLNTanh.apply(x2, w2, gamma, beta, E, 0.0, False)
+            o.backward(o.detach())
+        results["combiner_fwd+bwd(all)"] = timeit(run, iters=15)
+
+    ccv = Fn.CombinerLNTanh.apply(x, w, gamma, beta, E, 0.0, False).view(B, C, EP)
+    if "attention" in ops:
+        results["attention_fwd"] = timeit(
+            lambda: Fn.AttentionPool.apply(ccv, a, starts, E))
+
+    if "lsm" in ops:
+        cv, _ = Fn.AttentionPool.apply(ccv, a, starts, E)
+        logits = (cv.to(torch.bfloat16) @ wout.t()).contiguous()
+        results["lsm_fwd"] = timeit(
+            lambda: Fn.FusedLogSoftmaxNLL.apply(logits, label, weight))
+
+    if "adam" in ops:
+        p1 = term.clone().view(-1)
+        g1 = term.clone().view(-1)
+        master = p1.float()
+        m = torch.zeros_like(master)
+        v = torch.zeros_like(master)
+        from code2vec_amd.ops import ext
+        results["adam(term)"] = timeit(
+            lambda: ext().adam_step_bf16(p1, g1, master, m, v, 5, 0.01,
+                                         0.9, 0.999, 1e-8, 0.0))
+
+    if "wgrad" in ops:
+        dz = ccv.view(M, EP).contiguous()
+        partials = torch.empty(256, KP, EP, dtype=torch.float32, device=dev)
+        from code2vec_amd.ops import ext
+        results["wgrad"] = timeit(lambda: ext().wgrad(x, dz, partials))
+
+    for k, vv in results.items():
+        print(f"{k:24s} {vv:10.1f} us")
+
+
+if __name__ == "__main__":
+    main()
