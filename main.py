@@ -130,7 +130,13 @@ def train(args) -> None:
 
 def find_optimal_hyperparams(args) -> None:
     """Optuna HPO (reference main.py:429-488)."""
-    import optuna
+    try:
+        import optuna
+    except ImportError as e:
+        raise SystemExit(
+            "--find_hyperparams needs the optuna package, which is not "
+            "installed in this environment"
+        ) from e
 
     ctx = make_context(args)
     reader, option, builder = prepare(args, ctx)

@@ -158,3 +158,59 @@ def test_graphed_inference_matches_eager(dev):
     # tail batch smaller than capture size
     out_t, cv_t, attn_t = g.run(s.int()[:5], p.int()[:5], e.int()[:5], y[:5])
     assert torch.allclose(cv_t, cv_e[:5])
+
+
+def test_cli_end_to_end_gpu(tmp_path_factory):
+    """Full main.py train on GPU (HIP backend): 2 epochs on a synthetic
+    corpus, artifact formats, finite metrics."""
+    import main as cli
+    from code2vec_amd.data.synthetic import SyntheticSpec, write_synthetic_corpus
+
+    tmp_path = tmp_path_factory.mktemp("gpu_cli")
+    files = write_synthetic_corpus(
+        str(tmp_path / "data"),
+        SyntheticSpec(n_methods=64, n_terminals=120, n_paths=90,
+                      max_contexts=30, seed=9),
+    )
+    out_dir = tmp_path / "out"
+    cli.main([
+        "--corpus_path", files["corpus_path"],
+        "--path_idx_path", files["path_idx_path"],
+        "--terminal_idx_path", files["terminal_idx_path"],
+        "--model_path", str(out_dir),
+        "--vectors_path", str(out_dir / "code.vec"),
+        "--max_epoch", "2", "--batch_size", "16",
+        "--terminal_embed_size", "100", "--path_embed_size", "100",
+        "--encode_size", "100", "--max_path_length", "16",
+        "--print_sample_cycle", "0",
+    ])
+    lines = (out_dir / "code.vec").read_text().splitlines()
+    assert len(lines) == 1 + 64
+    count, esz = lines[0].split("\t")
+    assert (int(count), int(esz)) == (64, 100)
+    for row in lines[1:3]:
+        _, vec = row.split("\t")
+        vals = [float(v) for v in vec.split(" ")]
+        assert len(vals) == 100
+        assert all(v == v for v in vals)  # finite
+    sd = torch.load(out_dir / "code2vec.model", weights_only=True)
+    assert sd["terminal_embedding.weight"].shape[1] == 100
+    assert sd["input_linear.weight"].shape == (100, 300)
+
+
+def test_angular_margin_gpu(dev):
+    """Optional ArcFace head path on the HIP backend (reference
+    model/model.py:71-80)."""
+    opt = make_option(angular_margin_loss=True)
+    g = torch.Generator().manual_seed(6)
+    logical = init_logical_params(opt, g)
+    hip = Code2VecHIP(opt, logical, device=dev).train()
+    ref = Code2VecTorch(opt, logical).to(dev).train()
+    s, p, e, y = make_inputs(opt, 16, dev, seed=4)
+    w = torch.ones(opt.label_count, device=dev)
+    out_h, _, _ = hip(s, p, e, y)
+    out_r, _, _ = ref(s.long(), p.long(), e.long(), y)
+    assert relerr(out_h, out_r) < 6e-2
+    loss_h = hip.loss(out_h, y, w)
+    loss_h.backward()
+    assert torch.isfinite(hip.output_weight.grad.float()).all()

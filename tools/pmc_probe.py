@@ -1,0 +1,79 @@
+#!/usr/bin/env python3
+"""Minimal-dispatch probe for PMC counter collection (rocprofv3 --pmc ...).
+
+Runs each hot kernel a few times at the top11 flagship shape with as few
+auxiliary dispatches as possible (PMC collection serializes and replays
+every dispatch, so bench.py under --pmc is prohibitively slow).
+"""
+
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from code2vec_amd.ops import ext
+from code2vec_amd.ops import functional as Fn
+from code2vec_amd.ops import round_up
+
+dev = torch.device("cuda:0")
+B, C = 1024, 200
+T, P = 360632, 342846
+dt = dp = E = 100
+TS, PS, EP = round_up(dt), round_up(dp), round_up(E)
+KP = 2 * TS + PS
+M = B * C
+REPS = 4
+
+g = torch.Generator(device=dev).manual_seed(0)
+term = torch.randn(T, TS, generator=g, device=dev).to(torch.bfloat16)
+path = torch.randn(P, PS, generator=g, device=dev).to(torch.bfloat16)
+starts = torch.randint(1, T, (B, C), generator=g, device=dev, dtype=torch.int32)
+pth = torch.randint(1, P, (B, C), generator=g, device=dev, dtype=torch.int32)
+ends = torch.randint(1, T, (B, C), generator=g, device=dev, dtype=torch.int32)
+w = (torch.randn(EP, KP, generator=g, device=dev) * 0.05).to(torch.bfloat16)
+gamma = torch.rand(EP, generator=g, device=dev) + 0.5
+beta = torch.randn(EP, generator=g, device=dev) * 0.1
+a = torch.randn(EP, generator=g, device=dev) * 0.2
+
+x = torch.empty(M, KP, dtype=torch.bfloat16, device=dev)
+out = torch.empty(M, EP, dtype=torch.bfloat16, device=dev)
+z = torch.empty(M, EP, dtype=torch.bfloat16, device=dev)
+mean = torch.empty(M, dtype=torch.float32, device=dev)
+rstd = torch.empty(M, dtype=torch.float32, device=dev)
+cv = torch.empty(B, EP, dtype=torch.float32, device=dev)
+attn = torch.empty(B, C, dtype=torch.float32, device=dev)
+partials = torch.empty(256, KP, EP, dtype=torch.float32, device=dev)
+
+for _ in range(REPS):
+    ext().gather_concat_fwd(starts, pth, ends, term, path, x)
+for _ in range(REPS):
+    ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, 0.0, 0, 0, 1)
+ccv = out.view(B, C, EP)
+for _ in range(REPS):
+    ext().attention_fwd(ccv, a, starts, cv, attn, E)
+for _ in range(REPS):
+    ext().wgrad(x, out, partials)
+
+# scatter path (pre-grouped outside the timed kernels)
+idx_se = torch.cat([starts.view(-1), ends.view(-1)])
+sorted_se, perm_se, counts_se = Fn._group_by_index(idx_se, T)
+dterm32 = torch.zeros(T, TS, dtype=torch.float32, device=dev)
+for _ in range(REPS):
+    ext().embed_scatter_sorted(sorted_se, perm_se, x, dterm32, M, KP, 0,
+                               TS + PS, 16)
+dterm = torch.empty(T, TS, dtype=torch.bfloat16, device=dev)
+for _ in range(REPS):
+    ext().cast_clear_rows(dterm32, counts_se, dterm)
+
+p1 = term.view(-1)
+g1 = term.clone().view(-1)
+master = p1.float()
+m_ = torch.zeros_like(master)
+v_ = torch.zeros_like(master)
+for _ in range(REPS):
+    ext().adam_step_bf16(p1, g1, master, m_, v_, 5, 0.01, 0.9, 0.999, 1e-8, 0.0)
+
+torch.cuda.synchronize()
+print("pmc probe done")
