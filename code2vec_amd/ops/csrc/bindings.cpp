@@ -44,6 +44,8 @@ void launch_lsm_nll_bwd(const void*, const long*, const float*, const float*,
                         hipStream_t);
 void launch_wgrad(const void*, const void*, float*, long, int, int, int,
                   hipStream_t);
+void launch_dgrad(const void*, const void*, void*, long, int, int,
+                  hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
@@ -226,6 +228,17 @@ void logsoftmax_nll_bwd(torch::Tensor logits, torch::Tensor label,
                      cur_stream());
 }
 
+void dgrad(torch::Tensor dz, torch::Tensor w2, torch::Tensor dx) {
+  CHK_CUDA(dz); CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
+  CHK_CONTIG(w2); CHK_DT(w2, torch::kBFloat16);
+  CHK_CONTIG(dx); CHK_DT(dx, torch::kBFloat16);
+  const long M = dz.size(0);
+  const int EP = dz.size(1), KP = w2.size(0);
+  TORCH_CHECK(w2.size(1) == EP && dx.size(1) == KP, "dgrad shapes");
+  launch_dgrad(dz.data_ptr(), w2.data_ptr(), dx.data_ptr(), M, KP, EP,
+               cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -280,6 +293,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
   m.def("wgrad", &wgrad);
   m.def("colsum_bf16", &colsum_bf16);
+  m.def("dgrad", &dgrad);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);
 }

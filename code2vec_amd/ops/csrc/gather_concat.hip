@@ -99,44 +99,52 @@ template <int NP>  // ceil(S/128) column-pair iterations per lane
 __global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
     const int* __restrict__ sorted_idx, const long* __restrict__ perm,
     const bf16* __restrict__ gout, float* __restrict__ dtable, long N, long M,
-    int KP, int S, int off0, int off1, int R) {
+    int KP, int S, int off0, int off1, int R_unused) {
+  // fixed chunk RT: all RT gradient loads are issued up front (independent
+  // rows), THEN runs are scanned from registers — no load in the dependent
+  // run-scan chain.
+  constexpr int RT = 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_block = threadIdx.x / WAVE;
   const long wid = (long)blockIdx.x * (blockDim.x / WAVE) + wave_in_block;
-  const long c0 = wid * R;
+  const long c0 = wid * RT;
   if (c0 >= N) return;
-  const long c1 = min(c0 + R, N);
+  const int count = (int)min((long)RT, N - c0);
 
-  long t = c0;
-  // skip the head run if it started in an earlier chunk... no: process it,
-  // flagged as boundary-crossing (atomic flush).  Each wave handles exactly
-  // its chunk's entries, so every entry is accumulated exactly once.
-  while (t < c1) {
-    const int idx = sorted_idx[t];
-    const bool head_outside = (t == c0) && (c0 > 0) && (sorted_idx[c0 - 1] == idx);
-    float acc[NP][2];
+  int idx[RT];
+  bf16x2 gv[RT][NP];
 #pragma unroll
-    for (int i = 0; i < NP; ++i) acc[i][0] = acc[i][1] = 0.f;
-    long u = t;
-    while (u < c1 && sorted_idx[u] == idx) {
-      const long o = perm[u];
+  for (int t = 0; t < RT; ++t) idx[t] = t < count ? sorted_idx[c0 + t] : -1;
+  const int prev_idx = c0 > 0 ? sorted_idx[c0 - 1] : -1;
+  const int next_idx = c0 + RT < N ? sorted_idx[c0 + RT] : -1;
+#pragma unroll
+  for (int t = 0; t < RT; ++t) {
+    if (t < count) {
+      const long o = perm[c0 + t];
       const bf16* grow =
           o < M ? gout + o * KP + off0 : gout + (o - M) * KP + off1;
 #pragma unroll
       for (int i = 0; i < NP; ++i) {
         const int col = i * 128 + lane * 2;
-        if (col < S) {
-          const bf16x2 g = *(const bf16x2*)(grow + col);
-          acc[i][0] += bf2f(g[0]);
-          acc[i][1] += bf2f(g[1]);
-        }
+        bf16x2 g = {bf16(0.f), bf16(0.f)};
+        if (col < S) g = *(const bf16x2*)(grow + col);
+        gv[t][i] = g;
       }
-      ++u;
     }
-    const bool tail_outside = (u == c1) && (c1 < N) && (sorted_idx[c1] == idx);
-    if (idx != 0) {  // index 0 is <PAD/>: pad contexts carry zero grads
-      float* drow = dtable + (long)idx * S;
-      if (head_outside || tail_outside) {
+  }
+
+  // statically unrolled run scan (no dynamic register indexing): a running
+  // accumulator is flushed whenever the index changes
+  float acc[NP][2];
+#pragma unroll
+  for (int i = 0; i < NP; ++i) acc[i][0] = acc[i][1] = 0.f;
+  bool run_from_start = true;
+  int cur_id = idx[0];
+
+  auto flush = [&](int id, bool boundary) {
+    if (id != 0) {  // index 0 is <PAD/>: pad contexts carry zero grads
+      float* drow = dtable + (long)id * S;
+      if (boundary) {
 #pragma unroll
         for (int i = 0; i < NP; ++i) {
           const int col = i * 128 + lane * 2;
@@ -156,8 +164,29 @@ __global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
         }
       }
     }
-    t = u;
+#pragma unroll
+    for (int i = 0; i < NP; ++i) acc[i][0] = acc[i][1] = 0.f;
+  };
+
+#pragma unroll
+  for (int t = 0; t < RT; ++t) {
+    if (t < count) {
+      if (t > 0 && idx[t] != cur_id) {
+        // flush the run that just ended (head-crossing if it began at the
+        // chunk start AND continues from the previous chunk)
+        flush(cur_id, run_from_start && prev_idx == cur_id);
+        run_from_start = false;
+        cur_id = idx[t];
+      }
+#pragma unroll
+      for (int i = 0; i < NP; ++i) {
+        acc[i][0] += bf2f(gv[t][i][0]);
+        acc[i][1] += bf2f(gv[t][i][1]);
+      }
+    }
   }
+  // final run: tail-crossing if the next chunk starts with the same index
+  flush(cur_id, (run_from_start && prev_idx == cur_id) || next_idx == cur_id);
 }
 
 // ---------------------------------------------------------------------------

@@ -162,8 +162,18 @@ class CombinerLNTanh(torch.autograd.Function):
         )
         dgamma = dgamma_p.sum(dim=0)
         dbeta = dbeta_p.sum(dim=0)
-        # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned); w is [EP, KP]
-        dx = dz @ w
+        # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned).  A hand-written
+        # direct-fragment kernel (ops/csrc/dgrad.hip) measured 192 us vs
+        # rocBLAS's 64 us at the top11 shape (8-wave full-KP duplicates B
+        # L2 traffic) and is kept only as a reference; re-enable via
+        # C2V_CUSTOM_DGRAD=1 for experiments.
+        if os.environ.get("C2V_CUSTOM_DGRAD") == "1":
+            w2 = w.t().contiguous()
+            dx = torch.empty(M, w.shape[1], dtype=torch.bfloat16,
+                             device=w.device)
+            ext().dgrad(dz, w2, dx)
+        else:
+            dx = dz @ w
         # wgrad: custom split-K MFMA kernel for the skinny big-K shape
         # (hipBLASLt is ~3.5x off there); partial slabs summed here.
         KP = x.shape[1]
