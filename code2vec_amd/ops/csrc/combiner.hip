@@ -69,21 +69,65 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     arow[mi] = r < M ? r : (M - 1);
   }
   const int kj = (lane >> 4) * 8;  // this lane's k sub-offset within a K-step
-  // B fragment row for tile n: Wt row n*16 + (lane&15)
-  const bf16* wrow = Wt + (size_t)(lane & 15) * KP + kj;
+
+  // B tile staged in LDS, double-buffered, T14 split (load regs early,
+  // ds_write after MFMA).  Image: [e][k8] with e-stride 40 elements (80 B)
+  // and k in the low position — both the 16-B ds_writes (consecutive
+  // (e,k8) chunks) and the per-lane fragment ds_reads spread banks with
+  // no conflicts.  Fragment (n, lane): e = n*16 + (lane&15), k8 = lane>>4.
+  constexpr int ESTRIDE = 40;  // elements per e column (32 data + 8 pad)
+  bf16* lds_b = (bf16*)(smem + 2 * EP * sizeof(float));
+  const size_t bufsz = (size_t)EP * ESTRIDE;
+  constexpr int CPT = (NT + 3) / 4;  // 16-B chunks per thread per K-step
+  uint4 stage_regs[CPT];
+  auto stage_load = [&](int kk) {
+    const int chunks = 4 * EP;  // (EP cols) x (32/8 k-groups)
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      const int c = threadIdx.x + i * 256;
+      if (c < chunks) {
+        const int e = c >> 2;
+        const int k8 = c & 3;
+        stage_regs[i] =
+            *(const uint4*)(Wt + (size_t)e * KP + kk * 32 + k8 * 8);
+      }
+    }
+  };
+  auto stage_write = [&](int buf) {
+    const int chunks = 4 * EP;
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      const int c = threadIdx.x + i * 256;
+      if (c < chunks) {
+        const int e = c >> 2;
+        const int k8 = c & 3;
+        *(uint4*)(lds_b + buf * bufsz + (size_t)e * ESTRIDE + k8 * 8) =
+            stage_regs[i];
+      }
+    }
+  };
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
 
   for (int kk = 0; kk < NK; ++kk) {
+    const int buf = kk & 1;
+    if (kk + 1 < NK) stage_load(kk + 1);
     bf16x8 a[2];
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
       a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kk * 32 + kj);
 #pragma unroll
     for (int n = 0; n < NT; ++n) {
-      const bf16x8 b =
-          *(const bf16x8*)(wrow + (size_t)n * 16 * KP + kk * 32);
+      const bf16x8 b = *(const bf16x8*)(
+          lds_b + buf * bufsz +
+          (size_t)(n * 16 + (lane & 15)) * ESTRIDE + (lane >> 4) * 8);
       acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b, acc[0][n], 0, 0, 0);
       acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b, acc[1][n], 0, 0, 0);
     }
+    if (kk + 1 < NK) stage_write(buf ^ 1);
+    __syncthreads();
   }
 
   // ---- epilogue: LayerNorm(E) + tanh + dropout ----
@@ -93,7 +137,7 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   const float invE = 1.0f / (float)E;
   bf16* lds_t = nullptr;
   if (EPI == 1) {
-    lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) +
+    lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) + 2 * (size_t)EP * 40 +
             (size_t)wave * 32 * (EP + 8);
   }
   float mean_r[2][4], rstd_r[2][4];
@@ -321,7 +365,7 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
   const long grid = (M + 127) / 128;
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
   int epi = epilogue_mode;
-  int smem = 2 * EP * sizeof(float);
+  int smem = 2 * EP * sizeof(float) + 2 * EP * 40 * (int)sizeof(bf16);
   if (epi == 1) {
     const int bounce = 4 * 32 * (EP + 8) * (int)sizeof(bf16);
     if (smem + bounce <= 160 * 1024 - 2048) smem += bounce;
