@@ -265,6 +265,26 @@ def reference_state_dict_torch(model: Code2VecTorch) -> Dict[str, torch.Tensor]:
     return sd
 
 
+def logical_from_reference_state_dict(sd, option) -> Dict[str, torch.Tensor]:
+    """Convert a reference-format state_dict (keys of reference
+    model/model.py; our checkpoints use the same format) into logical
+    init tensors, so checkpoints load into either backend."""
+    p = {
+        "terminal_embedding": sd["terminal_embedding.weight"].float(),
+        "path_embedding": sd["path_embedding.weight"].float(),
+        "input_weight": sd["input_linear.weight"].float(),
+        "ln_gamma": sd["input_layer_norm.weight"].float(),
+        "ln_beta": sd["input_layer_norm.bias"].float(),
+        "attention_a": sd["attention_parameter"].float(),
+    }
+    if "output_linear" in sd:  # angular-margin head (Parameter, no bias)
+        p["output_weight"] = sd["output_linear"].float()
+    else:
+        p["output_weight"] = sd["output_linear.weight"].float()
+        p["output_bias"] = sd["output_linear.bias"].float()
+    return p
+
+
 def build_model(option, backend: str = "auto", logical=None, device=None):
     """Factory: 'hip' on CUDA devices, 'torch' on CPU (or forced)."""
     if backend == "auto":

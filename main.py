@@ -76,6 +76,9 @@ def build_parser() -> argparse.ArgumentParser:
                    choices=["auto", "hip", "torch"])
     p.add_argument("--bucket_mb", type=int, default=32,
                    help="gradient all-reduce bucket size (MiB)")
+    p.add_argument("--init_model", type=str, default=None,
+                   help="warm-start from a code2vec.model checkpoint "
+                        "(reference state_dict format)")
     return p
 
 
@@ -117,8 +120,14 @@ def prepare(args, ctx):
 def train(args) -> None:
     ctx = make_context(args)
     reader, option, builder = prepare(args, ctx)
-    gen = torch.Generator().manual_seed(args.random_seed)
-    logical = init_logical_params(option, gen)
+    if args.init_model:
+        from code2vec_amd.models.code2vec import logical_from_reference_state_dict
+
+        sd = torch.load(args.init_model, map_location="cpu", weights_only=True)
+        logical = logical_from_reference_state_dict(sd, option)
+    else:
+        gen = torch.Generator().manual_seed(args.random_seed)
+        logical = init_logical_params(option, gen)
     backend = args.backend
     if backend == "auto":
         backend = "hip" if ctx.device.type == "cuda" else "torch"

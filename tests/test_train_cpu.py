@@ -126,3 +126,41 @@ def test_code_vec_visualizer_parse_roundtrip(tmp_path, tiny_corpus):
     assert len(labels) == mat.shape[0] == 48
     assert mat.shape[1] == 16
     assert np.isfinite(mat).all()
+
+
+def test_checkpoint_warm_start(tmp_path, tiny_corpus):
+    """--init_model loads a previously exported reference-format
+    checkpoint (resume capability beyond the reference's save-only)."""
+    out1 = run_cli(tmp_path, tiny_corpus)
+    ckpt = out1 / "code2vec.model"
+    out2 = tmp_path / "out2"
+    argv = [
+        "--corpus_path", tiny_corpus["corpus_path"],
+        "--path_idx_path", tiny_corpus["path_idx_path"],
+        "--terminal_idx_path", tiny_corpus["terminal_idx_path"],
+        "--model_path", str(out2),
+        "--vectors_path", str(out2 / "code.vec"),
+        "--max_epoch", "1", "--batch_size", "16",
+        "--terminal_embed_size", "12", "--path_embed_size", "12",
+        "--encode_size", "16", "--max_path_length", "12",
+        "--no_cuda", "--print_sample_cycle", "0",
+        "--init_model", str(ckpt),
+    ]
+    cli.main(argv)
+    assert (out2 / "code.vec").exists()
+    # warm start actually used the checkpoint weights
+    import torch
+
+    sd1 = torch.load(ckpt, weights_only=True)
+    from code2vec_amd.models.code2vec import (
+        Code2VecTorch, logical_from_reference_state_dict,
+    )
+    from code2vec_amd.utils.options import Option
+
+    opt = Option(terminal_count=sd1["terminal_embedding.weight"].shape[0],
+                 path_count=sd1["path_embedding.weight"].shape[0],
+                 label_count=sd1["output_linear.weight"].shape[0],
+                 terminal_embed_size=12, path_embed_size=12, encode_size=16)
+    m = Code2VecTorch(opt, logical_from_reference_state_dict(sd1, opt))
+    assert torch.equal(m.terminal_embedding.detach(),
+                       sd1["terminal_embedding.weight"])
