@@ -6,6 +6,7 @@
 
 #include "common.h"
 
+template <int NT>  // NT=1: nontemporal loads/stores for the streamed state
 __global__ void adam_bf16_kernel(bf16* __restrict__ p,
                                  const bf16* __restrict__ g,
                                  float* __restrict__ master,
@@ -19,10 +20,19 @@ __global__ void adam_bf16_kernel(bf16* __restrict__ p,
   typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
   for (long i = i0; i < n; i += stride) {
     if (i + 4 <= n) {
-      f32x4v mm = *(const f32x4v*)(m + i);
-      f32x4v vv = *(const f32x4v*)(v + i);
-      f32x4v pp = *(const f32x4v*)(master + i);
-      const bf16x4 gg = *(const bf16x4*)(g + i);
+      f32x4v mm, vv, pp;
+      bf16x4 gg;
+      if (NT) {
+        mm = __builtin_nontemporal_load((const f32x4v*)(m + i));
+        vv = __builtin_nontemporal_load((const f32x4v*)(v + i));
+        pp = __builtin_nontemporal_load((const f32x4v*)(master + i));
+        gg = __builtin_nontemporal_load((const bf16x4*)(g + i));
+      } else {
+        mm = *(const f32x4v*)(m + i);
+        vv = *(const f32x4v*)(v + i);
+        pp = *(const f32x4v*)(master + i);
+        gg = *(const bf16x4*)(g + i);
+      }
       bf16x4 pout;
 #pragma unroll
       for (int k = 0; k < 4; ++k) {
@@ -32,10 +42,17 @@ __global__ void adam_bf16_kernel(bf16* __restrict__ p,
         pp[k] -= lr * (mm[k] * inv_bc1) / (sqrtf(vv[k] * inv_bc2) + eps);
         pout[k] = f2bf(pp[k]);
       }
-      *(f32x4v*)(m + i) = mm;
-      *(f32x4v*)(v + i) = vv;
-      *(f32x4v*)(master + i) = pp;
-      *(bf16x4*)(p + i) = pout;
+      if (NT) {
+        __builtin_nontemporal_store(mm, (f32x4v*)(m + i));
+        __builtin_nontemporal_store(vv, (f32x4v*)(v + i));
+        __builtin_nontemporal_store(pp, (f32x4v*)(master + i));
+        __builtin_nontemporal_store(pout, (bf16x4*)(p + i));
+      } else {
+        *(f32x4v*)(m + i) = mm;
+        *(f32x4v*)(v + i) = vv;
+        *(f32x4v*)(master + i) = pp;
+        *(bf16x4*)(p + i) = pout;
+      }
     } else {
       for (long j = i; j < n; ++j) {
         float grad = bf2f(g[j]) + wd * master[j];
@@ -73,10 +90,16 @@ void launch_adam_bf16(void* p, const void* g, float* master, float* m,
   const float inv_bc2 = 1.0f / (1.0f - powf(b2, (float)step));
   const int block = 256;
   const long want = (n / 4 + block - 1) / block;
-  const int grid = (int)min(want > 0 ? want : 1, (long)4096);
-  adam_bf16_kernel<<<grid, block, 0, stream>>>((bf16*)p, (const bf16*)g,
-                                               master, m, v, n, lr, b1, b2,
-                                               eps, wd, inv_bc1, inv_bc2);
+  const int grid = (int)min(want > 0 ? want : 1, (long)8192);
+  const char* nt_env = getenv("C2V_ADAM_NT");
+  if (nt_env == nullptr || nt_env[0] != '0')
+    adam_bf16_kernel<1><<<grid, block, 0, stream>>>(
+        (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,
+        inv_bc1, inv_bc2);
+  else
+    adam_bf16_kernel<0><<<grid, block, 0, stream>>>(
+        (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,
+        inv_bc1, inv_bc2);
 }
 
 void launch_adam_f32(float* p, const float* g, float* m, float* v, long n,
