@@ -19,11 +19,11 @@ void launch_gather_concat_fwd(const int*, const int*, const int*, const void*,
 void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
                               float*, float*, long, int, int, hipStream_t);
 void launch_embed_scatter_sorted(const int*, const long*, const void*, float*,
-                                 long, long, int, int, int, int, int,
-                                 hipStream_t);
+                                 void*, unsigned char*, long, long, int, int,
+                                 int, int, int, hipStream_t);
 void launch_count_indices(const int*, int*, long, hipStream_t);
-void launch_cast_clear_rows(float*, const int*, void*, long, int,
-                            hipStream_t);
+void launch_cast_clear_rows(float*, const int*, unsigned char*, void*, long,
+                            int, hipStream_t);
 void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
@@ -91,30 +91,35 @@ void gather_concat_bwd(torch::Tensor starts, torch::Tensor paths,
 
 void embed_scatter_sorted(torch::Tensor sorted_idx, torch::Tensor perm,
                           torch::Tensor gout, torch::Tensor dtable,
+                          torch::Tensor out_bf16, torch::Tensor flags,
                           int64_t M, int64_t KP, int64_t off0, int64_t off1,
                           int64_t R) {
   CHK_CUDA(sorted_idx); CHK_CONTIG(sorted_idx);
   CHK_DT(sorted_idx, torch::kInt32); CHK_DT(perm, torch::kInt64);
   CHK_DT(gout, torch::kBFloat16); CHK_CONTIG(gout);
   CHK_DT(dtable, torch::kFloat32); CHK_CONTIG(dtable);
+  CHK_DT(out_bf16, torch::kBFloat16); CHK_CONTIG(out_bf16);
+  CHK_DT(flags, torch::kUInt8);
   const long N = sorted_idx.numel();
   const int S = dtable.size(1);
   launch_embed_scatter_sorted(sorted_idx.data_ptr<int>(),
                               perm.data_ptr<long>(), gout.data_ptr(),
-                              dtable.data_ptr<float>(), N, M, (int)KP, S,
-                              (int)off0, (int)off1, (int)R, cur_stream());
+                              dtable.data_ptr<float>(), out_bf16.data_ptr(),
+                              flags.data_ptr<unsigned char>(), N, M, (int)KP,
+                              S, (int)off0, (int)off1, (int)R, cur_stream());
 }
 
 void cast_clear_rows(torch::Tensor dtable, torch::Tensor counts,
-                     torch::Tensor out) {
+                     torch::Tensor flags, torch::Tensor out) {
   CHK_CUDA(dtable); CHK_CONTIG(dtable); CHK_DT(dtable, torch::kFloat32);
   CHK_DT(counts, torch::kInt32); CHK_DT(out, torch::kBFloat16);
-  CHK_CONTIG(out);
+  CHK_DT(flags, torch::kUInt8); CHK_CONTIG(out);
   const long T = dtable.size(0);
   const int S = dtable.size(1);
-  TORCH_CHECK(counts.numel() >= T, "counts too small");
+  TORCH_CHECK(counts.numel() >= T && flags.numel() >= T, "counts/flags");
   launch_cast_clear_rows(dtable.data_ptr<float>(), counts.data_ptr<int>(),
-                         out.data_ptr(), T, S, cur_stream());
+                         flags.data_ptr<unsigned char>(), out.data_ptr(), T,
+                         S, cur_stream());
 }
 
 void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,

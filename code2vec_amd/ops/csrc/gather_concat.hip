@@ -98,8 +98,9 @@ __global__ void gather_concat_bwd_kernel(
 template <int NP>  // ceil(S/128) column-pair iterations per lane
 __global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
     const int* __restrict__ sorted_idx, const long* __restrict__ perm,
-    const bf16* __restrict__ gout, float* __restrict__ dtable, long N, long M,
-    int KP, int S, int off0, int off1, int R_unused) {
+    const bf16* __restrict__ gout, float* __restrict__ dtable,
+    bf16* __restrict__ out_bf16, unsigned char* __restrict__ flags, long N,
+    long M, int KP, int S, int off0, int off1, int R_unused) {
   // fixed chunk RT: all RT gradient loads are issued up front (independent
   // rows), THEN runs are scanned from registers — no load in the dependent
   // run-scan chain.
@@ -141,10 +142,14 @@ __global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
   bool run_from_start = true;
   int cur_id = idx[0];
 
+  // interior runs (the single run of an index fully inside this chunk)
+  // write the bf16 grad row DIRECTLY; boundary-crossing runs accumulate
+  // fp32 partials in the persistent scratch and set the row's flag for the
+  // combine pass (cast_clear_rows).
   auto flush = [&](int id, bool boundary) {
     if (id != 0) {  // index 0 is <PAD/>: pad contexts carry zero grads
-      float* drow = dtable + (long)id * S;
       if (boundary) {
+        float* drow = dtable + (long)id * S;
 #pragma unroll
         for (int i = 0; i < NP; ++i) {
           const int col = i * 128 + lane * 2;
@@ -153,13 +158,15 @@ __global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
             atomic_add_f32(drow + col + 1, acc[i][1]);
           }
         }
+        if (lane == 0) flags[id] = 1;
       } else {
+        bf16* brow = out_bf16 + (long)id * S;
 #pragma unroll
         for (int i = 0; i < NP; ++i) {
           const int col = i * 128 + lane * 2;
           if (col < S) {
-            float2 v = {acc[i][0], acc[i][1]};
-            *(float2*)(drow + col) = v;
+            bf16x2 v = {f2bf(acc[i][0]), f2bf(acc[i][1])};
+            *(bf16x2*)(brow + col) = v;
           }
         }
       }
@@ -226,6 +233,7 @@ __global__ void scatter_group_kernel(const int* __restrict__ idx,
 // written: emit bf16 zeros without touching the fp32 buffer.
 __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
                                        const int* __restrict__ counts,
+                                       unsigned char* __restrict__ flags,
                                        bf16* __restrict__ out, long T, int S) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -233,12 +241,16 @@ __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
   long row = (long)blockIdx.x * wpb + wave;
   const long stride = (long)gridDim.x * wpb;
   for (; row < T; row += stride) {
-    const bool touched = counts[row] > 0;
+    const bool boundary = flags[row] != 0 && row != 0;
+    // interior-touched rows were written bf16-direct by the scatter kernel;
+    // row 0 (<PAD/>) always reads as untouched (its grads are exactly zero)
+    const bool interior = !boundary && row != 0 && counts[row] > 0;
+    if (interior) continue;
     float* frow = dtable + row * S;
     bf16* orow = out + row * S;
     for (int c0 = lane * 2; c0 < S; c0 += WAVE * 2) {
       bf16x2 o = {bf16(0.f), bf16(0.f)};
-      if (touched) {
+      if (boundary) {
         const float2 v = *(const float2*)(frow + c0);
         o[0] = f2bf(v.x);
         o[1] = f2bf(v.y);
@@ -247,17 +259,19 @@ __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
       }
       *(bf16x2*)(orow + c0) = o;
     }
+    if (boundary && lane == 0) flags[row] = 0;
   }
 }
 
 extern "C" {
 
-void launch_cast_clear_rows(float* dtable, const int* counts, void* out,
-                            long T, int S, hipStream_t stream) {
+void launch_cast_clear_rows(float* dtable, const int* counts,
+                            unsigned char* flags, void* out, long T, int S,
+                            hipStream_t stream) {
   const int block = 256;
   const int wpb = block / WAVE;
   const int grid = (int)min((T + wpb - 1) / wpb, (long)8192);
-  cast_clear_rows_kernel<<<grid, block, 0, stream>>>(dtable, counts,
+  cast_clear_rows_kernel<<<grid, block, 0, stream>>>(dtable, counts, flags,
                                                      (bf16*)out, T, S);
 }
 
@@ -277,7 +291,8 @@ void launch_scatter_group(const int* idx, int* cursor, int* sorted_idx,
 }
 
 void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
-                                 const void* gout, float* dtable, long N,
+                                 const void* gout, float* dtable,
+                                 void* out_bf16, unsigned char* flags, long N,
                                  long M, int KP, int S, int off0, int off1,
                                  int R, hipStream_t stream) {
   if (R <= 0) R = 16;  // entries per wave-chunk
@@ -288,8 +303,8 @@ void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
 #define SCASE(n)                                                              \
   case n:                                                                     \
     embed_scatter_sorted_kernel<n><<<grid, 256, 0, stream>>>(                 \
-        sorted_idx, perm, (const bf16*)gout, dtable, N, M, KP, S, off0,       \
-        off1, R);                                                             \
+        sorted_idx, perm, (const bf16*)gout, dtable, (bf16*)out_bf16, flags,  \
+        N, M, KP, S, off0, off1, R);                                          \
     break;
   switch (np) {
     SCASE(1) SCASE(2) SCASE(3) SCASE(4)

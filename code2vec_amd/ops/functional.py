@@ -74,20 +74,23 @@ class GatherConcat(torch.autograd.Function):
         # no full-buffer memset per step
         dterm32 = _scratch_f32(term_shape, dev)
         dpath32 = _scratch_f32(path_shape, dev)
-        # sort-based segmented scatter (K13 v2): counting-sort groups the
-        # index lists by value, then run-owner waves write each touched row
-        # once (fp32 atomics only at chunk boundaries of heavy-hitter runs).
-        idx_se = torch.cat([starts.view(-1), ends.view(-1)])
-        sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
-        ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32,
-                                   M, KP, 0, TS + PS, _SCATTER_R)
-        sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
-        ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32,
-                                   M, KP, TS, TS, _SCATTER_R)
+        flags_t = _scratch_flags(term_shape[0], dev)
+        flags_p = _scratch_flags(path_shape[0], dev)
+        # sort-based segmented scatter (K13 v3): counting-sort groups the
+        # index lists; run-owner waves write interior rows' bf16 grads
+        # directly; boundary-crossing runs (heavy hitters) combine via
+        # flagged fp32 scratch in cast_clear_rows.
         dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
         dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
-        ext().cast_clear_rows(dterm32, counts_se, dterm)
-        ext().cast_clear_rows(dpath32, counts_p, dpath)
+        idx_se = torch.cat([starts.view(-1), ends.view(-1)])
+        sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
+        ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32, dterm,
+                                   flags_t, M, KP, 0, TS + PS, _SCATTER_R)
+        sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
+        ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
+                                   flags_p, M, KP, TS, TS, _SCATTER_R)
+        ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
+        ext().cast_clear_rows(dpath32, counts_p, flags_p, dpath)
         return None, None, None, dterm, dpath
 
 
@@ -101,6 +104,15 @@ def _scratch_f32(shape, device) -> torch.Tensor:
     buf = _scratch_cache.get(key)
     if buf is None:
         buf = torch.zeros(shape, dtype=torch.float32, device=device)
+        _scratch_cache[key] = buf
+    return buf
+
+
+def _scratch_flags(rows: int, device) -> torch.Tensor:
+    key = ("flags", rows, str(device))
+    buf = _scratch_cache.get(key)
+    if buf is None:
+        buf = torch.zeros(rows, dtype=torch.uint8, device=device)
         _scratch_cache[key] = buf
     return buf
 
