@@ -79,56 +79,49 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   bf16* lds_b = (bf16*)(smem + 2 * EP * sizeof(float));
   const size_t bufsz = (size_t)EP * ESTRIDE;
   constexpr int CPT = (NT + 3) / 4;  // 16-B chunks per thread per K-step
-  uint4 stage_regs[CPT];
-  auto stage_load = [&](int kk) {
-    const int chunks = 4 * EP;  // (EP cols) x (32/8 k-groups)
-#pragma unroll
-    for (int i = 0; i < CPT; ++i) {
-      const int c = threadIdx.x + i * 256;
-      if (c < chunks) {
-        const int e = c >> 2;
-        const int k8 = c & 3;
-        stage_regs[i] =
-            *(const uint4*)(Wt + (size_t)e * KP + kk * 32 + k8 * 8);
-      }
-    }
-  };
-  auto stage_write = [&](int buf) {
-    const int chunks = 4 * EP;
-#pragma unroll
-    for (int i = 0; i < CPT; ++i) {
-      const int c = threadIdx.x + i * 256;
-      if (c < chunks) {
-        const int e = c >> 2;
-        const int k8 = c & 3;
-        *(uint4*)(lds_b + buf * bufsz + (size_t)e * ESTRIDE + k8 * 8) =
-            stage_regs[i];
-      }
-    }
-  };
+  const int schunks = 4 * EP;  // (EP cols) x (32/8 k-groups)
+  const int s_e = threadIdx.x >> 2;
+  const int s_k8 = threadIdx.x & 3;
 
-  stage_load(0);
-  stage_write(0);
+  // direct global->LDS staging of K-tile kk into buffer buf (short-lived
+  // register round trip only; an address-taken staging array gets demoted
+  // to scratch INSIDE the loop by hipcc)
+#define STAGE(kk, buf)                                                        \
+  _Pragma("unroll") for (int i_ = 0; i_ < CPT; ++i_) {                        \
+    const int c_ = (int)threadIdx.x + i_ * 256;                               \
+    if (c_ < schunks)                                                         \
+      *(uint4*)(lds_b + (buf) * bufsz + (size_t)(s_e + i_ * 64) * ESTRIDE +   \
+                s_k8 * 8) =                                                   \
+          *(const uint4*)(Wt + (size_t)(s_e + i_ * 64) * KP + (kk) * 32 +     \
+                          s_k8 * 8);                                          \
+  }
+
+  STAGE(0, 0)
   __syncthreads();
 
   for (int kk = 0; kk < NK; ++kk) {
     const int buf = kk & 1;
-    if (kk + 1 < NK) stage_load(kk + 1);
+    if (kk + 1 < NK) STAGE(kk + 1, buf ^ 1)
     bf16x8 a[2];
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
       a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kk * 32 + kj);
+    // explicit 2-deep B pipeline: read fragment n+1 while n's MFMAs issue
+    const bf16* bbase =
+        lds_b + buf * bufsz + (size_t)(lane & 15) * ESTRIDE + (lane >> 4) * 8;
+    bf16x8 b_cur = *(const bf16x8*)(bbase);
+    bf16x8 b_nxt;
 #pragma unroll
     for (int n = 0; n < NT; ++n) {
-      const bf16x8 b = *(const bf16x8*)(
-          lds_b + buf * bufsz +
-          (size_t)(n * 16 + (lane & 15)) * ESTRIDE + (lane >> 4) * 8);
-      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b, acc[0][n], 0, 0, 0);
-      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b, acc[1][n], 0, 0, 0);
+      if (n + 1 < NT)
+        b_nxt = *(const bf16x8*)(bbase + (size_t)(n + 1) * 16 * ESTRIDE);
+      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b_cur, acc[0][n], 0, 0, 0);
+      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b_cur, acc[1][n], 0, 0, 0);
+      b_cur = b_nxt;
     }
-    if (kk + 1 < NK) stage_write(buf ^ 1);
     __syncthreads();
   }
+#undef STAGE
 
   // ---- epilogue: LayerNorm(E) + tanh + dropout ----
   // EPI==1: values staged per-wave in LDS (padded row stride) and flushed
