@@ -63,7 +63,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--config", type=str, default="top11", choices=CONFIGS)
-    ap.add_argument("--pool", type=int, default=4, help="synthetic batch pool size")
+    ap.add_argument("--pool", type=int, default=8, help="synthetic batch pool size")
     ap.add_argument("--mode", type=str, default="train", choices=["train", "infer"],
                     help="train: full step (the driver contract); infer: "
                          "hipGraph-captured batched code-vector export "
@@ -175,7 +175,8 @@ def main() -> None:
                 "label_vocab": cfg["label_count"],
                 "embed": cfg["embed"],
                 "encode": cfg["encode"],
-                "note": "step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam; top11 label vocab size unpublished, 30k assumed",
+                "note": ("step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam"
+                        + ("; top11 label vocab size unpublished, 30k assumed" if args.config == "top11" else "")),
             },
         }))
 
