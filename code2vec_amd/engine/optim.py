@@ -59,6 +59,8 @@ class FusedAdam:
             )
 
     def zero_grad(self, set_to_none: bool = False) -> None:
+        # NOTE: when grads are BucketedAllReduce views, use ddp.zero_grad()
+        # instead — set_to_none=True here would detach the bucket views.
         for p in self.params:
             if p.grad is not None:
                 if set_to_none:

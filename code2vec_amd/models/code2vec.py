@@ -209,10 +209,8 @@ class Code2VecHIP(nn.Module):
         return outputs, cv[:, : self.E], attn
 
     def loss(self, outputs, label, class_weight):
-        if self.option.angular_margin_loss:
-            # optional path: [B,L] bf16 elementwise head -> fused NLL still fine
-            return Fn.FusedLogSoftmaxNLL.apply(outputs, label, class_weight)
-        return Fn.FusedLogSoftmaxNLL.apply(outputs, label, class_weight)
+        return Fn.FusedLogSoftmaxNLL.apply(
+            outputs.contiguous(), label, class_weight)
 
     # ------------------------------------------------------------------
     def reference_state_dict(self) -> Dict[str, torch.Tensor]:

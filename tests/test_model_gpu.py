@@ -214,3 +214,65 @@ def test_angular_margin_gpu(dev):
     loss_h = hip.loss(out_h, y, w)
     loss_h.backward()
     assert torch.isfinite(hip.output_weight.grad.float()).all()
+
+
+@pytest.mark.parametrize("dt,dp,E", [(100, 100, 300), (72, 40, 160), (32, 32, 32)])
+def test_nondefault_dims_match_oracle(dev, dt, dp, E):
+    """Generality across padded shapes: default CLI encode_size is 300
+    (EP=320, NT=20); small/odd embed sizes exercise segment padding."""
+    opt = make_option(terminal_embed_size=dt, path_embed_size=dp,
+                      encode_size=E, max_path_length=24,
+                      terminal_count=800, path_count=700, label_count=150)
+    g = torch.Generator().manual_seed(12)
+    logical = init_logical_params(opt, g)
+    hip = Code2VecHIP(opt, logical, device=dev).train()
+    ref = Code2VecTorch(opt, logical).to(dev).train()
+    s, p, e, y = make_inputs(opt, 12, dev, seed=8)
+    w = torch.ones(opt.label_count, device=dev)
+    out_h, cv_h, attn_h = hip(s, p, e, y)
+    out_r, cv_r, attn_r = ref(s.long(), p.long(), e.long(), y)
+    assert relerr(attn_h, attn_r) < 4e-2
+    assert relerr(out_h, out_r) < 6e-2
+    loss_h = hip.loss(out_h, y, w)
+    loss_r = ref.loss(out_r, y, w)
+    assert abs(float(loss_h) - float(loss_r)) / float(loss_r) < 3e-2
+    loss_h.backward()
+    loss_r.backward()
+    E_ = opt.encode_size
+    assert relerr(hip.ln_gamma.grad[:E_], ref.ln_gamma.grad) < 8e-2
+    assert relerr(hip.terminal_embedding.grad[:, :dt].float(),
+                  ref.terminal_embedding.grad) < 8e-2
+
+
+def test_attention_grad_through_attn_output(dev):
+    """Backward when the returned attention itself receives a gradient
+    (the has_dattn path of attention_bwd)."""
+    from code2vec_amd.ops.functional import AttentionPool
+    from code2vec_amd.ops import reference as R
+    from code2vec_amd.ops import round_up
+
+    B, C, E = 6, 30, 100
+    EP = round_up(E)
+    g = torch.Generator().manual_seed(2)
+    ccv = torch.zeros(B, C, EP)
+    ccv[:, :, :E] = torch.tanh(torch.randn(B, C, E, generator=g))
+    a = torch.zeros(EP)
+    a[:E] = torch.randn(E, generator=g) * 0.2
+    starts = torch.randint(1, 50, (B, C), generator=g, dtype=torch.int32)
+    starts[:, -3:] = 0
+    ccv_h = ccv.to(dev, torch.bfloat16).requires_grad_(True)
+    a_h = a.to(dev).requires_grad_(True)
+    starts_d = starts.to(dev)
+    cv, attn = AttentionPool.apply(ccv_h, a_h, starts_d, E)
+    dcv = torch.randn_like(cv) * 0.1
+    dcv[:, E:] = 0
+    dattn = torch.randn_like(attn) * 0.1
+    torch.autograd.backward([cv, attn], [dcv, dattn])
+
+    ccv_r = ccv.float().to(dev).requires_grad_(True)
+    a_r = a.to(dev).requires_grad_(True)
+    mask = (starts_d > 0).float()
+    cv_r, attn_r = R.attention_code_vector(ccv_r, a_r, mask)
+    torch.autograd.backward([cv_r, attn_r], [dcv, dattn])
+    assert relerr(ccv_h.grad.float(), ccv_r.grad) < 6e-2
+    assert relerr(a_h.grad[:E], a_r.grad[:E]) < 6e-2

@@ -164,3 +164,9 @@ def test_checkpoint_warm_start(tmp_path, tiny_corpus):
     m = Code2VecTorch(opt, logical_from_reference_state_dict(sd1, opt))
     assert torch.equal(m.terminal_embedding.detach(),
                        sd1["terminal_embedding.weight"])
+
+
+@pytest.mark.parametrize("method", ["exact", "ave_subtoken"])
+def test_eval_methods_end_to_end(tmp_path, tiny_corpus, method):
+    out_dir = run_cli(tmp_path, tiny_corpus, extra=["--eval_method", method])
+    assert (out_dir / "code.vec").exists()
