@@ -23,7 +23,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import reference as R
-from ..ops import round_up
+from ..ops import round_up, seg_round
 from ..ops import functional as Fn
 
 
@@ -135,10 +135,12 @@ class Code2VecHIP(nn.Module):
             option.path_embed_size,
             option.encode_size,
         )
-        self.TS = round_up(dt)
-        self.PS = round_up(dp)
+        self.TS = seg_round(dt)
+        self.PS = seg_round(dp)
         self.EP = round_up(E)
-        self.KP = 2 * self.TS + self.PS
+        # combiner K: segments at offsets [0, TS, TS+PS], summed width
+        # padded to the MFMA K granularity (gather zero-fills the tail)
+        self.KP = round_up(2 * self.TS + self.PS, 32)
         self.E = E
 
         def pad2(t, rows, cols):

@@ -22,11 +22,20 @@ import os
 
 import torch
 
-PAD = 32  # element granularity of padded dims (64 B in bf16)
+PAD = 32  # element granularity of the encode dim (MFMA N tiles)
+SEG_PAD = 8  # element granularity of embedding rows / concat segments (16 B)
 
 
 def round_up(x: int, m: int = PAD) -> int:
     return (x + m - 1) // m * m
+
+
+def seg_round(x: int) -> int:
+    """Embedding-table row stride / concat-segment width: 16-B granules
+    (vectorized gathers) without the 32-element MFMA padding waste —
+    dt=100 stores as 104, not 128.  The combiner K dim pads the SUM of
+    segments up to 32 (KP = round_up(2*TS+PS, 32))."""
+    return (x + SEG_PAD - 1) // SEG_PAD * SEG_PAD
 
 
 _ext = None

@@ -15,7 +15,8 @@
 
 extern "C" {
 void launch_gather_concat_fwd(const int*, const int*, const int*, const void*,
-                              const void*, void*, long, int, int, hipStream_t);
+                              const void*, void*, long, int, int, int,
+                              hipStream_t);
 void launch_gather_concat_bwd(const int*, const int*, const int*, const void*,
                               float*, float*, long, int, int, hipStream_t);
 void launch_embed_scatter_sorted(const int*, const long*, const void*, float*,
@@ -67,12 +68,13 @@ void gather_concat_fwd(torch::Tensor starts, torch::Tensor paths,
   CHK_DT(out, torch::kBFloat16); CHK_CONTIG(out);
   const long M = starts.numel();
   const int TS = term.size(1), PS = path.size(1);
-  TORCH_CHECK(out.size(0) == M && out.size(1) == 2 * TS + PS, "out shape");
-  TORCH_CHECK(TS % 32 == 0 && PS % 32 == 0, "padded strides must be 32-mult");
+  TORCH_CHECK(out.size(0) == M && out.size(1) >= 2 * TS + PS, "out shape");
+  TORCH_CHECK(TS % 8 == 0 && PS % 8 == 0, "segment strides must be 8-mult");
+  TORCH_CHECK(out.size(1) % 8 == 0, "out width must be 8-mult");
   launch_gather_concat_fwd(starts.data_ptr<int>(), paths.data_ptr<int>(),
                            ends.data_ptr<int>(), term.data_ptr(),
                            path.data_ptr(), out.data_ptr(), M, TS, PS,
-                           cur_stream());
+                           (int)out.size(1), cur_stream());
 }
 
 void gather_concat_bwd(torch::Tensor starts, torch::Tensor paths,

@@ -15,8 +15,8 @@ __global__ void gather_concat_fwd_kernel(
     const int* __restrict__ starts, const int* __restrict__ paths,
     const int* __restrict__ ends, const bf16* __restrict__ term,
     const bf16* __restrict__ path, bf16* __restrict__ out,
-    long M, int TS, int PS) {
-  const int KP = 2 * TS + PS;
+    long M, int TS, int PS, int KP) {
+  // KP >= 2*TS+PS (combiner K padding); tail chunks are zero-filled
   const int chunks = KP / 8;  // 16-B chunks per row
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -34,10 +34,10 @@ __global__ void gather_concat_fwd_kernel(
     uint4* orow = (uint4*)(out + row * KP);
     const int ts8 = TS / 8, ps8 = PS / 8;
     for (int c = lane; c < chunks; c += WAVE) {
-      uint4 v;
+      uint4 v = {0, 0, 0, 0};
       if (c < ts8) v = srow[c];
       else if (c < ts8 + ps8) v = prow[c - ts8];
-      else v = erow[c - ts8 - ps8];
+      else if (c < 2 * ts8 + ps8) v = erow[c - ts8 - ps8];
       orow[c] = v;
     }
   }
@@ -317,13 +317,13 @@ void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
 void launch_gather_concat_fwd(const int* starts, const int* paths,
                               const int* ends, const void* term,
                               const void* path, void* out, long M, int TS,
-                              int PS, hipStream_t stream) {
+                              int PS, int KP, hipStream_t stream) {
   const int block = 256;
   const int waves_per_block = block / WAVE;
   int grid = (int)min((M + waves_per_block - 1) / waves_per_block, (long)16384);
   gather_concat_fwd_kernel<<<grid, block, 0, stream>>>(
       starts, paths, ends, (const bf16*)term, (const bf16*)path, (bf16*)out,
-      M, TS, PS);
+      M, TS, PS, KP);
 }
 
 void launch_gather_concat_bwd(const int* starts, const int* paths,

@@ -21,7 +21,7 @@ import torch
 _FWD_EPI = int(os.environ.get("C2V_FWD_EPI", "1"))
 _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 
-from . import ext
+from . import ext, round_up
 
 _rng_state = {"seed": None, "offset": 0}
 
@@ -53,7 +53,8 @@ class GatherConcat(torch.autograd.Function):
         TS = term_w.shape[1]
         PS = path_w.shape[1]
         out = torch.empty(
-            B * C, 2 * TS + PS, dtype=torch.bfloat16, device=starts.device
+            B * C, round_up(2 * TS + PS, 32), dtype=torch.bfloat16,
+            device=starts.device
         )
         ext().gather_concat_fwd(starts, paths, ends, term_w, path_w, out)
         ctx.save_for_backward(starts, paths, ends)
@@ -67,8 +68,8 @@ class GatherConcat(torch.autograd.Function):
         dev = grad_out.device
         M = starts.numel()
         TS, PS = term_shape[1], path_shape[1]
-        KP = 2 * TS + PS
         gout = grad_out.contiguous()
+        KP = gout.shape[1]
         # persistent fp32 scatter scratch — zeroed once; the fused
         # cast-and-clear kernel re-zeroes exactly the rows it consumed, so
         # no full-buffer memset per step
