@@ -68,6 +68,10 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     long r = row0 + wave * 32 + mi * 16 + (lane & 15);
     arow[mi] = r < M ? r : (M - 1);
   }
+#ifdef C2V_EXP_FIXED_A  // experiment: L2-hot A reads (pattern diagnosis)
+  arow[0] = (lane & 15);
+  arow[1] = 16 + (lane & 15);
+#endif
   const int kj = (lane >> 4) * 8;  // this lane's k sub-offset within a K-step
 
   // B tile staged in LDS, double-buffered, T14 split (load regs early,
@@ -185,7 +189,7 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
             } else {
               const float xhat = (zv - mean) * rstd;
               const float u = xhat * lds_gamma[col] + lds_beta[col];
-              float y = tanhf(u);
+              float y = fast_tanh(u);
               if (p > 0.0f) {
                 const float u01 = rng_uniform(
                     seed, offset + (unsigned long long)row * EP + col);

@@ -17,17 +17,27 @@ __device__ __forceinline__ float bf2f(bf16 x) { return (float)x; }
 __device__ __forceinline__ bf16 f2bf(float x) { return (bf16)x; }
 
 // ---------------------------------------------------------------------------
-// Counter-based RNG for dropout (recomputable in backward from (seed, offset)).
-// splitmix64 finalizer on the element's global index — statistically ample
-// for Bernoulli masks, deterministic, stateless.
+// Counter-based RNG for dropout masks: deterministic, stateless, cheap
+// (32-bit finalizer; the backward never replays it — the mask is recovered
+// from the saved forward output, so only mask-distribution quality matters).
 __device__ __forceinline__ float rng_uniform(unsigned long long seed,
                                              unsigned long long idx) {
-  unsigned long long z = seed + 0x9E3779B97F4A7C15ull * (idx + 1ull);
-  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
-  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-  z = z ^ (z >> 31);
-  // take the top 24 bits -> [0, 1)
-  return (float)(z >> 40) * (1.0f / 16777216.0f);
+  unsigned int h = (unsigned int)(idx ^ (idx >> 32)) * 0x9E3779B9u;
+  h ^= (unsigned int)seed ^ (unsigned int)(seed >> 32) * 0x85EBCA6Bu;
+  h ^= h >> 16;
+  h *= 0x7FEB352Du;
+  h ^= h >> 15;
+  h *= 0x846CA68Bu;
+  h ^= h >> 16;
+  return (float)(h >> 8) * (1.0f / 16777216.0f);
+}
+
+// tanh via one v_exp (exact identity; skips libm's precise-path branches):
+// tanh(x) = sign(x) * (1 - 2/(e^{2|x|} + 1))
+__device__ __forceinline__ float fast_tanh(float x) {
+  const float ax = fabsf(x);
+  const float t = 1.0f - 2.0f / (__expf(2.0f * ax) + 1.0f);
+  return copysignf(t, x);
 }
 
 // ---------------------------------------------------------------------------
