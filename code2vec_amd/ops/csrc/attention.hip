@@ -5,53 +5,19 @@
 //   attn = softmax(scores, dim=C)
 //   cv[b,e] = sum_c attn[b,c] * ccv[b,c,e]
 //
-// One workgroup per batch row; the [C, EP] ccv slice is staged in LDS once
-// and reused by the score and pool phases (C=200, EP=128 -> 50 KiB).
+// One workgroup (256 threads = 4 waves) per batch row; the [C, EP] ccv
+// slice is staged in LDS once and reused by every phase.  Work is
+// row-oriented: each 16-lane group owns one context at a time (8 columns
+// per lane -> bf16x8 vector accesses, coalesced global writes in backward).
 
 #include "common.h"
 
 #define NINF_F (-3.4e38f)
 
-// ---------------------------------------------------------------------------
-template <bool STAGE_LDS>
-__global__ __launch_bounds__(256) void attention_fwd_kernel(
-    const bf16* __restrict__ ccv, const float* __restrict__ a,
-    const int* __restrict__ starts, float* __restrict__ cv,
-    float* __restrict__ attn, int B, int C, int EP, int E) {
-  const int b = blockIdx.x;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [C*EP bf16 tile (optional)] [EP f32 a] [C f32 scores] [64 f32 red]
-  bf16* tile = (bf16*)smem;
-  float* lds_a = (float*)(smem + (STAGE_LDS ? (size_t)C * EP * 2 : 0));
-  float* scores = lds_a + EP;
-  float* red = scores + C;
-
-  const bf16* src = ccv + (long)b * C * EP;
-  for (int e = threadIdx.x; e < EP; e += blockDim.x) lds_a[e] = a[e];
-  if (STAGE_LDS) {
-    const int total = C * EP / 8;
-    const uint4* s4 = (const uint4*)src;
-    uint4* d4 = (uint4*)tile;
-    for (int i = threadIdx.x; i < total; i += blockDim.x) d4[i] = s4[i];
-  }
-  __syncthreads();
-
-  // phase 1: scores (thread-per-context dot over EP; pad cols of a are 0)
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
-    float s = 0.f;
-    for (int e = 0; e < EP; e += 8) {
-      bf16 v[8];
-      *(uint4*)v = *(const uint4*)(row + e);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) s += bf2f(v[j]) * lds_a[e + j];
-    }
-    const float mask = starts[(long)b * C + c] > 0 ? 1.0f : 0.0f;
-    scores[c] = s * mask + (1.0f - mask) * NINF_F;
-  }
-  __syncthreads();
-
-  // phase 2: block softmax over C
+// block softmax over scores[C] (in LDS), returns nothing (scores -> attn)
+__device__ __forceinline__ void block_softmax(float* scores, float* red,
+                                              float* attn_out, int C,
+                                              long base) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int nwaves = blockDim.x / WAVE;
@@ -85,23 +51,99 @@ __global__ __launch_bounds__(256) void attention_fwd_kernel(
   const float inv_sum = 1.0f / red[8];
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
     scores[c] *= inv_sum;
-    attn[(long)b * C + c] = scores[c];
+    attn_out[base + c] = scores[c];
+  }
+  __syncthreads();
+}
+
+// per-16-lane-group dot of a ccv row against a vector held in LDS
+// (each lane covers cols (l&15)*8 + j + 128*i)
+__device__ __forceinline__ float group_row_dot(const bf16* row,
+                                               const float* vec, int EP,
+                                               int lane16) {
+  float s = 0.f;
+  for (int e0 = lane16 * 8; e0 < EP; e0 += 128) {
+    bf16 v[8];
+    *(uint4*)v = *(const uint4*)(row + e0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += bf2f(v[j]) * vec[e0 + j];
+  }
+  return s;
+}
+
+// ---------------------------------------------------------------------------
+template <bool STAGE_LDS>
+__global__ __launch_bounds__(256) void attention_fwd_kernel(
+    const bf16* __restrict__ ccv, const float* __restrict__ a,
+    const int* __restrict__ starts, float* __restrict__ cv,
+    float* __restrict__ attn, int B, int C, int EP, int E) {
+  const int b = blockIdx.x;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: [ccv tile (opt)] [a f32 EP] [scores C] [red 64] [partials 4*EP]
+  bf16* tile = (bf16*)smem;
+  float* lds_a = (float*)(smem + (STAGE_LDS ? (size_t)C * EP * 2 : 0));
+  float* scores = lds_a + EP;
+  float* red = scores + C;
+  float* partials = red + 64;
+
+  const bf16* src = ccv + (long)b * C * EP;
+  for (int e = threadIdx.x; e < EP; e += blockDim.x) lds_a[e] = a[e];
+  if (STAGE_LDS) {
+    const int total = C * EP / 8;
+    const uint4* s4 = (const uint4*)src;
+    uint4* d4 = (uint4*)tile;
+    for (int i = threadIdx.x; i < total; i += blockDim.x) d4[i] = s4[i];
   }
   __syncthreads();
 
-  // phase 3: cv[e] = sum_c attn[c]*ccv[c,e] (thread-per-column, coalesced)
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int lane16 = lane & 15;
+  const int group = (threadIdx.x >> 4);  // 16 groups of 16 lanes
+
+  // phase 1: scores — each 16-lane group owns one context per iteration
+  for (int c = group; c < C; c += 16) {
+    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
+    float s = group_row_dot(row, lds_a, EP, lane16);
+    s = group16_reduce_sum(s);
+    if (lane16 == 0) {
+      const float mask = starts[(long)b * C + c] > 0 ? 1.0f : 0.0f;
+      scores[c] = s * mask + (1.0f - mask) * NINF_F;
+    }
+  }
+  __syncthreads();
+
+  // phase 2: softmax (scores -> attn probabilities, also written out)
+  block_softmax(scores, red, attn, C, (long)b * C);
+
+  // phase 3: cv[e] = sum_c attn[c]*ccv[c,e] — each wave accumulates a
+  // contiguous context span into per-lane column pairs, then cross-wave sum
+  const int span = (C + 3) / 4;
+  const int c_lo = wave * span;
+  const int c_hi = min(C, c_lo + span);
+  for (int e0 = lane * 2; e0 < EP; e0 += WAVE * 2) {
+    float acc0 = 0.f, acc1 = 0.f;
+    for (int c = c_lo; c < c_hi; ++c) {
+      const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
+      const bf16x2 v = *(const bf16x2*)(row + e0);
+      const float at = scores[c];
+      acc0 += at * bf2f(v[0]);
+      acc1 += at * bf2f(v[1]);
+    }
+    partials[wave * EP + e0] = acc0;
+    partials[wave * EP + e0 + 1] = acc1;
+  }
+  __syncthreads();
   for (int e = threadIdx.x; e < EP; e += blockDim.x) {
-    float acc = 0.f;
-    const bf16* base = (STAGE_LDS ? tile : src) + e;
-    for (int c = 0; c < C; ++c) acc += scores[c] * bf2f(base[(long)c * EP]);
-    cv[(long)b * EP + e] = acc;
+    cv[(long)b * EP + e] = partials[e] + partials[EP + e] +
+                           partials[2 * EP + e] + partials[3 * EP + e];
   }
 }
 
 // ---------------------------------------------------------------------------
 // Backward.  g[c] = dot(dcv, ccv[c]) (+ dattn[c]); sum_g = sum_c attn[c]*g[c];
-// ds[c] = attn[c]*(g[c]-sum_g);  dccv[c,e] = attn[c]*dcv[e] + ds[c]*mask*a[e];
-// da[e] += sum_c ds[c]*mask[c]*ccv[c,e].
+// ds[c] = attn[c]*(g[c]-sum_g)*mask;  dccv[c,e] = attn[c]*dcv[e] + ds[c]*a[e];
+// da[e] = sum_c ds[c]*ccv[c,e]  (per-block partials row, summed host-side).
 template <bool STAGE_LDS>
 __global__ __launch_bounds__(256) void attention_bwd_kernel(
     const float* __restrict__ dcv, const float* __restrict__ dattn,
@@ -114,8 +156,9 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   bf16* tile = (bf16*)smem;
   float* lds_dcv = (float*)(smem + (STAGE_LDS ? (size_t)C * EP * 2 : 0));
   float* lds_a = lds_dcv + EP;
-  float* ds = lds_a + EP;     // [C]
-  float* red = ds + C;        // [64]
+  float* ds = lds_a + EP;        // [C]
+  float* red = ds + C;           // [64]
+  float* partials = red + 64;    // [4][EP] (da)
 
   const bf16* src = ccv + (long)b * C * EP;
   for (int e = threadIdx.x; e < EP; e += blockDim.x) {
@@ -130,30 +173,28 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   }
   __syncthreads();
 
-  // g[c] and partial sum_g
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const int nwaves = blockDim.x / WAVE;
+  const int lane16 = lane & 15;
+  const int group = (threadIdx.x >> 4);
+
+  // phase A: g[c] (group-per-context) and block-reduced sum_g
   float part = 0.f;
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+  for (int c = group; c < C; c += 16) {
     const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
-    float g = 0.f;
-    for (int e = 0; e < EP; e += 8) {
-      bf16 v[8];
-      *(uint4*)v = *(const uint4*)(row + e);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) g += bf2f(v[j]) * lds_dcv[e + j];
+    float g = group_row_dot(row, lds_dcv, EP, lane16);
+    g = group16_reduce_sum(g);
+    if (lane16 == 0) {
+      if (has_dattn) g += dattn[(long)b * C + c];
+      ds[c] = g;  // temporarily g
+      part += attn[(long)b * C + c] * g;
     }
-    if (has_dattn) g += dattn[(long)b * C + c];
-    const float at = attn[(long)b * C + c];
-    ds[c] = g;  // temporarily g
-    part += at * g;
   }
   part = wave_reduce_sum(part);
   if (lane == 0) red[wave] = part;
   __syncthreads();
   if (wave == 0) {
-    float v = lane < nwaves ? red[lane] : 0.f;
+    float v = lane < (blockDim.x / WAVE) ? red[lane] : 0.f;
     v = wave_reduce_sum(v);
     if (lane == 0) red[0] = v;
   }
@@ -166,22 +207,48 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   }
   __syncthreads();
 
-  // dccv + da partials (thread-per-column)
-  for (int e = threadIdx.x; e < EP; e += blockDim.x) {
-    const float dcv_e = lds_dcv[e];
-    const float a_e = lds_a[e];
-    float da_acc = 0.f;
-    const bf16* base = (STAGE_LDS ? tile : src) + e;
-    bf16* dbase = dccv + (long)b * C * EP + e;
-    for (int c = 0; c < C; ++c) {
-      const float at = attn[(long)b * C + c];
-      const float d = at * dcv_e + ds[c] * a_e;
-      dbase[(long)c * EP] = f2bf(d);
-      da_acc += ds[c] * bf2f(base[(long)c * EP]);
+  // phase B: dccv rows (group-per-context, coalesced bf16x8 writes) and
+  // per-lane da partial accumulators
+  // da accumulation: this lane's columns are (lane16*8 + j) + 128*i;
+  // statically unrolled i (dynamic array indexing would go to scratch)
+  float da_acc[3][8] = {};  // EP <= 384
+  for (int c = group; c < C; c += 16) {
+    const float at = attn[(long)b * C + c];
+    const float dsc = ds[c];
+    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
+    bf16* drow = dccv + ((long)b * C + c) * EP;
+#pragma unroll
+    for (int i = 0; i < 3; ++i) {
+      const int e0 = lane16 * 8 + 128 * i;
+      if (e0 < EP) {
+        bf16 v[8], o[8];
+        *(uint4*)v = *(const uint4*)(row + e0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          o[j] = f2bf(at * lds_dcv[e0 + j] + dsc * lds_a[e0 + j]);
+          da_acc[i][j] += dsc * bf2f(v[j]);
+        }
+        *(uint4*)(drow + e0) = *(uint4*)o;
+      }
     }
-    // per-block partials row (summed host-side; avoids same-address atomics)
-    da[(long)b * EP + e] = e < E ? da_acc : 0.f;
   }
+  // combine da partials: each group's lane holds 8 cols (+128*i); groups
+  // 0..15 across 4 waves all accumulated different context spans -> LDS sum
+  __syncthreads();  // ds/tile reads done; reuse partials region
+  for (int e = threadIdx.x; e < EP; e += blockDim.x) partials[e] = 0.f;
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 3; ++i) {
+    const int e_base = lane16 * 8 + 128 * i;
+    if (e_base < EP) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        atomicAdd(&partials[e_base + j], da_acc[i][j]);
+    }
+  }
+  __syncthreads();
+  for (int e = threadIdx.x; e < EP; e += blockDim.x)
+    da[(long)b * EP + e] = e < E ? partials[e] : 0.f;
 }
 
 extern "C" {
@@ -192,6 +259,7 @@ static size_t attn_smem(int C, int EP, bool stage, bool bwd) {
   if (bwd) s += (size_t)EP * 4;   // a (bwd extra)
   s += (size_t)C * 4;             // scores / ds
   s += 64 * 4;                    // reduction scratch
+  s += (size_t)4 * EP * 4;        // partials
   return s;
 }
 
