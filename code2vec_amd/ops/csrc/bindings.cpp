@@ -30,6 +30,15 @@ void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
                          float, unsigned long long, unsigned long long, int,
                          hipStream_t);
+void launch_gather_combiner_fwd(const int*, const int*, const int*,
+                                const void*, const void*, int, int,
+                                const void*, const float*, const float*,
+                                void*, void*, float*, float*, long, int, int,
+                                int, float, unsigned long long,
+                                unsigned long long, hipStream_t);
+void launch_wgrad_gather(const int*, const int*, const int*, const void*,
+                         const void*, int, int, const void*, float*, long,
+                         int, int, int, hipStream_t);
 void launch_combiner_bwd(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*, void*,
                          float*, float*, long, int, int, float, hipStream_t);
@@ -143,6 +152,46 @@ void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                       EP, (int)E, (float)p, (unsigned long long)seed,
                       (unsigned long long)offset, (int)epilogue_mode,
                       cur_stream());
+}
+
+void gather_combiner_fwd(torch::Tensor starts, torch::Tensor paths,
+                         torch::Tensor ends, torch::Tensor term,
+                         torch::Tensor path, torch::Tensor w,
+                         torch::Tensor gamma, torch::Tensor beta,
+                         torch::Tensor out, torch::Tensor z,
+                         torch::Tensor mean, torch::Tensor rstd, int64_t KP,
+                         int64_t E, double p, int64_t seed, int64_t offset) {
+  CHK_CUDA(starts); CHK_CONTIG(starts); CHK_DT(starts, torch::kInt32);
+  CHK_DT(term, torch::kBFloat16); CHK_CONTIG(term);
+  CHK_DT(path, torch::kBFloat16); CHK_CONTIG(path);
+  CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
+  CHK_DT(out, torch::kBFloat16); CHK_DT(z, torch::kBFloat16);
+  const long M = starts.numel();
+  const int EP = w.size(0);
+  const int TS = term.size(1), PS = path.size(1);
+  TORCH_CHECK(w.size(1) == KP && KP >= 2 * TS + PS, "shapes");
+  launch_gather_combiner_fwd(
+      starts.data_ptr<int>(), paths.data_ptr<int>(), ends.data_ptr<int>(),
+      term.data_ptr(), path.data_ptr(), TS, PS, w.data_ptr(),
+      gamma.data_ptr<float>(), beta.data_ptr<float>(), out.data_ptr(),
+      z.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), M,
+      (int)KP, EP, (int)E, (float)p, (unsigned long long)seed,
+      (unsigned long long)offset, cur_stream());
+}
+
+void wgrad_gather(torch::Tensor starts, torch::Tensor paths,
+                  torch::Tensor ends, torch::Tensor term, torch::Tensor path,
+                  torch::Tensor dz, torch::Tensor partials, int64_t KP) {
+  CHK_CUDA(dz); CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
+  CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
+  const long M = starts.numel();
+  const int EP = dz.size(1);
+  const int TS = term.size(1), PS = path.size(1);
+  TORCH_CHECK(partials.size(1) == KP && partials.size(2) == EP, "partials");
+  launch_wgrad_gather(starts.data_ptr<int>(), paths.data_ptr<int>(),
+                      ends.data_ptr<int>(), term.data_ptr(), path.data_ptr(),
+                      TS, PS, dz.data_ptr(), partials.data_ptr<float>(), M,
+                      (int)KP, EP, partials.size(0), cur_stream());
 }
 
 void combiner_bwd(torch::Tensor dout, torch::Tensor z, torch::Tensor out,
@@ -293,6 +342,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_by_index", &group_by_index);
   m.def("cast_clear_rows", &cast_clear_rows);
   m.def("combiner_fwd", &combiner_fwd);
+  m.def("gather_combiner_fwd", &gather_combiner_fwd);
+  m.def("wgrad_gather", &wgrad_gather);
   m.def("combiner_bwd", &combiner_bwd);
   m.def("attention_fwd", &attention_fwd);
   m.def("attention_bwd", &attention_bwd);

@@ -33,14 +33,21 @@
 //   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
 //   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
 //   C/D: lane l, reg r holds C[row = (l>>4)*4 + r][col = l&15]
-template <int NT, int EPI>  // NT = EP/16 col tiles; EPI: 0 scalar stores, 1 LDS-bounce
+// GATHER=1 fuses K1/K2 into the GEMM: the A fragment reads the embedding
+// tables directly (segment-resolved), never materializing the concat
+// tensor.  Segment boundaries are 8-element multiples, so a 16-B fragment
+// piece never straddles one.
+template <int NT, int EPI, int GATHER>
 __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ Wt,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     bf16* __restrict__ out, bf16* __restrict__ z_save,
     float* __restrict__ mean_save, float* __restrict__ rstd_save,
     long M, int KP, int E, float p, float inv1mp,
-    unsigned long long seed, unsigned long long offset) {
+    unsigned long long seed, unsigned long long offset,
+    const int* __restrict__ g_starts, const int* __restrict__ g_paths,
+    const int* __restrict__ g_ends, const bf16* __restrict__ g_term,
+    const bf16* __restrict__ g_path, int TS, int PS) {
   const int EP = NT * 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -68,10 +75,16 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     long r = row0 + wave * 32 + mi * 16 + (lane & 15);
     arow[mi] = r < M ? r : (M - 1);
   }
-#ifdef C2V_EXP_FIXED_A  // experiment: L2-hot A reads (pattern diagnosis)
-  arow[0] = (lane & 15);
-  arow[1] = 16 + (lane & 15);
-#endif
+  // fused-gather mode: this lane's three embedding rows, loaded once
+  const bf16* seg_base[2][3];
+  if (GATHER) {
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+      seg_base[mi][0] = g_term + (size_t)g_starts[arow[mi]] * TS;
+      seg_base[mi][1] = g_path + (size_t)g_paths[arow[mi]] * PS;
+      seg_base[mi][2] = g_term + (size_t)g_ends[arow[mi]] * TS;
+    }
+  }
   const int kj = (lane >> 4) * 8;  // this lane's k sub-offset within a K-step
 
   // B tile staged in LDS, double-buffered, T14 split (load regs early,
@@ -107,9 +120,22 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     const int buf = kk & 1;
     if (kk + 1 < NK) STAGE(kk + 1, buf ^ 1)
     bf16x8 a[2];
+    const int kfrag = kk * 32 + kj;
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi)
-      a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kk * 32 + kj);
+    for (int mi = 0; mi < 2; ++mi) {
+      if (GATHER) {
+        bf16x8 v = {};
+        if (kfrag < TS)
+          v = *(const bf16x8*)(seg_base[mi][0] + kfrag);
+        else if (kfrag < TS + PS)
+          v = *(const bf16x8*)(seg_base[mi][1] + kfrag - TS);
+        else if (kfrag < 2 * TS + PS)
+          v = *(const bf16x8*)(seg_base[mi][2] + kfrag - TS - PS);
+        a[mi] = v;
+      } else {
+        a[mi] = *(const bf16x8*)(X + arow[mi] * KP + kfrag);
+      }
+    }
     // explicit 2-deep B pipeline: read fragment n+1 while n's MFMAs issue
     const bf16* bbase =
         lds_b + buf * bufsz + (size_t)(lane & 15) * ESTRIDE + (lane >> 4) * 8;
@@ -353,15 +379,20 @@ __global__ __launch_bounds__(256) void combiner_bwd_kernel(
 
 extern "C" {
 
-void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
+void launch_combiner_fwd_impl(const void* X, const void* W,
+                         const float* gamma,
                          const float* beta, void* out, void* z, float* mean,
                          float* rstd, long M, int KP, int EP, int E, float p,
                          unsigned long long seed, unsigned long long offset,
-                         int epilogue_mode, hipStream_t stream) {
+                         int epilogue_mode, int gather, const int* starts,
+                         const int* paths, const int* ends, const void* term,
+                         const void* path, int TS, int PS,
+                         hipStream_t stream) {
   const int NT = EP / 16;
   const long grid = (M + 127) / 128;
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
   int epi = epilogue_mode;
+  (void)gather;
   int smem = 2 * EP * sizeof(float) + 2 * EP * 40 * (int)sizeof(bf16);
   if (epi == 1) {
     const int bounce = 4 * 32 * (EP + 8) * (int)sizeof(bf16);
@@ -370,14 +401,21 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
   }
 #define CASE(nt)                                                              \
   case nt:                                                                    \
-    if (epi == 1)                                                             \
-      combiner_fwd_kernel<nt, 1><<<grid, 256, smem, stream>>>(                \
+    if (gather)                                                               \
+      combiner_fwd_kernel<nt, 1, 1><<<grid, 256, smem, stream>>>(             \
+          nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset, starts, paths,       \
+          ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
+    else if (epi == 1)                                                        \
+      combiner_fwd_kernel<nt, 1, 0><<<grid, 256, smem, stream>>>(             \
           (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset);                     \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset, nullptr, nullptr,    \
+          nullptr, nullptr, nullptr, 0, 0);                                   \
     else                                                                      \
-      combiner_fwd_kernel<nt, 0><<<grid, 256, smem, stream>>>(                \
+      combiner_fwd_kernel<nt, 0, 0><<<grid, 256, smem, stream>>>(             \
           (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset);                     \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset, nullptr, nullptr,    \
+          nullptr, nullptr, nullptr, 0, 0);                                   \
     break;
   switch (NT) {
     CASE(2) CASE(4) CASE(6) CASE(8) CASE(10)
@@ -386,6 +424,31 @@ void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
       printf("combiner_fwd: unsupported EP=%d\n", EP);
   }
 #undef CASE
+}
+
+void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
+                         const float* beta, void* out, void* z, float* mean,
+                         float* rstd, long M, int KP, int EP, int E, float p,
+                         unsigned long long seed, unsigned long long offset,
+                         int epilogue_mode, hipStream_t stream) {
+  launch_combiner_fwd_impl(X, W, gamma, beta, out, z, mean, rstd, M, KP, EP,
+                           E, p, seed, offset, epilogue_mode, 0, nullptr,
+                           nullptr, nullptr, nullptr, nullptr, 0, 0, stream);
+}
+
+void launch_gather_combiner_fwd(const int* starts, const int* paths,
+                                const int* ends, const void* term,
+                                const void* path, int TS, int PS,
+                                const void* W, const float* gamma,
+                                const float* beta, void* out, void* z,
+                                float* mean, float* rstd, long M, int KP,
+                                int EP, int E, float p,
+                                unsigned long long seed,
+                                unsigned long long offset,
+                                hipStream_t stream) {
+  launch_combiner_fwd_impl(nullptr, W, gamma, beta, out, z, mean, rstd, M,
+                           KP, EP, E, p, seed, offset, 1, 1, starts, paths,
+                           ends, term, path, TS, PS, stream);
 }
 
 void launch_combiner_bwd(const void* dout, const void* z, const void* out,

@@ -15,10 +15,15 @@
 
 #include "common.h"
 
-template <int NT, int IT>  // NT = EP/16 col tiles; IT = KP/16/8 row tiles/wave
+// GATHER=1: the X operand is re-gathered from the embedding tables during
+// staging (the fused forward never materializes the concat tensor).
+template <int NT, int IT, int GATHER>
 __global__ __launch_bounds__(512) void wgrad_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ dZ,
-    float* __restrict__ partials, long M, int KP, int EP, long rows_per_block) {
+    float* __restrict__ partials, long M, int KP, int EP, long rows_per_block,
+    const int* __restrict__ g_starts, const int* __restrict__ g_paths,
+    const int* __restrict__ g_ends, const bf16* __restrict__ g_term,
+    const bf16* __restrict__ g_path, int TS, int PS) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const long m0 = (long)blockIdx.x * rows_per_block;
@@ -54,8 +59,21 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
       uint4 v = {0, 0, 0, 0};
       if (c < chunks_x) {
         const int krow = c / (KP / 8);
-        if (krow < rows)
-          v = *(const uint4*)(X + (mb + krow) * KP + (c % (KP / 8)) * 8);
+        const int col8 = (c % (KP / 8)) * 8;
+        if (krow < rows) {
+          if (GATHER) {
+            const long r = mb + krow;
+            if (col8 < TS)
+              v = *(const uint4*)(g_term + (size_t)g_starts[r] * TS + col8);
+            else if (col8 < TS + PS)
+              v = *(const uint4*)(g_path + (size_t)g_paths[r] * PS + col8 - TS);
+            else if (col8 < 2 * TS + PS)
+              v = *(const uint4*)(g_term + (size_t)g_ends[r] * TS + col8 -
+                                  TS - PS);
+          } else {
+            v = *(const uint4*)(X + (mb + krow) * KP + col8);
+          }
+        }
       }
       xr[i] = v;
     }
@@ -156,8 +174,10 @@ __global__ __launch_bounds__(512) void wgrad_kernel(
 
 extern "C" {
 
-void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
-                  int KP, int EP, int nblocks, hipStream_t stream) {
+void launch_wgrad_impl(const void* X, const void* dZ, float* partials,
+                  long M, int KP, int EP, int nblocks, const int* starts,
+                  const int* paths, const int* ends, const void* term,
+                  const void* path, int TS, int PS, hipStream_t stream) {
   const long rows_per_block = ((M + nblocks - 1) / nblocks + 31) / 32 * 32;
   const int smem = 2 * ((KP / 16) + (EP / 16)) * (64 * 8 + 8) * 2;
   const int NT = EP / 16;
@@ -165,9 +185,16 @@ void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
   // supported shapes: KP/16 divisible into 8 waves; dispatch common cases
 #define WCASE(nt, it)                                                        \
   if (NT == nt && IT == it) {                                                \
-    wgrad_kernel<nt, it><<<nblocks, 512, smem, stream>>>(                    \
-        (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,                \
-        rows_per_block);                                                     \
+    if (starts)                                                              \
+      wgrad_kernel<nt, it, 1><<<nblocks, 512, smem, stream>>>(               \
+          nullptr, (const bf16*)dZ, partials, M, KP, EP, rows_per_block,     \
+          starts, paths, ends, (const bf16*)term, (const bf16*)path, TS,     \
+          PS);                                                               \
+    else                                                                     \
+      wgrad_kernel<nt, it, 0><<<nblocks, 512, smem, stream>>>(               \
+          (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,              \
+          rows_per_block, nullptr, nullptr, nullptr, nullptr, nullptr, 0,    \
+          0);                                                                \
     return;                                                                  \
   }
   // NT > 8 (encode > 128) overflows the register budget -> callers fall
@@ -178,6 +205,21 @@ void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
   WCASE(2, 1) WCASE(2, 2) WCASE(2, 3) WCASE(2, 4)
 #undef WCASE
   printf("wgrad: unsupported KP=%d EP=%d\n", KP, EP);
+}
+
+void launch_wgrad(const void* X, const void* dZ, float* partials, long M,
+                  int KP, int EP, int nblocks, hipStream_t stream) {
+  launch_wgrad_impl(X, dZ, partials, M, KP, EP, nblocks, nullptr, nullptr,
+                    nullptr, nullptr, nullptr, 0, 0, stream);
+}
+
+void launch_wgrad_gather(const int* starts, const int* paths,
+                         const int* ends, const void* term, const void* path,
+                         int TS, int PS, const void* dZ, float* partials,
+                         long M, int KP, int EP, int nblocks,
+                         hipStream_t stream) {
+  launch_wgrad_impl(nullptr, dZ, partials, M, KP, EP, nblocks, starts, paths,
+                    ends, term, path, TS, PS, stream);
 }
 
 }  // extern "C"

@@ -20,6 +20,8 @@ import torch
 # 1 = per-wave LDS bounce + coalesced 16-B flush (kbench-selected)
 _FWD_EPI = int(os.environ.get("C2V_FWD_EPI", "1"))
 _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
+# fused gather+combiner forward (C2V_NO_FUSE=1 falls back to two kernels)
+FUSE_GATHER_COMBINER = os.environ.get("C2V_NO_FUSE") != "1"
 
 from . import ext, round_up
 
@@ -65,34 +67,38 @@ class GatherConcat(torch.autograd.Function):
     def backward(ctx, grad_out):
         starts, paths, ends = ctx.saved_tensors
         term_shape, path_shape = ctx.shapes
-        dev = grad_out.device
-        M = starts.numel()
-        TS, PS = term_shape[1], path_shape[1]
-        gout = grad_out.contiguous()
-        KP = gout.shape[1]
-        # persistent fp32 scatter scratch — zeroed once; the fused
-        # cast-and-clear kernel re-zeroes exactly the rows it consumed, so
-        # no full-buffer memset per step
-        dterm32 = _scratch_f32(term_shape, dev)
-        dpath32 = _scratch_f32(path_shape, dev)
-        flags_t = _scratch_flags(term_shape[0], dev)
-        flags_p = _scratch_flags(path_shape[0], dev)
-        # sort-based segmented scatter (K13 v3): counting-sort groups the
-        # index lists; run-owner waves write interior rows' bf16 grads
-        # directly; boundary-crossing runs (heavy hitters) combine via
-        # flagged fp32 scratch in cast_clear_rows.
-        dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
-        dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
-        idx_se = torch.cat([starts.view(-1), ends.view(-1)])
-        sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
-        ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32, dterm,
-                                   flags_t, M, KP, 0, TS + PS, _SCATTER_R)
-        sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
-        ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
-                                   flags_p, M, KP, TS, TS, _SCATTER_R)
-        ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
-        ext().cast_clear_rows(dpath32, counts_p, flags_p, dpath)
+        dterm, dpath = _scatter_embedding_grads(
+            starts, paths, ends, grad_out.contiguous(), term_shape, path_shape
+        )
         return None, None, None, dterm, dpath
+
+
+def _scatter_embedding_grads(starts, paths, ends, gout, term_shape, path_shape):
+    """K13 v4: sort-based segmented scatter of a [M, KP] grad into bf16
+    dense embedding grads.  Counting-sort groups the index lists; run-owner
+    waves write interior rows' bf16 grads directly; boundary-crossing runs
+    (heavy hitters) combine via flagged persistent fp32 scratch in
+    cast_clear_rows (which also re-zeroes what it consumed)."""
+    dev = gout.device
+    M = starts.numel()
+    TS, PS = term_shape[1], path_shape[1]
+    KP = gout.shape[1]
+    dterm32 = _scratch_f32(term_shape, dev)
+    dpath32 = _scratch_f32(path_shape, dev)
+    flags_t = _scratch_flags(term_shape[0], dev)
+    flags_p = _scratch_flags(path_shape[0], dev)
+    dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
+    dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
+    idx_se = torch.cat([starts.view(-1), ends.view(-1)])
+    sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
+    ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32, dterm,
+                               flags_t, M, KP, 0, TS + PS, _SCATTER_R)
+    sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
+    ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
+                               flags_p, M, KP, TS, TS, _SCATTER_R)
+    ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
+    ext().cast_clear_rows(dpath32, counts_p, flags_p, dpath)
+    return dterm, dpath
 
 
 _scratch_cache = {}
@@ -199,6 +205,68 @@ class CombinerLNTanh(torch.autograd.Function):
         else:
             dw = (x.t() @ dz).t().contiguous()
         return dx, dw, dgamma, dbeta, None, None, None
+
+
+class FusedGatherCombiner(torch.autograd.Function):
+    """K1-K6 fully fused: embedding gather + MFMA combiner GEMM + LN + tanh
+    + dropout in ONE kernel — the [M, KP] concat tensor is never
+    materialized on the forward path.  Backward: fused LN/tanh/dropout
+    chain -> dz; dW via the re-gathering split-K wgrad; dX (for the
+    embedding scatter) via one rocBLAS GEMM; embedding grads via the
+    sort-based scatter."""
+
+    @staticmethod
+    def forward(ctx, starts, paths, ends, term_w, path_w, w, gamma, beta,
+                E: int, p: float, training: bool):
+        M = starts.numel()
+        EP, KP = w.shape
+        dev = starts.device
+        out = torch.empty(M, EP, dtype=torch.bfloat16, device=dev)
+        z = torch.empty(M, EP, dtype=torch.bfloat16, device=dev)
+        mean = torch.empty(M, dtype=torch.float32, device=dev)
+        rstd = torch.empty(M, dtype=torch.float32, device=dev)
+        p_eff = float(p) if training else 0.0
+        seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
+        ext().gather_combiner_fwd(starts, paths, ends, term_w, path_w, w,
+                                  gamma, beta, out, z, mean, rstd, KP, E,
+                                  p_eff, seed, offset)
+        ctx.save_for_backward(starts, paths, ends, term_w, path_w, w, gamma,
+                              beta, z, mean, rstd, out)
+        ctx.meta = (E, p_eff)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (starts, paths, ends, term_w, path_w, w, gamma, beta, z, mean, rstd,
+         out) = ctx.saved_tensors
+        E, p = ctx.meta
+        M, EP = z.shape
+        KP = w.shape[1]
+        dz = torch.empty(M, EP, dtype=torch.bfloat16, device=z.device)
+        nblocks = min((M + 4 - 1) // 4, 1024)
+        dgamma_p = torch.empty(nblocks, EP, dtype=torch.float32, device=z.device)
+        dbeta_p = torch.empty(nblocks, EP, dtype=torch.float32, device=z.device)
+        ext().combiner_bwd(dout.contiguous(), z, out, mean, rstd, gamma,
+                           beta, dz, dgamma_p, dbeta_p, E, p)
+        dgamma = dgamma_p.sum(dim=0)
+        dbeta = dbeta_p.sum(dim=0)
+        # dW: re-gathering split-K wgrad (X is never materialized)
+        if EP <= 128 and KP <= 512:
+            partials = torch.empty(256, KP, EP, dtype=torch.float32,
+                                   device=z.device)
+            ext().wgrad_gather(starts, paths, ends, term_w, path_w, dz,
+                               partials, KP)
+            dw = partials.sum(dim=0).t().contiguous().to(torch.bfloat16)
+        else:
+            x = GatherConcat.apply(starts, paths, ends, term_w, path_w)
+            dw = (x.t() @ dz).t().contiguous()
+        # dX for the embedding scatter: one plain GEMM
+        dx = dz @ w
+        dterm, dpath = _scatter_embedding_grads(
+            starts, paths, ends, dx, term_w.shape, path_w.shape
+        )
+        return (None, None, None, dterm, dpath, dw, dgamma, dbeta,
+                None, None, None)
 
 
 class AttentionPool(torch.autograd.Function):
