@@ -83,10 +83,10 @@ def _scatter_embedding_grads(starts, paths, ends, gout, term_shape, path_shape):
     M = starts.numel()
     TS, PS = term_shape[1], path_shape[1]
     KP = gout.shape[1]
-    dterm32 = _scratch_f32(term_shape, dev)
-    dpath32 = _scratch_f32(path_shape, dev)
-    flags_t = _scratch_flags(term_shape[0], dev)
-    flags_p = _scratch_flags(path_shape[0], dev)
+    dterm32 = _scratch_f32("term", term_shape, dev)
+    dpath32 = _scratch_f32("path", path_shape, dev)
+    flags_t = _scratch_flags("term", term_shape[0], dev)
+    flags_p = _scratch_flags("path", path_shape[0], dev)
     dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
     dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
     idx_se = torch.cat([starts.view(-1), ends.view(-1)])
@@ -104,10 +104,11 @@ def _scatter_embedding_grads(starts, paths, ends, gout, term_shape, path_shape):
 _scratch_cache = {}
 
 
-def _scratch_f32(shape, device) -> torch.Tensor:
+def _scratch_f32(tag: str, shape, device) -> torch.Tensor:
     """Persistent fp32 scatter scratch (invariant: all-zero between steps —
-    maintained by cast_clear_rows)."""
-    key = (tuple(shape), str(device))
+    maintained by cast_clear_rows).  ``tag`` keeps the term/path buffers
+    distinct even when both tables have identical shapes."""
+    key = (tag, tuple(shape), str(device))
     buf = _scratch_cache.get(key)
     if buf is None:
         buf = torch.zeros(shape, dtype=torch.float32, device=device)
@@ -115,8 +116,8 @@ def _scratch_f32(shape, device) -> torch.Tensor:
     return buf
 
 
-def _scratch_flags(rows: int, device) -> torch.Tensor:
-    key = ("flags", rows, str(device))
+def _scratch_flags(tag: str, rows: int, device) -> torch.Tensor:
+    key = ("flags", tag, rows, str(device))
     buf = _scratch_cache.get(key)
     if buf is None:
         buf = torch.zeros(rows, dtype=torch.uint8, device=device)

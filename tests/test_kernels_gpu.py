@@ -347,3 +347,49 @@ class TestCombinerDropoutBwd:
         out_ref.backward(gout[:, :E].to(torch.bfloat16).float())
         assert relerr(xh.grad, xr.grad) < 5e-2
         assert relerr(wh.grad[:E, :], wr.grad[:E, :]) < 5e-2
+
+
+class TestScatterProperty:
+    def test_group_by_index_is_a_sort(self, dev):
+        from code2vec_amd.ops.functional import _group_by_index
+
+        g = torch.Generator(device=dev).manual_seed(0)
+        idx = torch.randint(0, 997, (50_000,), generator=g, device=dev,
+                            dtype=torch.int32)
+        sorted_idx, perm, counts = _group_by_index(idx, 1000)
+        ref, _ = torch.sort(idx)
+        assert torch.equal(sorted_idx, ref)
+        # perm is a permutation mapping back to the original values
+        assert torch.equal(idx[perm], sorted_idx)
+        assert int(counts[:1000].sum()) == 50_000
+
+    @pytest.mark.parametrize("n_distinct", [7, 1000, 200_000])
+    def test_scatter_matches_index_add_oracle(self, dev, n_distinct):
+        """Heavy-hitter runs (n_distinct=7 -> runs span hundreds of chunks)
+        exercise the boundary fp32-atomic path; large n_distinct the
+        interior bf16-direct path."""
+        from code2vec_amd.ops import functional as Fn
+
+        T, S, M = 250_000, 104, 60_000
+        KP = 320
+        g = torch.Generator(device=dev).manual_seed(3)
+        starts = torch.randint(1, n_distinct + 1, (M,), generator=g,
+                               device=dev, dtype=torch.int32)
+        ends = torch.randint(1, n_distinct + 1, (M,), generator=g,
+                             device=dev, dtype=torch.int32)
+        paths = torch.randint(1, n_distinct + 1, (M,), generator=g,
+                              device=dev, dtype=torch.int32)
+        gout = (torch.randn(M, KP, generator=g, device=dev) * 0.1).to(
+            torch.bfloat16)
+        dterm, dpath = Fn._scatter_embedding_grads(
+            starts, paths, ends, gout, (T, S), (T, S))
+
+        ref_t = torch.zeros(T, S, dtype=torch.float32, device=dev)
+        ref_t.index_add_(0, starts.long(), gout[:, :S].float())
+        ref_t.index_add_(0, ends.long(), gout[:, 208:208 + S].float())
+        ref_p = torch.zeros(T, S, dtype=torch.float32, device=dev)
+        ref_p.index_add_(0, paths.long(), gout[:, 104:104 + S].float())
+        err_t = (dterm.float() - ref_t).norm() / (ref_t.norm() + 1e-9)
+        err_p = (dpath.float() - ref_p).norm() / (ref_p.norm() + 1e-9)
+        assert float(err_t) < 2e-2, float(err_t)
+        assert float(err_p) < 2e-2, float(err_p)
