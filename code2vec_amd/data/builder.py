@@ -202,14 +202,32 @@ class DatasetBuilder:
         self._flat_cache[tag] = cached
         return cached
 
+    def _pinned_pool(self, tag: str, N: int, C: int):
+        """Persistent (optionally pinned) per-epoch output buffers: the
+        native builder writes straight into them, so the loader's H2D path
+        needs no per-epoch pin_memory copy of ~GB index arrays."""
+        key = ("pool", tag, N, C)
+        bufs = self._flat_cache.get(key)
+        if bufs is None:
+            pin = _torch.cuda.is_available()
+            bufs = tuple(
+                _torch.empty(N, C, dtype=_torch.int32, pin_memory=pin)
+                for _ in range(3)
+            )
+            self._flat_cache[key] = bufs
+        elif _torch.cuda.is_available():
+            # the previous epoch's async H2D copies may still read these
+            # buffers; fence before overwriting (NOTE: an EpochData aliases
+            # its pool — it is invalidated by the next refresh of its split)
+            _torch.cuda.synchronize()
+        return bufs
+
     def _build_native(self, base, tag, shard_idx, epoch, stream) -> EpochData:
         offsets_t, contexts_t, labels, ids = self._get_flat(base, tag)
         C = self.option.max_path_length
         N = len(shard_idx)
         item_idx = _torch.from_numpy(np.ascontiguousarray(shard_idx, dtype=np.int64))
-        starts = _torch.empty(N, C, dtype=_torch.int32)
-        paths = _torch.empty(N, C, dtype=_torch.int32)
-        ends = _torch.empty(N, C, dtype=_torch.int32)
+        starts, paths, ends = self._pinned_pool(tag, N, C)
         seed = hash((self.seed, epoch, stream, self.rank)) & 0x7FFFFFFFFFFFFFFF
         method_token_index = self.reader.terminal_vocab.stoi["@method_0"]
         _native.build_method_epoch(

@@ -45,10 +45,13 @@ class BatchIterator:
             [i if i is not None else -1 for i in data.ids], dtype=torch.int64
         )
         if self.prefetch:
-            self.starts = self.starts.pin_memory()
-            self.paths = self.paths.pin_memory()
-            self.ends = self.ends.pin_memory()
-            self.labels = self.labels.pin_memory()
+            def _pin(t):
+                return t if t.is_pinned() else t.pin_memory()
+
+            self.starts = _pin(self.starts)
+            self.paths = _pin(self.paths)
+            self.ends = _pin(self.ends)
+            self.labels = _pin(self.labels)
             self._copy_stream = torch.cuda.Stream(self.device)
 
     def __len__(self) -> int:

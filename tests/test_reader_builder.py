@@ -78,14 +78,17 @@ def test_builder_resamples_per_epoch(tiny_corpus):
     r = make_reader(tiny_corpus)
     opt = make_option(r, max_path_length=4)  # force truncation
     b = DatasetBuilder(r, opt, seed=11)
-    d0 = b.refresh_train_dataset(epoch=0)
+    # NOTE: an EpochData aliases the builder's persistent buffers and is
+    # invalidated by the next refresh of the same split — copy to compare.
+    d0_paths = b.refresh_train_dataset(epoch=0).paths.copy()
+    d0_labels = list(b.refresh_train_dataset(epoch=0).labels.tolist())
     d1 = b.refresh_train_dataset(epoch=1)
     # same items, different resampled contexts
-    assert d0.labels.tolist() == d1.labels.tolist()
-    assert not np.array_equal(d0.paths, d1.paths)
+    assert d0_labels == d1.labels.tolist()
+    assert not np.array_equal(d0_paths, d1.paths)
     # deterministic per (seed, epoch)
     d0b = b.refresh_train_dataset(epoch=0)
-    assert np.array_equal(d0.paths, d0b.paths)
+    assert np.array_equal(d0_paths, d0b.paths)
 
 
 def test_builder_question_replacement(tiny_corpus):
