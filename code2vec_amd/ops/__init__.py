@@ -18,9 +18,6 @@ Layout conventions (all HIP kernels assume these):
 from __future__ import annotations
 
 import importlib
-import os
-
-import torch
 
 PAD = 32  # element granularity of the encode dim (MFMA N tiles)
 SEG_PAD = 8  # element granularity of embedding rows / concat segments (16 B)

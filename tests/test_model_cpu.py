@@ -4,7 +4,6 @@
 import math
 
 import numpy as np
-import pytest
 import torch
 import torch.nn.functional as F
 

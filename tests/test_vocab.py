@@ -1,8 +1,6 @@
 """Vocab semantics parity tests (reference model/dataset.py:52-93,
 model/dataset_reader.py:15-41)."""
 
-import pytest
-
 from code2vec_amd.data.vocab import (
     QUESTION_TOKEN_INDEX,
     Vocab,
