@@ -23,6 +23,13 @@ _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 # fused gather+combiner forward (C2V_NO_FUSE=1 falls back to two kernels)
 FUSE_GATHER_COMBINER = os.environ.get("C2V_NO_FUSE") != "1"
 
+# Early-gradient callbacks (parallel/ddp.py registers one per big parameter,
+# keyed by param.data_ptr()): the embedding backward invokes them the moment
+# a table's grad tensor is complete, so at DP>1 the terminal-table
+# all-reduce overlaps with the path table's scatter pipeline instead of
+# waiting for the whole backward to finish.
+EARLY_GRAD_CALLBACKS = {}
+
 from . import ext, round_up
 
 _rng_state = {"seed": None, "offset": 0}
@@ -61,6 +68,7 @@ class GatherConcat(torch.autograd.Function):
         ext().gather_concat_fwd(starts, paths, ends, term_w, path_w, out)
         ctx.save_for_backward(starts, paths, ends)
         ctx.shapes = (term_w.shape, path_w.shape)
+        ctx.param_keys = (term_w.data_ptr(), path_w.data_ptr())
         return out
 
     @staticmethod
@@ -68,12 +76,14 @@ class GatherConcat(torch.autograd.Function):
         starts, paths, ends = ctx.saved_tensors
         term_shape, path_shape = ctx.shapes
         dterm, dpath = _scatter_embedding_grads(
-            starts, paths, ends, grad_out.contiguous(), term_shape, path_shape
+            starts, paths, ends, grad_out.contiguous(), term_shape,
+            path_shape, ctx.param_keys[0], ctx.param_keys[1]
         )
         return None, None, None, dterm, dpath
 
 
-def _scatter_embedding_grads(starts, paths, ends, gout, term_shape, path_shape):
+def _scatter_embedding_grads(starts, paths, ends, gout, term_shape,
+                             path_shape, term_key=None, path_key=None):
     """K13 v4: sort-based segmented scatter of a [M, KP] grad into bf16
     dense embedding grads.  Counting-sort groups the index lists; run-owner
     waves write interior rows' bf16 grads directly; boundary-crossing runs
@@ -97,7 +107,15 @@ def _scatter_embedding_grads(starts, paths, ends, gout, term_shape, path_shape):
     ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
                                flags_p, M, KP, TS, TS, _SCATTER_R)
     ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
+    if term_key is not None:
+        cb = EARLY_GRAD_CALLBACKS.get(term_key)
+        if cb is not None:
+            cb(dterm)
     ext().cast_clear_rows(dpath32, counts_p, flags_p, dpath)
+    if path_key is not None:
+        cb = EARLY_GRAD_CALLBACKS.get(path_key)
+        if cb is not None:
+            cb(dpath)
     return dterm, dpath
 
 
@@ -264,7 +282,8 @@ class FusedGatherCombiner(torch.autograd.Function):
         # dX for the embedding scatter: one plain GEMM
         dx = dz @ w
         dterm, dpath = _scatter_embedding_grads(
-            starts, paths, ends, dx, term_w.shape, path_w.shape
+            starts, paths, ends, dx, term_w.shape, path_w.shape,
+            term_w.data_ptr(), path_w.data_ptr()
         )
         return (None, None, None, dterm, dpath, dw, dgamma, dbeta,
                 None, None, None)

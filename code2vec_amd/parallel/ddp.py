@@ -67,6 +67,9 @@ class BucketedAllReduce:
         ]
         self._direct_set = set(id(p) for p in self.direct_params)
         self._direct_handles = []
+        self._early_reduced = set()  # data_ptrs all-reduced via callbacks
+        if self.enabled:
+            self._register_early_callbacks()
 
         # reverse order ~ autograd completion order; one dtype per bucket
         self.buckets: List[Bucket] = []
@@ -102,19 +105,36 @@ class BucketedAllReduce:
                 self._hooks.append(h)
         self._reset_pending()
 
+    def _register_early_callbacks(self):
+        """Let the embedding backward start a big tensor's all-reduce the
+        moment that grad is complete (ops/functional.EARLY_GRAD_CALLBACKS),
+        overlapping comm with the rest of backward."""
+        from ..ops import functional as Fn
+
+        for p in self.direct_params:
+            def cb(grad, _param=p, _self=self):
+                h = dist.all_reduce(grad, op=dist.ReduceOp.SUM,
+                                    group=_self.group, async_op=True)
+                _self._direct_handles.append((h, _param, grad.data_ptr()))
+                _self._early_reduced.add(grad.data_ptr())
+            Fn.EARLY_GRAD_CALLBACKS[p.data_ptr()] = cb
+
     def _reset_pending(self):
         for b in self.buckets:
             b.pending = len(b.params)
             b.handle = None
         self._direct_handles = []
+        self._early_reduced = set()
 
     def _on_grad_ready(self, param) -> None:
         if id(param) in self._direct_set:
+            if param.grad.data_ptr() in self._early_reduced:
+                return  # already launched from inside the backward
             h = dist.all_reduce(
                 param.grad, op=dist.ReduceOp.SUM, group=self.group,
                 async_op=True,
             )
-            self._direct_handles.append((h, param))
+            self._direct_handles.append((h, param, param.grad.data_ptr()))
             return
         b = self._param_bucket[id(param)]
         b.pending -= 1
@@ -127,8 +147,16 @@ class BucketedAllReduce:
         """Wait for outstanding all-reduces and average; call every step."""
         if not self.enabled:
             return
-        for h, p in self._direct_handles:
+        for h, p, gptr in self._direct_handles:
             h.wait()
+            if p.grad is None or p.grad.data_ptr() != gptr:
+                # autograd cloned instead of stealing the grad tensor; the
+                # reduction landed in the wrong buffer — fail loudly rather
+                # than silently train on unreduced gradients
+                raise RuntimeError(
+                    "early-reduced gradient was not adopted as p.grad "
+                    "(autograd cloned it); disable early callbacks"
+                )
             p.grad.div_(self.world_size)
         for b in self.buckets:
             if b.handle is not None:

@@ -174,3 +174,58 @@ def test_trainer_world2_runs(tmp_path):
         assert pr.exitcode == 0
     obj = float(open(result_file).read())
     assert 0.0 <= obj <= 1.0
+
+
+def _worker_early_cb(rank, world, port, result_file):
+    """Exercise the early-gradient-callback path (the one the embedding
+    backward uses on GPU): cb launches the all-reduce before the hook."""
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+    from code2vec_amd.ops import functional as Fn
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    p = torch.nn.Parameter(torch.zeros(4096))
+    ddp = BucketedAllReduce([p], world, direct_threshold=1024)
+    assert p.data_ptr() in Fn.EARLY_GRAD_CALLBACKS
+
+    # step 1: grad arrives via the early callback during "backward"
+    g = torch.full((4096,), float(rank + 1))
+    Fn.EARLY_GRAD_CALLBACKS[p.data_ptr()](g)
+    p.grad = g  # autograd adopts the same tensor
+    ddp._on_grad_ready(p)  # the hook must NOT reduce again
+    ddp.finish()
+    expected = (sum(range(1, world + 1)) / world)
+    assert torch.allclose(p.grad, torch.full((4096,), expected)), p.grad[0]
+
+    # step 2: no callback (hook path) still works after reset
+    ddp.zero_grad()
+    g2 = torch.full((4096,), float(10 * (rank + 1)))
+    p.grad = g2
+    ddp._on_grad_ready(p)
+    ddp.finish()
+    expected2 = 10 * sum(range(1, world + 1)) / world
+    assert torch.allclose(p.grad, torch.full((4096,), expected2)), p.grad[0]
+
+    if rank == 0:
+        with open(result_file, "w") as f:
+            f.write("ok")
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_early_grad_callback_world2(tmp_path):
+    ctx = mp.get_context("spawn")
+    result_file = str(tmp_path / "cb.txt")
+    procs = [ctx.Process(target=_worker_early_cb,
+                         args=(r, 2, 29533, result_file)) for r in range(2)]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    assert open(result_file).read() == "ok"
