@@ -160,7 +160,10 @@ class DatasetBuilder:
             and self.reader.infer_method
             and not self.reader.infer_variable
         ):
-            return self._build_native(base, tag, shard_idx, epoch, stream)
+            # pool double-buffered by epoch parity: epoch N+1 can build in a
+            # background thread while N's buffers feed H2D copies
+            return self._build_native(base, tag, f"{tag}{epoch % 2}",
+                                      shard_idx, epoch, stream)
         items = [base[i] for i in shard_idx]
         return self.build_data(items, self.option.max_path_length,
                                epoch=epoch, stream=stream)
@@ -222,12 +225,13 @@ class DatasetBuilder:
             _torch.cuda.synchronize()
         return bufs
 
-    def _build_native(self, base, tag, shard_idx, epoch, stream) -> EpochData:
-        offsets_t, contexts_t, labels, ids = self._get_flat(base, tag)
+    def _build_native(self, base, flat_tag, pool_tag, shard_idx, epoch,
+                      stream) -> EpochData:
+        offsets_t, contexts_t, labels, ids = self._get_flat(base, flat_tag)
         C = self.option.max_path_length
         N = len(shard_idx)
         item_idx = _torch.from_numpy(np.ascontiguousarray(shard_idx, dtype=np.int64))
-        starts, paths, ends = self._pinned_pool(tag, N, C)
+        starts, paths, ends = self._pinned_pool(pool_tag, N, C)
         seed = hash((self.seed, epoch, stream, self.rank)) & 0x7FFFFFFFFFFFFFFF
         method_token_index = self.reader.terminal_vocab.stoi["@method_0"]
         _native.build_method_epoch(

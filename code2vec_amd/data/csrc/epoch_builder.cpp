@@ -60,6 +60,10 @@ void build_method_epoch(torch::Tensor offsets, torch::Tensor contexts,
   const int32_t mtok = (int32_t)method_token_index;
   const int32_t qtok = (int32_t)question_index;
 
+  // release the GIL: the trainer builds epoch N+1 on a worker thread while
+  // epoch N trains on the GPU
+  pybind11::gil_scoped_release release;
+
 #pragma omp parallel
   {
     std::vector<int32_t> idx;

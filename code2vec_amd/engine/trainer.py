@@ -14,6 +14,8 @@ from typing import Optional
 import numpy as np
 import torch
 
+from concurrent.futures import ThreadPoolExecutor
+
 from ..data.builder import DatasetBuilder
 from ..models.code2vec import (
     Code2VecHIP,
@@ -110,6 +112,10 @@ class Trainer:
         self.ddp.broadcast_parameters()
 
         self.roctx = _Roctx(self.device.type == "cuda")
+        # epoch N+1's host-side rebuild runs here while N trains (the
+        # native C++ builder releases the GIL)
+        self._build_pool = ThreadPoolExecutor(max_workers=1)
+        self._next_train = None
         self.summary_writer = None
         if config.env == "tensorboard" and ctx.is_rank0:
             from tensorboardX import SummaryWriter  # optional dep
@@ -218,7 +224,11 @@ class Trainer:
     # ------------------------------------------------------------------
     def _make_loader(self, train: bool, epoch: int) -> BatchIterator:
         if train:
-            data = self.builder.refresh_train_dataset(epoch)
+            if self._next_train is not None and self._next_train[0] == epoch:
+                data = self._next_train[1].result()
+            else:
+                data = self.builder.refresh_train_dataset(epoch)
+            self._next_train = None
         else:
             data = self.builder.refresh_test_dataset(epoch)
         return BatchIterator(
@@ -231,6 +241,14 @@ class Trainer:
 
     def _train_epoch(self, epoch: int):
         loader = self._make_loader(train=True, epoch=epoch)
+        # kick off next epoch's rebuild in the background.  NOTE: the
+        # builder's pinned pool is double-buffered per (tag, parity) so the
+        # background build never overwrites the epoch currently training.
+        self._next_train = (
+            epoch + 1,
+            self._build_pool.submit(self.builder.refresh_train_dataset,
+                                    epoch + 1),
+        )
         model = self.model
         model.train()
         total_loss = torch.zeros((), dtype=torch.float64, device=self.device)
