@@ -23,11 +23,10 @@ _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 # fused gather+combiner forward (C2V_NO_FUSE=1 falls back to two kernels)
 FUSE_GATHER_COMBINER = os.environ.get("C2V_NO_FUSE") != "1"
 
-# Early-gradient callbacks (parallel/ddp.py registers one per big parameter,
-# keyed by param.data_ptr()): the embedding backward invokes them the moment
-# a table's grad tensor is complete, so at DP>1 the terminal-table
-# all-reduce overlaps with the path table's scatter pipeline instead of
-# waiting for the whole backward to finish.
+# Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
+# backward invokes them the moment a table's grad tensor is complete.
+# parallel/ddp.py no longer registers any (see the note in _on_grad_ready);
+# the hook point remains for schemes that own their buffers.
 EARLY_GRAD_CALLBACKS = {}
 
 from . import ext, round_up

@@ -67,9 +67,6 @@ class BucketedAllReduce:
         ]
         self._direct_set = set(id(p) for p in self.direct_params)
         self._direct_handles = []
-        self._early_reduced = set()  # data_ptrs all-reduced via callbacks
-        if self.enabled:
-            self._register_early_callbacks()
 
         # reverse order ~ autograd completion order; one dtype per bucket
         self.buckets: List[Bucket] = []
@@ -105,31 +102,19 @@ class BucketedAllReduce:
                 self._hooks.append(h)
         self._reset_pending()
 
-    def _register_early_callbacks(self):
-        """Let the embedding backward start a big tensor's all-reduce the
-        moment that grad is complete (ops/functional.EARLY_GRAD_CALLBACKS),
-        overlapping comm with the rest of backward."""
-        from ..ops import functional as Fn
-
-        for p in self.direct_params:
-            def cb(grad, _param=p, _self=self):
-                h = dist.all_reduce(grad, op=dist.ReduceOp.SUM,
-                                    group=_self.group, async_op=True)
-                _self._direct_handles.append((h, _param, grad.data_ptr()))
-                _self._early_reduced.add(grad.data_ptr())
-            Fn.EARLY_GRAD_CALLBACKS[p.data_ptr()] = cb
-
     def _reset_pending(self):
         for b in self.buckets:
             b.pending = len(b.params)
             b.handle = None
         self._direct_handles = []
-        self._early_reduced = set()
 
     def _on_grad_ready(self, param) -> None:
         if id(param) in self._direct_set:
-            if param.grad.data_ptr() in self._early_reduced:
-                return  # already launched from inside the backward
+            # NOTE: an "early reduce" of the grad tensor from inside the
+            # embedding backward was tried and REVERTED: autograd does not
+            # reliably adopt the returned tensor as p.grad (it may clone),
+            # so an in-place reduction on the produced tensor can be lost —
+            # caught by tests/test_ddp_gpu_gloo.py.  Reduce at hook time.
             h = dist.all_reduce(
                 param.grad, op=dist.ReduceOp.SUM, group=self.group,
                 async_op=True,
@@ -150,12 +135,9 @@ class BucketedAllReduce:
         for h, p, gptr in self._direct_handles:
             h.wait()
             if p.grad is None or p.grad.data_ptr() != gptr:
-                # autograd cloned instead of stealing the grad tensor; the
-                # reduction landed in the wrong buffer — fail loudly rather
-                # than silently train on unreduced gradients
                 raise RuntimeError(
-                    "early-reduced gradient was not adopted as p.grad "
-                    "(autograd cloned it); disable early callbacks"
+                    "direct-reduced gradient tensor changed between the "
+                    "hook and finish() — refusing to average the wrong buffer"
                 )
             p.grad.div_(self.world_size)
         for b in self.buckets:

@@ -191,18 +191,19 @@ def _worker_early_cb(rank, world, port, result_file):
 
     p = torch.nn.Parameter(torch.zeros(4096))
     ddp = BucketedAllReduce([p], world, direct_threshold=1024)
-    assert p.data_ptr() in Fn.EARLY_GRAD_CALLBACKS
+    # early callbacks were removed (see ddp._on_grad_ready note): no
+    # registration must happen
+    assert p.data_ptr() not in Fn.EARLY_GRAD_CALLBACKS
 
-    # step 1: grad arrives via the early callback during "backward"
+    # step 1: hook-path direct reduce
     g = torch.full((4096,), float(rank + 1))
-    Fn.EARLY_GRAD_CALLBACKS[p.data_ptr()](g)
-    p.grad = g  # autograd adopts the same tensor
-    ddp._on_grad_ready(p)  # the hook must NOT reduce again
+    p.grad = g
+    ddp._on_grad_ready(p)
     ddp.finish()
     expected = (sum(range(1, world + 1)) / world)
     assert torch.allclose(p.grad, torch.full((4096,), expected)), p.grad[0]
 
-    # step 2: no callback (hook path) still works after reset
+    # step 2: second cycle after reset
     ddp.zero_grad()
     g2 = torch.full((4096,), float(10 * (rank + 1)))
     p.grad = g2
