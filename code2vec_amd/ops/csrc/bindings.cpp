@@ -56,6 +56,8 @@ void launch_wgrad(const void*, const void*, float*, long, int, int, int,
                   hipStream_t);
 void launch_dgrad(const void*, const void*, void*, long, int, int,
                   hipStream_t);
+void launch_head_wgrad(const void*, const void*, void*, long, long, int,
+                       hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
@@ -295,6 +297,18 @@ void dgrad(torch::Tensor dz, torch::Tensor w2, torch::Tensor dx) {
                cur_stream());
 }
 
+void head_wgrad(torch::Tensor dlogits, torch::Tensor cv, torch::Tensor dw) {
+  CHK_CUDA(dlogits); CHK_CONTIG(dlogits); CHK_DT(dlogits, torch::kBFloat16);
+  CHK_CONTIG(cv); CHK_DT(cv, torch::kBFloat16);
+  CHK_CONTIG(dw); CHK_DT(dw, torch::kBFloat16);
+  const long B = dlogits.size(0), L = dlogits.size(1);
+  const int EP = cv.size(1);
+  TORCH_CHECK(dw.size(0) == L && dw.size(1) == EP, "dw shape");
+  TORCH_CHECK(B % 32 == 0 && L % 8 == 0, "head_wgrad shape gates");
+  launch_head_wgrad(dlogits.data_ptr(), cv.data_ptr(), dw.data_ptr(), L, B,
+                    EP, cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -351,6 +365,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
   m.def("wgrad", &wgrad);
   m.def("colsum_bf16", &colsum_bf16);
+  m.def("head_wgrad", &head_wgrad);
   m.def("dgrad", &dgrad);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);

@@ -23,7 +23,7 @@ extern "C" {
 
 void launch_colsum_bf16(const void* x, float* out, long B, long L,
                         hipStream_t stream) {
-  const int rows_per_block = 128;
+  const int rows_per_block = 32;  // wider grid: latency-bound otherwise
   const int lblocks = (int)((L + 255) / 256);
   const int rblocks = (int)((B + rows_per_block - 1) / rows_per_block);
   colsum_bf16_kernel<<<lblocks * rblocks, 256, 0, stream>>>(

@@ -341,8 +341,18 @@ class OutputHead(torch.autograd.Function):
     def backward(ctx, dlogits):
         cv, w = ctx.saved_tensors
         dlogits = dlogits.contiguous()
-        dcv = dlogits @ w                      # [B, EP] bf16
-        dw = dlogits.t() @ cv                  # [L, EP] bf16
+        dcv = dlogits @ w                      # [B, EP] bf16 (rocBLAS)
+        B, L = dlogits.shape
+        EP = cv.shape[1]
+        # default OFF: measured slower than TunableOp rocBLAS at the
+        # top11 shape (single-buffered staging; see PERF.md)
+        if (os.environ.get("C2V_HEAD_WGRAD", "0") == "1"
+                and B % 32 == 0 and L % 8 == 0 and EP % 32 == 0
+                and EP <= 128):
+            dw = torch.empty(L, EP, dtype=torch.bfloat16, device=w.device)
+            ext().head_wgrad(dlogits, cv, dw)
+        else:
+            dw = dlogits.t() @ cv              # [L, EP] bf16 (rocBLAS)
         dbias = torch.zeros(w.shape[0], dtype=torch.float32,
                             device=w.device)
         ext().colsum_bf16(dlogits, dbias)
