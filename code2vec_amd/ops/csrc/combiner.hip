@@ -37,7 +37,7 @@
 // tables directly (segment-resolved), never materializing the concat
 // tensor.  Segment boundaries are 8-element multiples, so a 16-B fragment
 // piece never straddles one.
-template <int NT, int EPI, int GATHER>
+template <int NT, int EPI, int GATHER, int MT = 2>  // MT: 16-row M tiles per wave
 __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ Wt,
     const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -51,7 +51,7 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   const int EP = NT * 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const long row0 = (long)blockIdx.x * 128;
+  const long row0 = (long)blockIdx.x * (MT * 64);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* lds_gamma = (float*)smem;
@@ -63,23 +63,23 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   __syncthreads();
 
   const int NK = KP / 32;
-  f32x4 acc[2][NT];
+  f32x4 acc[MT][NT];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int mi = 0; mi < MT; ++mi)
 #pragma unroll
     for (int n = 0; n < NT; ++n) acc[mi][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  long arow[2];
+  long arow[MT];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi) {
-    long r = row0 + wave * 32 + mi * 16 + (lane & 15);
+  for (int mi = 0; mi < MT; ++mi) {
+    long r = row0 + wave * (MT * 16) + mi * 16 + (lane & 15);
     arow[mi] = r < M ? r : (M - 1);
   }
   // fused-gather mode: this lane's three embedding rows, loaded once
-  const bf16* seg_base[2][3];
+  const bf16* seg_base[MT][3];
   if (GATHER) {
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+    for (int mi = 0; mi < MT; ++mi) {
       seg_base[mi][0] = g_term + (size_t)g_starts[arow[mi]] * TS;
       seg_base[mi][1] = g_path + (size_t)g_paths[arow[mi]] * PS;
       seg_base[mi][2] = g_term + (size_t)g_ends[arow[mi]] * TS;
@@ -119,10 +119,10 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   for (int kk = 0; kk < NK; ++kk) {
     const int buf = kk & 1;
     if (kk + 1 < NK) STAGE(kk + 1, buf ^ 1)
-    bf16x8 a[2];
+    bf16x8 a[MT];
     const int kfrag = kk * 32 + kj;
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+    for (int mi = 0; mi < MT; ++mi) {
       if (GATHER) {
         bf16x8 v = {};
         if (kfrag < TS)
@@ -145,8 +145,9 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     for (int n = 0; n < NT; ++n) {
       if (n + 1 < NT)
         b_nxt = *(const bf16x8*)(bbase + (size_t)(n + 1) * 16 * ESTRIDE);
-      acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], b_cur, acc[0][n], 0, 0, 0);
-      acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], b_cur, acc[1][n], 0, 0, 0);
+#pragma unroll
+      for (int mi = 0; mi < MT; ++mi)
+        acc[mi][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b_cur, acc[mi][n], 0, 0, 0);
       b_cur = b_nxt;
     }
     __syncthreads();
@@ -161,14 +162,14 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   bf16* lds_t = nullptr;
   if (EPI == 1) {
     lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) + 2 * (size_t)EP * 40 +
-            (size_t)wave * 32 * (EP + 8);
+            (size_t)wave * (MT * 16) * (EP + 8);
   }
-  float mean_r[2][4], rstd_r[2][4];
+  float mean_r[MT][4], rstd_r[MT][4];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi) {
+  for (int mi = 0; mi < MT; ++mi) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const long row = row0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+      const long row = row0 + wave * (MT * 16) + mi * 16 + (lane >> 4) * 4 + r;
       float s1 = 0.f, s2 = 0.f;
 #pragma unroll
       for (int n = 0; n < NT; ++n) {
@@ -196,11 +197,11 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   for (int pass = 0; pass < 2; ++pass) {
     bf16* dst_base = pass == 0 ? z_save : out;
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+    for (int mi = 0; mi < MT; ++mi) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row_l = mi * 16 + (lane >> 4) * 4 + r;
-        const long row = row0 + wave * 32 + row_l;
+        const long row = row0 + wave * (MT * 16) + row_l;
         const float mean = mean_r[mi][r];
         const float rstd = rstd_r[mi][r];
         if (EPI == 0 && row >= M) continue;
@@ -233,12 +234,12 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
       }
     }
     if (EPI == 1) {
-      // flush the wave's [32, EP] tile as 16-B chunks (coalesced)
-      const int chunks = 32 * EP / 8;
+      // flush the wave's [MT*16, EP] tile as 16-B chunks (coalesced)
+      const int chunks = MT * 16 * EP / 8;
       for (int c = lane; c < chunks; c += WAVE) {
         const int row_l = c / (EP / 8);
         const int col8 = (c % (EP / 8)) * 8;
-        const long row = row0 + wave * 32 + row_l;
+        const long row = row0 + wave * (MT * 16) + row_l;
         if (row < M) {
           const uint4 v =
               *(const uint4*)(lds_t + (size_t)row_l * (EP + 8) + col8);
@@ -389,20 +390,27 @@ void launch_combiner_fwd_impl(const void* X, const void* W,
                          const void* path, int TS, int PS,
                          hipStream_t stream) {
   const int NT = EP / 16;
-  const long grid = (M + 127) / 128;
+  const char* mt_env = getenv("C2V_COMBINER_MT");
+  const int mt = (mt_env && mt_env[0] == '1') ? 1 : 2;
+  const long grid = (M + mt * 64 - 1) / (mt * 64);
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
   int epi = epilogue_mode;
   (void)gather;
   int smem = 2 * EP * sizeof(float) + 2 * EP * 40 * (int)sizeof(bf16);
   if (epi == 1) {
-    const int bounce = 4 * 32 * (EP + 8) * (int)sizeof(bf16);
+    const int bounce = 4 * mt * 16 * (EP + 8) * (int)sizeof(bf16);
     if (smem + bounce <= 160 * 1024 - 2048) smem += bounce;
     else epi = 0;
   }
 #define CASE(nt)                                                              \
   case nt:                                                                    \
-    if (gather)                                                               \
-      combiner_fwd_kernel<nt, 1, 1><<<grid, 256, smem, stream>>>(             \
+    if (gather && mt == 1)                                                    \
+      combiner_fwd_kernel<nt, 1, 1, 1><<<grid, 256, smem, stream>>>(          \
+          nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
+          mean, rstd, M, KP, E, p, inv1mp, seed, offset, starts, paths,       \
+          ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
+    else if (gather)                                                          \
+      combiner_fwd_kernel<nt, 1, 1, 2><<<grid, 256, smem, stream>>>(          \
           nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
           mean, rstd, M, KP, E, p, inv1mp, seed, offset, starts, paths,       \
           ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
