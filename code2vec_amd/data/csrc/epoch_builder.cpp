@@ -102,10 +102,156 @@ void build_method_epoch(torch::Tensor offsets, torch::Tensor contexts,
   }
 }
 
-// Corpus line-oriented fast path parse of the "paths:" triples section is in
-// Python (reader.py); numeric-heavy per-epoch work lives here.
+// ---------------------------------------------------------------------------
+// Native corpus parse (reference model/dataset_reader.py:72-128 semantics).
+// Single C++ pass over the file (GIL released); label normalization and
+// vocab building stay in Python (regex parity).  Start/end terminal
+// indexes are returned +1-shifted (the @question offset); paths unshifted.
+
+#include <fstream>
+#include <string>
+
+namespace {
+struct Span { size_t off; size_t len; };
+
+inline const char* strip(const char* b, const char* e, size_t* len) {
+  while (b < e && (*b == ' ' || *b == '\r' || *b == '\t')) ++b;
+  while (e > b && (e[-1] == ' ' || e[-1] == '\r' || e[-1] == '\t'))
+    --e;
+  *len = (size_t)(e - b);
+  return b;
+}
+
+inline long parse_int(const char* p, const char* e, const char** out) {
+  long v = 0;
+  bool neg = false;
+  if (p < e && *p == '-') { neg = true; ++p; }
+  while (p < e && *p >= '0' && *p <= '9') v = v * 10 + (*p++ - '0');
+  *out = p;
+  return neg ? -v : v;
+}
+}  // namespace
+
+pybind11::dict parse_corpus(const std::string& path) {
+  std::string buf;
+  std::vector<int64_t> ids;
+  std::vector<Span> labels, sources;
+  std::vector<int32_t> triples;
+  std::vector<int64_t> rec_off;          // per-record triple offsets
+  std::vector<int64_t> var_rec_off;      // per-record var offsets
+  std::vector<Span> var_orig, var_alias;
+  {
+    pybind11::gil_scoped_release release;
+    std::ifstream f(path, std::ios::binary);
+    if (!f) throw std::runtime_error("cannot open corpus: " + path);
+    f.seekg(0, std::ios::end);
+    buf.resize((size_t)f.tellg());
+    f.seekg(0);
+    f.read(&buf[0], (std::streamsize)buf.size());
+
+    bool in_record = false;
+    int mode = 0;
+    auto flush = [&]() {
+      if (in_record) {
+        rec_off.push_back((int64_t)(triples.size() / 3));
+        var_rec_off.push_back((int64_t)var_orig.size());
+        in_record = false;
+      }
+    };
+    const char* base = buf.data();
+    const char* end = base + buf.size();
+    const char* line = base;
+    while (line <= end) {
+      const char* nl = (const char*)memchr(line, '\n', (size_t)(end - line));
+      const char* le = nl ? nl : end;
+      size_t len;
+      const char* lb = strip(line, le, &len);
+      if (len == 0) {
+        flush();
+      } else {
+        if (!in_record) {
+          in_record = true;
+          mode = 0;
+          ids.push_back(-1);
+          labels.push_back({0, 0});
+          sources.push_back({0, 0});
+        }
+        if (lb[0] == '#') {
+          const char* q;
+          ids.back() = parse_int(lb + 1, lb + len, &q);
+        } else if (len >= 6 && memcmp(lb, "label:", 6) == 0) {
+          labels.back() = {(size_t)(lb + 6 - base), len - 6};
+        } else if (len >= 6 && memcmp(lb, "class:", 6) == 0) {
+          sources.back() = {(size_t)(lb + 6 - base), len - 6};
+        } else if (len >= 6 && memcmp(lb, "paths:", 6) == 0) {
+          mode = 1;
+        } else if (len >= 5 && memcmp(lb, "vars:", 5) == 0) {
+          mode = 2;
+        } else if (len >= 4 && memcmp(lb, "doc:", 4) == 0) {
+          // parsed and discarded
+        } else if (mode == 1) {
+          const char* q = lb;
+          long s = parse_int(q, lb + len, &q);
+          if (q < lb + len && *q == '\t') ++q;
+          long pth = parse_int(q, lb + len, &q);
+          if (q < lb + len && *q == '\t') ++q;
+          long e2 = parse_int(q, lb + len, &q);
+          triples.push_back((int32_t)(s + 1));  // +@question shift
+          triples.push_back((int32_t)pth);
+          triples.push_back((int32_t)(e2 + 1));
+        } else if (mode == 2) {
+          const char* tab =
+              (const char*)memchr(lb, '\t', len);
+          if (tab) {
+            var_orig.push_back({(size_t)(lb - base), (size_t)(tab - lb)});
+            var_alias.push_back({(size_t)(tab + 1 - base),
+                                 len - (size_t)(tab + 1 - lb)});
+          }
+        }
+      }
+      if (!nl) break;
+      line = nl + 1;
+    }
+    flush();
+  }
+
+  // with the GIL: assemble python objects
+  namespace py = pybind11;
+  const int64_t n = (int64_t)ids.size();
+  auto ids_t = torch::from_blob(ids.data(), {n}, torch::kInt64).clone();
+  std::vector<int64_t> off(n + 1, 0);
+  for (int64_t i = 0; i < n; ++i) off[i + 1] = rec_off[(size_t)i];
+  auto off_t = torch::from_blob(off.data(), {n + 1}, torch::kInt64).clone();
+  std::vector<int64_t> voff(n + 1, 0);
+  for (int64_t i = 0; i < n; ++i) voff[i + 1] = var_rec_off[(size_t)i];
+  auto voff_t = torch::from_blob(voff.data(), {n + 1}, torch::kInt64).clone();
+  auto tri_t = torch::from_blob(triples.data(),
+                                {(int64_t)(triples.size() / 3), 3},
+                                torch::kInt32).clone();
+  py::list label_l, source_l, vorig_l, valias_l;
+  for (auto& sp : labels)
+    label_l.append(py::str(buf.data() + sp.off, sp.len));
+  for (auto& sp : sources)
+    source_l.append(py::str(buf.data() + sp.off, sp.len));
+  for (auto& sp : var_orig)
+    vorig_l.append(py::str(buf.data() + sp.off, sp.len));
+  for (auto& sp : var_alias)
+    valias_l.append(py::str(buf.data() + sp.off, sp.len));
+
+  py::dict out;
+  out["ids"] = ids_t;
+  out["labels"] = label_l;
+  out["sources"] = source_l;
+  out["offsets"] = off_t;
+  out["contexts"] = tri_t;
+  out["var_offsets"] = voff_t;
+  out["var_originals"] = vorig_l;
+  out["var_aliases"] = valias_l;
+  return out;
+}
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("build_method_epoch", &build_method_epoch,
         "per-epoch resample+pad (OpenMP)");
+  m.def("parse_corpus", &parse_corpus, "native corpus parse");
 }

@@ -30,6 +30,12 @@ from .vocab import (
 
 logger = logging.getLogger(__name__)
 
+try:  # native C++ corpus parser (GIL-released single pass)
+    import torch as _torch  # noqa: F401  (loads libc10 for the extension)
+    from . import _c2v_host as _native
+except Exception:  # noqa: BLE001
+    _native = None
+
 
 @dataclass
 class CodeItem:
@@ -89,10 +95,60 @@ class CorpusReader:
 
         self.label_vocab = Vocab()
         self.items: List[CodeItem] = []
-        self._load(corpus_path)
+        if _native is not None:
+            self._load_native(corpus_path)
+        else:
+            self._load(corpus_path)
 
         logger.info("label vocab size: %d", len(self.label_vocab))
         logger.info("corpus: %d", len(self.items))
+
+    def _load_native(self, corpus_path: str) -> None:
+        """Native parse (data/csrc/epoch_builder.cpp::parse_corpus) +
+        Python-side label normalization/vocab build (regex parity with the
+        reference); path_contexts become zero-copy views of one flat
+        array."""
+        res = _native.parse_corpus(corpus_path)
+        ids = res["ids"].numpy()
+        offsets = res["offsets"].numpy()
+        voffsets = res["var_offsets"].numpy()
+        contexts = res["contexts"].numpy()
+        labels = res["labels"]
+        sources = res["sources"]
+        vorig = res["var_originals"]
+        valias = res["var_aliases"]
+        label_vocab = self.label_vocab
+        infer_method = self.infer_method
+        infer_variable = self.infer_variable
+        norm_cache = {}
+
+        def normalized(raw: str):
+            hit = norm_cache.get(raw)
+            if hit is None:
+                n = normalize_method_name(raw)
+                hit = (n.lower(), get_method_subtokens(n))
+                norm_cache[raw] = hit
+            return hit
+
+        for i in range(len(ids)):
+            item = CodeItem()
+            item.id = int(ids[i]) if ids[i] >= 0 else None
+            raw = labels[i]
+            if raw:
+                item.label = raw
+                lower, subtokens = normalized(raw)
+                item.normalized_label = lower
+                if infer_method:
+                    label_vocab.append(lower, subtokens=subtokens)
+            item.source = sources[i] or None
+            item.path_contexts = contexts[offsets[i]:offsets[i + 1]]
+            for v in range(voffsets[i], voffsets[i + 1]):
+                alias_name = valias[v]
+                lower, subtokens = normalized(vorig[v])
+                item.aliases[alias_name] = lower
+                if infer_variable and alias_name.startswith("@var_"):
+                    label_vocab.append(lower, subtokens=subtokens)
+            self.items.append(item)
 
     def _load(self, corpus_path: str) -> None:
         items = self.items

@@ -178,3 +178,75 @@ def test_native_builder_matches_python_semantics(tiny_corpus):
                            native.ends[row, :n]], axis=1).tolist()]
         if len(allowed) == pcs.shape[0]:
             assert len(set(trips)) == len(trips)
+
+
+def test_reader_edge_cases(tmp_path):
+    """Parser robustness: empty paths section, no-vars method, missing id,
+    no trailing blank line, doc lines, blank-line runs."""
+    corpus = tmp_path / "corpus.txt"
+    corpus.write_text(
+        "#1\n"
+        "label:emptyPaths\n"
+        "class:A.java\n"
+        "paths:\n"
+        "\n"
+        "\n"
+        "label:noIdNoVars\n"
+        "doc: some doc text\n"
+        "paths:\n"
+        "1\t1\t2\n"
+        "2\t2\t1\n",  # no trailing blank line
+        encoding="utf-8",
+    )
+    terms = tmp_path / "terms.txt"
+    terms.write_text("0\t<PAD/>\n1\t@method_0\n2\tfoo\n3\tbar\n",
+                     encoding="utf-8")
+    pathsf = tmp_path / "paths.txt"
+    pathsf.write_text("0\t<PAD/>\n1\tP1\n2\tP2\n", encoding="utf-8")
+
+    r = CorpusReader(str(corpus), str(pathsf), str(terms))
+    assert len(r.items) == 2
+    assert r.items[0].path_contexts.shape == (0, 3)  # empty bag
+    assert r.items[1].id is None
+    assert r.items[1].path_contexts.shape == (2, 3)
+    # +1 question shift on start/end, not on path
+    assert r.items[1].path_contexts[0].tolist() == [2, 1, 3]
+
+    # a zero-context method builds an all-pad row; model handles it
+    opt = make_option(r, max_path_length=4)
+    b = DatasetBuilder(r, opt, seed=1, split_ratio=0.0)
+    data = b.refresh_train_dataset(0)
+    assert len(data) == 2
+    row_empty = data.ids.index(1)
+    assert (data.starts[row_empty] == 0).all()
+
+
+def test_native_parser_matches_python(tiny_corpus):
+    """C++ parse_corpus produces identical records/vocabs to the Python
+    parser (including +1 question shift, aliases, doc-skip)."""
+    from code2vec_amd.data import reader as RD
+
+    if RD._native is None:
+        pytest.skip("native parser not built")
+    r_nat = make_reader(tiny_corpus)
+    # force the python path
+    r_py = CorpusReader.__new__(CorpusReader)
+    r_py.path_vocab = r_nat.path_vocab
+    r_py.terminal_vocab = r_nat.terminal_vocab
+    r_py.infer_method = True
+    r_py.infer_variable = False
+    from code2vec_amd.data.vocab import Vocab
+    r_py.label_vocab = Vocab()
+    r_py.items = []
+    r_py._load(tiny_corpus["corpus_path"])
+
+    assert len(r_nat.items) == len(r_py.items)
+    assert r_nat.label_vocab.stoi == r_py.label_vocab.stoi
+    assert r_nat.label_vocab.itosubtokens == r_py.label_vocab.itosubtokens
+    for a, b in zip(r_nat.items, r_py.items):
+        assert a.id == b.id
+        assert a.label == b.label
+        assert a.normalized_label == b.normalized_label
+        assert a.source == b.source
+        assert a.aliases == b.aliases
+        assert np.array_equal(a.path_contexts, b.path_contexts)
