@@ -44,6 +44,8 @@ def _try_load():
     if _ext is not None or _ext_err is not None:
         return
     try:
+        import torch  # noqa: F401  (the extension links libtorch/libc10)
+
         _ext = importlib.import_module("code2vec_amd.ops._c2v_hip")
     except Exception as e:  # noqa: BLE001
         _ext_err = e
