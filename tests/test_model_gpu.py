@@ -276,3 +276,15 @@ def test_attention_grad_through_attn_output(dev):
     torch.autograd.backward([cv_r, attn_r], [dcv, dattn])
     assert relerr(ccv_h.grad.float(), ccv_r.grad) < 6e-2
     assert relerr(a_h.grad[:E], a_r.grad[:E]) < 6e-2
+
+
+def test_torch_backend_on_gpu(dev):
+    """--backend torch runs the fp32 reference math on the GPU (oracle mode
+    on device); one step must work and produce finite grads."""
+    opt = make_option()
+    model = Code2VecTorch(opt).to(dev).train()
+    s, p, e, y = make_inputs(opt, 8, dev, seed=1)
+    out, _, _ = model(s.long(), p.long(), e.long(), y)
+    loss = model.loss(out, y, torch.ones(opt.label_count, device=dev))
+    loss.backward()
+    assert torch.isfinite(model.terminal_embedding.grad).all()

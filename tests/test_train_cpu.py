@@ -200,3 +200,39 @@ def test_print_sample_runs(tmp_path, tiny_corpus, caplog):
     # either no correct prediction (silent) or a full sample dump
     if caplog.records:
         assert any("label" in r.getMessage() for r in caplog.records)
+
+
+def test_seed_determinism_end_to_end(tiny_corpus):
+    """Same seed => identical epoch-0 loss (the reference's unseeded
+    run-to-run variance is a design delta we intentionally remove)."""
+    def one_loss():
+        reader = CorpusReader(
+            tiny_corpus["corpus_path"], tiny_corpus["path_idx_path"],
+            tiny_corpus["terminal_idx_path"],
+        )
+        opt = Option(
+            terminal_count=len(reader.terminal_vocab),
+            path_count=len(reader.path_vocab),
+            label_count=len(reader.label_vocab),
+            max_path_length=12, terminal_embed_size=12, path_embed_size=12,
+            encode_size=16, dropout_prob=0.0, batch_size=16,
+            device=torch.device("cpu"),
+        )
+        builder = DatasetBuilder(reader, opt, seed=21)
+        g = torch.Generator().manual_seed(21)
+        model = build_model(opt, backend="torch",
+                            logical=init_logical_params(opt, g))
+        ctx = DistContext(0, 1, 0, torch.device("cpu"))
+
+        class A:
+            max_epoch = 1; lr = 0.01; beta_min = 0.9; beta_max = 0.999
+            weight_decay = 0.0; model_path = "/tmp/c2v_det"
+            vectors_path = "/tmp/c2v_det/code.vec"; test_result_path = None
+            env = None; print_sample_cycle = 0; eval_method = "subtoken"
+            random_seed = 21; batch_size = 16
+
+        t = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
+        loss, _ = t._train_epoch(0)
+        return loss
+
+    assert one_loss() == one_loss()
