@@ -8,9 +8,11 @@
 // mfma_f32_16x16x32_bf16 tiles, fp32 accumulation, LDS-staged B operand,
 // in-register row reductions for the LN statistics.
 //
-// Layouts: X [M, KP] bf16 row-major; W [KP, EP] bf16 row-major (B-operand);
-// z/out [M, EP] bf16; gamma/beta [EP] fp32 (pad cols zero); valid cols = E.
-// Pad cols of W are zero => z pad cols are exactly zero => LN sums over all
+// Layouts: X [M, KP] bf16 row-major (or gathered on the fly from the
+// embedding tables in the fused GATHER=1 variant); W is stored TRANSPOSED
+// [EP, KP] bf16 so the MFMA B fragment is memory-contiguous; z/out [M, EP]
+// bf16; gamma/beta [EP] fp32 (pad rows/cols zero); valid cols = E.
+// Pad rows of W are zero => z pad cols are exactly zero => LN sums over all
 // EP cols equal sums over the E valid cols.
 
 #include "common.h"

@@ -3,9 +3,15 @@
 // nn.Embedding for backward — reimagined as HBM-streaming CDNA4 kernels.)
 //
 // Layout: out[row, :] = [ term[starts[row]] | path[paths[row]] | term[ends[row]] ]
-// with segment strides TS / PS / TS (each a multiple of 32 bf16 = 64 B), so
-// every lane moves aligned 16-byte chunks.  Pad columns of the tables are
-// zero, so pad columns of the output are zero by construction.
+// with segment strides TS / PS / TS (8-element multiples = 16-B granules;
+// dt=100 stores 104-wide) and a zero-filled tail up to KP (the combiner's
+// 32-element K padding).  Pad columns of the tables are zero, so pad
+// columns of the output are zero by construction.
+//
+// NOTE: the standalone forward gather and the atomic backward below are the
+// reference/fallback path — the default forward fuses the gather into the
+// combiner GEMM (combiner.hip GATHER=1) and the default backward is the
+// sort-based scatter further down.
 
 #include "common.h"
 
@@ -43,9 +49,10 @@ __global__ void gather_concat_fwd_kernel(
   }
 }
 
-// Backward: scatter-add grad rows into fp32 dense grad tables.
-// Pad contexts (starts == 0) carry exactly-zero grads (attention mask math)
-// and are skipped entirely.  fp32 accumulation via native atomics.
+// Legacy backward (atomic-rate-bound; superseded by the sort-based scatter
+// below — kept as the simple reference implementation for A/B): scatter-add
+// grad rows into fp32 dense grad tables.  Pad contexts (starts == 0) carry
+// exactly-zero grads (attention mask math) and are skipped entirely.
 __global__ void gather_concat_bwd_kernel(
     const int* __restrict__ starts, const int* __restrict__ paths,
     const int* __restrict__ ends, const bf16* __restrict__ gout,
