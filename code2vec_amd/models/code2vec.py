@@ -190,13 +190,22 @@ class Code2VecHIP(nn.Module):
         starts = starts.to(torch.int32)
         paths = paths.to(torch.int32)
         ends = ends.to(torch.int32)
-        x = Fn.GatherConcat.apply(
-            starts, paths, ends, self.terminal_embedding, self.path_embedding
-        )
-        y = Fn.CombinerLNTanh.apply(
-            x, self.input_weight, self.ln_gamma, self.ln_beta,
-            self.E, self.dropout_p, self.training,
-        )
+        if Fn.FUSE_GATHER_COMBINER:
+            # fused K1-K6: the concat tensor is never materialized
+            y = Fn.FusedGatherCombiner.apply(
+                starts, paths, ends, self.terminal_embedding,
+                self.path_embedding, self.input_weight, self.ln_gamma,
+                self.ln_beta, self.E, self.dropout_p, self.training,
+            )
+        else:
+            x = Fn.GatherConcat.apply(
+                starts, paths, ends, self.terminal_embedding,
+                self.path_embedding
+            )
+            y = Fn.CombinerLNTanh.apply(
+                x, self.input_weight, self.ln_gamma, self.ln_beta,
+                self.E, self.dropout_p, self.training,
+            )
         ccv = y.view(B, C, self.EP)
         cv, attn = Fn.AttentionPool.apply(ccv, self.attention_a, starts, self.E)
         if opt.angular_margin_loss:

@@ -20,8 +20,14 @@ import torch
 # 1 = per-wave LDS bounce + coalesced 16-B flush (kbench-selected)
 _FWD_EPI = int(os.environ.get("C2V_FWD_EPI", "1"))
 _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
-# fused gather+combiner forward (C2V_NO_FUSE=1 falls back to two kernels)
-FUSE_GATHER_COMBINER = os.environ.get("C2V_NO_FUSE") != "1"
+# Fused gather+combiner forward, opt-in via C2V_FUSE=1.  Measured OFF-better
+# on top11 (1.415 vs 1.478 ms/step): fusing saves the 61 us gather kernel but
+# costs +16 us in combiner_fwd (scattered row loads inside the MFMA loop) and
+# +109 us in wgrad (GATHER=1 re-gathers X every split-K pass), net +64 us.
+# The split gather kernel streams at HBM BW, so there is no bandwidth win to
+# recover.  Kept (correct, GPU-tested) for small-table configs where the
+# gather output (B*C*KP bf16) no longer fits comfortably.
+FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
 # backward invokes them the moment a table's grad tensor is complete.

@@ -87,6 +87,39 @@ def test_loss_and_grads_match_oracle(dev):
     assert relerr(hip.output_weight.grad[:, :E], ref.output_weight.grad) < 6e-2
 
 
+def test_fused_gather_combiner_matches_unfused(dev, monkeypatch):
+    """C2V_FUSE=1 path (gather fused into the combiner GEMM + wgrad) must be
+    bit-for-bit equivalent in eval forward and match the unfused backward
+    closely (both are bf16 MFMA; only summation order differs)."""
+    import code2vec_amd.ops.functional as Fn
+
+    opt = make_option()
+    g = torch.Generator().manual_seed(7)
+    logical = init_logical_params(opt, g)
+    s, p, e, y = make_inputs(opt, 24, dev, seed=5)
+    w = torch.ones(opt.label_count, device=dev)
+
+    results = []
+    for fuse in (False, True):
+        monkeypatch.setattr(Fn, "FUSE_GATHER_COMBINER", fuse)
+        m = Code2VecHIP(opt, logical, device=dev).train()
+        out, _, _ = m(s, p, e, y)
+        loss = m.loss(out, y, w)
+        loss.backward()
+        results.append((
+            out.detach(), float(loss), m.terminal_embedding.grad.clone(),
+            m.path_embedding.grad.clone(), m.input_weight.grad.clone(),
+            m.ln_gamma.grad.clone(),
+        ))
+    (o0, l0, tg0, pg0, wg0, gg0), (o1, l1, tg1, pg1, wg1, gg1) = results
+    assert torch.equal(o0, o1)  # same kernel math for the activations
+    assert abs(l0 - l1) / abs(l0) < 1e-3
+    assert relerr(tg1, tg0) < 2e-2
+    assert relerr(pg1, pg0) < 2e-2
+    assert relerr(wg1, wg0) < 2e-2
+    assert relerr(gg1, gg0) < 2e-2
+
+
 def test_pad_regions_stay_zero_after_steps(dev):
     opt = make_option(dropout_prob=0.25)
     hip = Code2VecHIP(opt, device=dev).train()
