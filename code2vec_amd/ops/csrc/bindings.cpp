@@ -59,6 +59,11 @@ void launch_dgrad(const void*, const void*, void*, long, int, int,
 void launch_head_wgrad(const void*, const void*, void*, long, long, int,
                        hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
+void launch_head_fwd(const void*, const void*, const float*, void*, float*,
+                     float*, long, long, int, hipStream_t);
+void launch_lsm_finalize(const void*, const float*, const float*, const long*,
+                         const float*, float*, float*, int, long, int,
+                         hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
 void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
@@ -309,6 +314,53 @@ void head_wgrad(torch::Tensor dlogits, torch::Tensor cv, torch::Tensor dw) {
                     EP, cur_stream());
 }
 
+// logits = cv @ w^T + bias; pm/ps (optional, pass undefined tensors to
+// skip) receive per-(row, 64-col-block) online-softmax partials.
+void head_fwd(torch::Tensor cv, torch::Tensor w, torch::Tensor bias,
+              torch::Tensor out, torch::Tensor pm, torch::Tensor ps) {
+  CHK_CUDA(cv); CHK_CONTIG(cv); CHK_DT(cv, torch::kBFloat16);
+  CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
+  CHK_DT(bias, torch::kFloat32); CHK_CONTIG(bias);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  const long B = cv.size(0), L = w.size(0);
+  const int EP = cv.size(1);
+  TORCH_CHECK(w.size(1) == EP && EP % 32 == 0, "head_fwd EP");
+  TORCH_CHECK(out.size(0) == B && out.size(1) == L && bias.numel() == L,
+              "head_fwd shapes");
+  float* pmp = nullptr;
+  float* psp = nullptr;
+  if (pm.defined() && pm.numel() > 0) {
+    const long GXL = (L + 255) / 256;  // [lab_block, B] layout
+    CHK_DT(pm, torch::kFloat32); CHK_CONTIG(pm);
+    TORCH_CHECK(pm.numel() == B * GXL && ps.numel() == B * GXL, "partials");
+    pmp = pm.data_ptr<float>();
+    psp = ps.data_ptr<float>();
+  }
+  launch_head_fwd(cv.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
+                  out.data_ptr(), pmp, psp, B, L, EP, cur_stream());
+}
+
+void logsoftmax_nll_finalize(torch::Tensor logits, torch::Tensor pm,
+                             torch::Tensor ps, torch::Tensor label,
+                             torch::Tensor weight, torch::Tensor lse,
+                             torch::Tensor acc) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_DT(pm, torch::kFloat32); CHK_CONTIG(pm);
+  CHK_DT(lse, torch::kFloat32);
+  const int B = logits.size(0);
+  const long L = logits.size(1);
+  const int GX = (int)((L + 255) / 256);
+  TORCH_CHECK(pm.numel() == (long)B * GX && ps.numel() == (long)B * GX,
+              "partials");
+  const float* wp = weight.defined() && weight.numel() > 0
+                        ? weight.data_ptr<float>()
+                        : nullptr;
+  launch_lsm_finalize(logits.data_ptr(), pm.data_ptr<float>(),
+                      ps.data_ptr<float>(), label.data_ptr<long>(), wp,
+                      lse.data_ptr<float>(), acc.data_ptr<float>(), B, L, GX,
+                      cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -365,6 +417,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
   m.def("wgrad", &wgrad);
   m.def("colsum_bf16", &colsum_bf16);
+  m.def("head_fwd", &head_fwd);
+  m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
   m.def("head_wgrad", &head_wgrad);
   m.def("dgrad", &dgrad);
   m.def("adam_step_bf16", &adam_step_bf16);

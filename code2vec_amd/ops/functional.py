@@ -28,6 +28,10 @@ _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 # recover.  Kept (correct, GPU-tested) for small-table configs where the
 # gather output (B*C*KP bf16) no longer fits comfortably.
 FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
+# custom output-head forward (+ fused loss statistics); C2V_HEAD_FWD=0
+# falls back to hipBLASLt linear + full-pass loss forward
+_HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
+_NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
 # backward invokes them the moment a table's grad tensor is complete.
@@ -338,8 +342,29 @@ class OutputHead(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, cv_bf16, w, bias):
-        logits = torch.nn.functional.linear(
-            cv_bf16, w, bias.to(torch.bfloat16))
+        B, EP = cv_bf16.shape
+        L = w.shape[0]
+        if _HEAD_FWD and cv_bf16.is_cuda and EP % 32 == 0:
+            # custom MFMA forward: both fragments are contiguous in memory
+            # (K = EP is the fast axis of both cv and w), so this streams at
+            # the C-write bound where hipBLASLt runs a K=128 GEMM pipeline.
+            # When grads are on (training), the epilogue also emits the
+            # per-row online-softmax partials the NLL loss consumes, saving
+            # the loss kernel's full re-read of logits.
+            logits = torch.empty(B, L, dtype=torch.bfloat16,
+                                 device=cv_bf16.device)
+            if cv_bf16.requires_grad or w.requires_grad:
+                gx = (L + 255) // 256  # one partial per 256-label block
+                pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
+                ps = torch.empty_like(pm)
+            else:
+                pm = ps = _NONE_T
+            ext().head_fwd(cv_bf16, w, bias.float(), logits, pm, ps)
+            if pm is not _NONE_T:
+                logits._c2v_lsm_partials = (pm, ps)
+        else:
+            logits = torch.nn.functional.linear(
+                cv_bf16, w, bias.to(torch.bfloat16))
         ctx.save_for_backward(cv_bf16, w)
         return logits
 
@@ -378,7 +403,17 @@ class FusedLogSoftmaxNLL(torch.autograd.Function):
         lse = torch.empty(B, dtype=torch.float32, device=logits.device)
         # acc[0] = sum(w_y * nll), acc[1] = sum(w_y)
         acc = torch.zeros(2, dtype=torch.float32, device=logits.device)
-        ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc)
+        partials = getattr(logits, "_c2v_lsm_partials", None)
+        if partials is not None:
+            # the head-forward epilogue already reduced per-row (max, sum)
+            # partials over the logits — merge them instead of re-reading
+            # the whole [B, L] matrix
+            pm, ps = partials
+            del logits._c2v_lsm_partials
+            ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight,
+                                          lse, acc)
+        else:
+            ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc)
         loss = acc[0] / acc[1]
         ctx.save_for_backward(logits, label, weight, lse, acc)
         return loss

@@ -279,6 +279,66 @@ class TestLogSoftmaxNLL:
 
 
 # ---------------------------------------------------------------------------
+class TestHeadFwd:
+    @pytest.mark.parametrize("B,L,EP", [(24, 700, 128), (100, 7321, 128),
+                                        (300, 30000, 128), (24, 997, 320)])
+    def test_matches_linear_oracle(self, dev, B, L, EP):
+        from code2vec_amd.ops import ext
+
+        g = torch.Generator().manual_seed(11)
+        cv = (torch.randn(B, EP, generator=g) * 0.5).to(dev, torch.bfloat16)
+        w = (torch.randn(L, EP, generator=g) * 0.1).to(dev, torch.bfloat16)
+        bias = torch.randn(L, generator=g).to(dev)
+        out = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+        gx = (L + 255) // 256
+        pm = torch.empty(gx, B, dtype=torch.float32, device=dev)
+        ps = torch.empty_like(pm)
+        ext().head_fwd(cv, w, bias, out, pm, ps)
+
+        ref = cv.float() @ w.float().t() + bias
+        assert relerr(out.float(), ref) < 2e-2
+
+        # the fused (max, sumexp) partials must reproduce the row lse of the
+        # stored bf16 logits exactly (same rounding, different summation)
+        m = pm.max(dim=0, keepdim=True).values
+        lse = (ps * torch.exp(pm - m)).sum(dim=0).log() + m[0]
+        lse_ref = torch.logsumexp(out.float(), dim=1)
+        assert relerr(lse, lse_ref) < 1e-4
+
+    def test_output_head_plus_loss_fused_path(self, dev):
+        """OutputHead (custom fwd, stats on) + FusedLogSoftmaxNLL
+        (finalize path) vs the fp32 oracle end to end."""
+        from code2vec_amd.ops.functional import (
+            FusedLogSoftmaxNLL, OutputHead)
+
+        B, L, EP = 64, 7321, 128
+        g = torch.Generator().manual_seed(13)
+        cv = (torch.randn(B, EP, generator=g) * 0.5).to(dev, torch.bfloat16)
+        w = (torch.randn(L, EP, generator=g) * 0.1).to(dev, torch.bfloat16)
+        bias = torch.randn(L, generator=g).to(dev)
+        label = torch.randint(0, L, (B,), generator=g).to(dev)
+        weight = (torch.rand(L, generator=g) + 0.5).to(dev)
+
+        cvh = cv.clone().requires_grad_(True)
+        wh = w.clone().requires_grad_(True)
+        bh = bias.clone().requires_grad_(True)
+        logits = OutputHead.apply(cvh, wh, bh)
+        assert hasattr(logits, "_c2v_lsm_partials")
+        loss = FusedLogSoftmaxNLL.apply(logits, label, weight)
+        loss.backward()
+
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        ref = R.logsoftmax_nll(cvr @ wr.t() + br, label, weight)
+        ref.backward()
+        assert abs(float(loss) - float(ref)) / abs(float(ref)) < 1e-2
+        assert relerr(cvh.grad.float(), cvr.grad) < 3e-2
+        assert relerr(wh.grad.float(), wr.grad) < 3e-2
+        assert relerr(bh.grad.float(), br.grad) < 3e-2
+
+
+# ---------------------------------------------------------------------------
 class TestAdam:
     def test_bf16_master_matches_torch_adam(self, dev):
         from code2vec_amd.engine.optim import FusedAdam
