@@ -29,8 +29,12 @@ _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 # gather output (B*C*KP bf16) no longer fits comfortably.
 FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
 # custom output-head forward (+ fused loss statistics); C2V_HEAD_FWD=0
-# falls back to hipBLASLt linear + full-pass loss forward
+# falls back to hipBLASLt linear + full-pass loss forward.  Measured
+# faster up to mid-size label vocabs (top11 L=30k: 69 -> 45 us for the
+# head+loss pair) but ~100 us/step slower at java-large's L=261k, so
+# vocabs past C2V_HEAD_FWD_MAXL take the library path.
 _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
+_HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
@@ -344,7 +348,8 @@ class OutputHead(torch.autograd.Function):
     def forward(ctx, cv_bf16, w, bias):
         B, EP = cv_bf16.shape
         L = w.shape[0]
-        if _HEAD_FWD and cv_bf16.is_cuda and EP % 32 == 0:
+        if (_HEAD_FWD and cv_bf16.is_cuda and EP % 32 == 0
+                and L <= _HEAD_FWD_MAXL):
             # custom MFMA forward: both fragments are contiguous in memory
             # (K = EP is the fast axis of both cv and w), so this streams at
             # the C-write bound where hipBLASLt runs a K=128 GEMM pipeline.
