@@ -58,6 +58,10 @@ void launch_dgrad(const void*, const void*, void*, long, int, int,
                   hipStream_t);
 void launch_head_wgrad(const void*, const void*, void*, long, long, int,
                        hipStream_t);
+void launch_head_dgrad(const void*, const void*, float*, long, long,
+                       hipStream_t);
+void launch_transpose_w(const void*, void*, long, hipStream_t);
+void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
                      float*, long, long, int, hipStream_t);
@@ -361,6 +365,41 @@ void logsoftmax_nll_finalize(torch::Tensor logits, torch::Tensor pm,
                       cur_stream());
 }
 
+// dcv split-K partials: partials[ceil(L/512), B, 128] f32; wt = W
+// transposed [EP=128, L]
+void head_dgrad(torch::Tensor dlogits, torch::Tensor wt,
+                torch::Tensor partials) {
+  CHK_CUDA(dlogits); CHK_CONTIG(dlogits); CHK_DT(dlogits, torch::kBFloat16);
+  CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
+  CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
+  const long B = dlogits.size(0), L = dlogits.size(1);
+  TORCH_CHECK(wt.size(0) == 128 && wt.size(1) == L, "wt must be [128, L]");
+  TORCH_CHECK(L % 8 == 0, "head_dgrad needs L % 8 == 0");
+  TORCH_CHECK(partials.numel() == (L + 511) / 512 * B * 128, "partials");
+  launch_head_dgrad(dlogits.data_ptr(), wt.data_ptr(),
+                    partials.data_ptr<float>(), B, L, cur_stream());
+}
+
+void transpose_w(torch::Tensor w, torch::Tensor wt) {
+  CHK_CUDA(w); CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
+  CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
+  const long L = w.size(0);
+  TORCH_CHECK(w.size(1) == 128 && wt.size(0) == 128 && wt.size(1) == L &&
+              L % 8 == 0, "transpose_w shapes");
+  launch_transpose_w(w.data_ptr(), wt.data_ptr(), L, cur_stream());
+}
+
+void slab_sum_bf16(torch::Tensor partials, torch::Tensor out) {
+  CHK_CUDA(partials); CHK_CONTIG(partials);
+  CHK_DT(partials, torch::kFloat32);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  const long N = out.numel();
+  const long S = partials.numel() / N;
+  TORCH_CHECK(S * N == partials.numel() && N % 4 == 0, "slab shapes");
+  launch_slab_sum_bf16(partials.data_ptr<float>(), out.data_ptr(), (int)S,
+                       N, cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -420,6 +459,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_fwd", &head_fwd);
   m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
   m.def("head_wgrad", &head_wgrad);
+  m.def("head_dgrad", &head_dgrad);
+  m.def("transpose_w", &transpose_w);
+  m.def("slab_sum_bf16", &slab_sum_bf16);
   m.def("dgrad", &dgrad);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);

@@ -32,8 +32,6 @@
 #define HF_BATCH 64  // batch cols per block
 #define NINF (-3.0e38f)
 
-typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
-
 // merge (m2, s2) into (m, s): online-softmax pair combine
 #define HF_MERGE(m, s, m2, s2)                       \
   do {                                               \

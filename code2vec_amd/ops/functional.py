@@ -35,6 +35,8 @@ FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
 # vocabs past C2V_HEAD_FWD_MAXL take the library path.
 _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
 _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
+# custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
+_HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
@@ -143,6 +145,16 @@ def _scratch_f32(tag: str, shape, device) -> torch.Tensor:
     buf = _scratch_cache.get(key)
     if buf is None:
         buf = torch.zeros(shape, dtype=torch.float32, device=device)
+        _scratch_cache[key] = buf
+    return buf
+
+
+def _scratch_bf16(tag: str, shape, device) -> torch.Tensor:
+    """Persistent bf16 scratch (fully overwritten by its producer)."""
+    key = ("bf16", tag, tuple(shape), str(device))
+    buf = _scratch_cache.get(key)
+    if buf is None:
+        buf = torch.empty(shape, dtype=torch.bfloat16, device=device)
         _scratch_cache[key] = buf
     return buf
 
@@ -377,9 +389,23 @@ class OutputHead(torch.autograd.Function):
     def backward(ctx, dlogits):
         cv, w = ctx.saved_tensors
         dlogits = dlogits.contiguous()
-        dcv = dlogits @ w                      # [B, EP] bf16 (rocBLAS)
         B, L = dlogits.shape
         EP = cv.shape[1]
+        if (_HEAD_DGRAD and EP == 128 and L % 8 == 0 and dlogits.is_cuda
+                and L <= _HEAD_FWD_MAXL):  # big L: slab traffic dominates
+            # split-K MFMA dcv (head_dgrad.hip): hipBLASLt runs this
+            # skinny-output huge-K GEMM ~6x off the traffic floor.  W is
+            # transposed once so both fragments are contiguous loads.
+            wt = _scratch_bf16("head_wt", (128, L), w.device)
+            ext().transpose_w(w, wt)
+            split = (L + 511) // 512
+            partials = _scratch_f32("head_dgrad", (split, B, 128), w.device)
+            ext().head_dgrad(dlogits, wt, partials)
+            dcv = torch.empty(B, 128, dtype=torch.bfloat16,
+                              device=w.device)
+            ext().slab_sum_bf16(partials, dcv)
+        else:
+            dcv = dlogits @ w                  # [B, EP] bf16 (rocBLAS)
         # default OFF: measured slower than TunableOp rocBLAS at the
         # top11 shape (single-buffered staging; see PERF.md)
         if (os.environ.get("C2V_HEAD_WGRAD", "0") == "1"

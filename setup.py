@@ -23,6 +23,7 @@ SRC = [
     "code2vec_amd/ops/csrc/wgrad.hip",
     "code2vec_amd/ops/csrc/colsum.hip",
     "code2vec_amd/ops/csrc/head_fwd.hip",
+    "code2vec_amd/ops/csrc/head_dgrad.hip",
     "code2vec_amd/ops/csrc/dgrad.hip",
     "code2vec_amd/ops/csrc/head_wgrad.hip",
 ]

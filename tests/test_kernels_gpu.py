@@ -305,6 +305,46 @@ class TestHeadFwd:
         lse_ref = torch.logsumexp(out.float(), dim=1)
         assert relerr(lse, lse_ref) < 1e-4
 
+    @pytest.mark.parametrize("B,L", [(64, 1024), (300, 29992), (1024, 2048)])
+    def test_head_dgrad_matches_oracle(self, dev, B, L):
+        from code2vec_amd.ops import ext
+
+        g = torch.Generator().manual_seed(17)
+        dlogits = (torch.randn(B, L, generator=g) * 0.02).to(
+            dev, torch.bfloat16)
+        w = (torch.randn(L, 128, generator=g) * 0.1).to(dev, torch.bfloat16)
+        split = (L + 511) // 512
+        partials = torch.full((split, B, 128), float("nan"),
+                              dtype=torch.float32, device=dev)
+        ext().head_dgrad(dlogits, w.t().contiguous(), partials)
+        dcv = partials.sum(dim=0)
+        ref = dlogits.float() @ w.float()
+        assert relerr(dcv, ref) < 2e-2
+
+    def test_output_head_backward_custom_dcv(self, dev):
+        """OutputHead.backward's custom split-K dcv path (L % 8 == 0,
+        EP = 128) vs the fp32 oracle."""
+        from code2vec_amd.ops.functional import OutputHead
+
+        B, L, EP = 96, 4096, 128
+        g = torch.Generator().manual_seed(19)
+        cv = (torch.randn(B, EP, generator=g) * 0.5).to(dev, torch.bfloat16)
+        w = (torch.randn(L, EP, generator=g) * 0.1).to(dev, torch.bfloat16)
+        bias = torch.randn(L, generator=g).to(dev)
+        cvh = cv.clone().requires_grad_(True)
+        wh = w.clone().requires_grad_(True)
+        bh = bias.clone().requires_grad_(True)
+        dl = (torch.randn(B, L, generator=g) * 0.05).to(dev, torch.bfloat16)
+        OutputHead.apply(cvh, wh, bh).backward(dl)
+
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        (cvr @ wr.t() + br).backward(dl.float())
+        assert relerr(cvh.grad.float(), cvr.grad) < 3e-2
+        assert relerr(wh.grad.float(), wr.grad) < 3e-2
+        assert relerr(bh.grad.float(), br.grad) < 3e-2
+
     def test_output_head_plus_loss_fused_path(self, dev):
         """OutputHead (custom fwd, stats on) + FusedLogSoftmaxNLL
         (finalize path) vs the fp32 oracle end to end."""

@@ -58,6 +58,19 @@ def main():
                                                      lse, acc))
     t_fin = time_op(lambda: ext().logsoftmax_nll_finalize(out, pm, ps, label,
                                                           weight, lse, acc))
+    # head backward dcv: library GEMM vs split-K custom
+    t_dcv_lib = time_op(lambda: out @ w)
+    wt = w.t().contiguous()
+    split = (L + 511) // 512
+    partials = torch.empty(split, B, 128, dtype=torch.float32, device=dev)
+    dcv_out = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
+    t_tr = time_op(lambda: ext().transpose_w(w, wt))
+    t_dcv = time_op(lambda: ext().head_dgrad(out, wt, partials))
+    t_red = time_op(lambda: ext().slab_sum_bf16(partials, dcv_out))
+    print(f"dcv rocBLAS:                  {t_dcv_lib:7.1f} us")
+    print(f"dcv custom (tr+k+reduce):     {t_tr + t_dcv + t_red:7.1f} us "
+          f"({t_tr:.1f} + {t_dcv:.1f} + {t_red:.1f})")
+
     bw = (B * L * 2) / 1e9
     print(f"linear (hipBLASLt+TunableOp): {t_lin:7.1f} us")
     print(f"head_fwd no-stats:            {t_hf:7.1f} us "
