@@ -238,6 +238,11 @@ __global__ void scatter_group_kernel(const int* __restrict__ idx,
 // tensor AND re-zero the touched fp32 rows (so the scratch is reusable next
 // step without a full-buffer memset).  Rows with counts == 0 were never
 // written: emit bf16 zeros without touching the fp32 buffer.
+// Each wave screens 64 CONSECUTIVE rows with coalesced per-lane
+// flags/counts loads, then ballots and processes only the rows needing
+// work (a wave-per-row mapping makes every metadata read a solo
+// cache-line touch — 360k of them — and ran latency-bound at 37 us vs
+// ~8 us of actual byte traffic).
 __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
                                        const int* __restrict__ counts,
                                        unsigned char* __restrict__ flags,
@@ -245,28 +250,42 @@ __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int wpb = blockDim.x / WAVE;
-  long row = (long)blockIdx.x * wpb + wave;
+  const long nchunks = (T + WAVE - 1) / WAVE;
+  long chunk = (long)blockIdx.x * wpb + wave;
   const long stride = (long)gridDim.x * wpb;
-  for (; row < T; row += stride) {
-    const bool boundary = flags[row] != 0 && row != 0;
-    // interior-touched rows were written bf16-direct by the scatter kernel;
-    // row 0 (<PAD/>) always reads as untouched (its grads are exactly zero)
-    const bool interior = !boundary && row != 0 && counts[row] > 0;
-    if (interior) continue;
-    float* frow = dtable + row * S;
-    bf16* orow = out + row * S;
-    for (int c0 = lane * 2; c0 < S; c0 += WAVE * 2) {
-      bf16x2 o = {bf16(0.f), bf16(0.f)};
-      if (boundary) {
-        const float2 v = *(const float2*)(frow + c0);
-        o[0] = f2bf(v.x);
-        o[1] = f2bf(v.y);
-        float2 zz = {0.f, 0.f};
-        *(float2*)(frow + c0) = zz;
+  for (; chunk < nchunks; chunk += stride) {
+    const long r0 = chunk * WAVE;
+    const long rl = r0 + lane;
+    const bool valid = rl < T;
+    const bool boundary = valid && rl != 0 && flags[rl] != 0;
+    // interior-touched rows were written bf16-direct by the scatter
+    // kernel; row 0 (<PAD/>) always reads as untouched (its grads are
+    // exactly zero)
+    const bool interior =
+        valid && !boundary && rl != 0 && counts[rl] > 0;
+    const bool zero_row = valid && !boundary && !interior;
+    if (boundary) flags[rl] = 0;
+    const unsigned long long mb = __ballot(boundary);
+    unsigned long long m = mb | __ballot(zero_row);
+    while (m) {
+      const int r = __ffsll((long long)m) - 1;
+      m &= m - 1;
+      const long row = r0 + r;
+      const bool bnd = (mb >> r) & 1;
+      float* frow = dtable + row * S;
+      bf16* orow = out + row * S;
+      for (int c0 = lane * 2; c0 < S; c0 += WAVE * 2) {
+        bf16x2 o = {bf16(0.f), bf16(0.f)};
+        if (bnd) {
+          const float2 v = *(const float2*)(frow + c0);
+          o[0] = f2bf(v.x);
+          o[1] = f2bf(v.y);
+          float2 zz = {0.f, 0.f};
+          *(float2*)(frow + c0) = zz;
+        }
+        *(bf16x2*)(orow + c0) = o;
       }
-      *(bf16x2*)(orow + c0) = o;
     }
-    if (boundary && lane == 0) flags[row] = 0;
   }
 }
 
@@ -277,7 +296,8 @@ void launch_cast_clear_rows(float* dtable, const int* counts,
                             hipStream_t stream) {
   const int block = 256;
   const int wpb = block / WAVE;
-  const int grid = (int)min((T + wpb - 1) / wpb, (long)8192);
+  const long chunks = (T + WAVE - 1) / WAVE;
+  const int grid = (int)min((chunks + wpb - 1) / wpb, (long)8192);
   cast_clear_rows_kernel<<<grid, block, 0, stream>>>(dtable, counts, flags,
                                                      (bf16*)out, T, S);
 }
