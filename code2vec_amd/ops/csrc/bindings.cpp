@@ -62,6 +62,7 @@ void launch_head_dgrad(const void*, const void*, float*, long, long,
                        hipStream_t);
 void launch_transpose_w(const void*, void*, long, hipStream_t);
 void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
+void launch_slab_sum_f32(const float*, float*, int, int, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
                      float*, long, long, int, hipStream_t);
@@ -400,6 +401,16 @@ void slab_sum_bf16(torch::Tensor partials, torch::Tensor out) {
                        N, cur_stream());
 }
 
+void slab_sum_f32(torch::Tensor p, torch::Tensor out) {
+  CHK_CUDA(p); CHK_CONTIG(p); CHK_DT(p, torch::kFloat32);
+  CHK_CONTIG(out); CHK_DT(out, torch::kFloat32);
+  const int N = (int)out.numel();
+  const int S = (int)(p.numel() / N);
+  TORCH_CHECK((long)S * N == p.numel(), "slab_sum_f32 shapes");
+  launch_slab_sum_f32(p.data_ptr<float>(), out.data_ptr<float>(), S, N,
+                      cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -462,6 +473,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_dgrad", &head_dgrad);
   m.def("transpose_w", &transpose_w);
   m.def("slab_sum_bf16", &slab_sum_bf16);
+  m.def("slab_sum_f32", &slab_sum_f32);
   m.def("dgrad", &dgrad);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);

@@ -19,7 +19,29 @@ __global__ __launch_bounds__(256) void colsum_bf16_kernel(
   atomic_add_f32(out + col, acc);
 }
 
+// Tiny partial-slab reduce: p[S, N] f32 -> out[N] f32, one block per
+// column (N is small: EP/KP-scale; S <= ~1024 reduction-partial rows).
+// torch's generic reduce spends 10 us per call on these; this is ~2 us.
+__global__ __launch_bounds__(256) void slab_sum_f32_kernel(
+    const float* __restrict__ p, float* __restrict__ out, int S, int N) {
+  const int col = blockIdx.x;
+  float acc = 0.f;
+  for (int s = threadIdx.x; s < S; s += 256)
+    acc += p[(long)s * N + col];
+  acc = wave_reduce_sum(acc);
+  __shared__ float red[4];
+  const int wave = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) red[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) out[col] = red[0] + red[1] + red[2] + red[3];
+}
+
 extern "C" {
+
+void launch_slab_sum_f32(const float* p, float* out, int S, int N,
+                         hipStream_t stream) {
+  slab_sum_f32_kernel<<<N, 256, 0, stream>>>(p, out, S, N);
+}
 
 void launch_colsum_bf16(const void* x, float* out, long B, long L,
                         hipStream_t stream) {

@@ -168,6 +168,15 @@ def _scratch_flags(tag: str, rows: int, device) -> torch.Tensor:
     return buf
 
 
+def _slab_sum(p: torch.Tensor) -> torch.Tensor:
+    """Sum reduction-partial slabs [S, N] -> [N] f32 (block-per-column
+    kernel; torch's generic reduce costs ~10 us per call at these tiny
+    shapes)."""
+    out = torch.empty(p.shape[-1], dtype=torch.float32, device=p.device)
+    ext().slab_sum_f32(p, out)
+    return out
+
+
 def _group_by_index(idx: torch.Tensor, table_rows: int):
     """Counting sort: returns (sorted_idx i32, perm i64, counts i32)
     grouping equal indexes contiguously (ascending)."""
@@ -223,8 +232,8 @@ class CombinerLNTanh(torch.autograd.Function):
             dout.contiguous(), z, out, mean, rstd, gamma, beta, dz, dgamma_p,
             dbeta_p, E, p,
         )
-        dgamma = dgamma_p.sum(dim=0)
-        dbeta = dbeta_p.sum(dim=0)
+        dgamma = _slab_sum(dgamma_p)
+        dbeta = _slab_sum(dbeta_p)
         # dgrad: plain GEMM -> rocBLAS (TunableOp-tuned).  A hand-written
         # direct-fragment kernel (ops/csrc/dgrad.hip) measured 192 us vs
         # rocBLAS's 64 us at the top11 shape (8-wave full-KP duplicates B
@@ -292,8 +301,8 @@ class FusedGatherCombiner(torch.autograd.Function):
         dbeta_p = torch.empty(nblocks, EP, dtype=torch.float32, device=z.device)
         ext().combiner_bwd(dout.contiguous(), z, out, mean, rstd, gamma,
                            beta, dz, dgamma_p, dbeta_p, E, p)
-        dgamma = dgamma_p.sum(dim=0)
-        dbeta = dbeta_p.sum(dim=0)
+        dgamma = _slab_sum(dgamma_p)
+        dbeta = _slab_sum(dbeta_p)
         # dW: re-gathering split-K wgrad (X is never materialized)
         if EP <= 128 and KP <= 512:
             partials = torch.empty(256, KP, EP, dtype=torch.float32,
@@ -348,7 +357,7 @@ class AttentionPool(torch.autograd.Function):
             dcv.contiguous(), dattn.contiguous() if has_dattn else dattn,
             ccv, a, starts, attn, dccv, da_p, ctx.E, has_dattn,
         )
-        return dccv, da_p.sum(dim=0), None, None
+        return dccv, _slab_sum(da_p), None, None
 
 
 class OutputHead(torch.autograd.Function):
