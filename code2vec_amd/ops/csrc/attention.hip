@@ -72,27 +72,38 @@ __device__ __forceinline__ float group_row_dot(const bf16* row,
 }
 
 // ---------------------------------------------------------------------------
+// ts = round8(E) <= EP: only the valid columns are staged/dotted — the
+// EP-pad columns of ccv are zero and a/dcv pad is zero, so they never
+// contribute; trimming them keeps the tile under the 53.3-KB line for
+// 3 blocks/CU (54.8 KB = 2 blocks, a 33% occupancy loss).  Outputs in
+// [ts, EP) are written as explicit zeros.
 template <bool STAGE_LDS>
 __global__ __launch_bounds__(256) void attention_fwd_kernel(
     const bf16* __restrict__ ccv, const float* __restrict__ a,
     const int* __restrict__ starts, float* __restrict__ cv,
-    float* __restrict__ attn, int B, int C, int EP, int E) {
-  const int b = blockIdx.x;
+    float* __restrict__ attn, int B, int C, int EP, int E, int ts) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // layout: [ccv tile (opt)] [a f32 EP] [scores C] [red 64] [partials 4*EP]
   bf16* tile = (bf16*)smem;
-  float* lds_a = (float*)(smem + (STAGE_LDS ? (size_t)C * EP * 2 : 0));
+  float* lds_a = (float*)(smem + (STAGE_LDS ? (size_t)C * ts * 2 : 0));
   float* scores = lds_a + EP;
   float* red = scores + C;
   float* partials = red + 64;
+  const int rstride = STAGE_LDS ? ts : EP;  // tile vs global row stride
 
+  // persistent grid: only 3 blocks fit per CU at this LDS size, so a
+  // block-per-method grid (B=1024) runs an underfilled tail round;
+  // looping methods over occupancy*CU blocks balances across CUs
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
   const bf16* src = ccv + (long)b * C * EP;
   for (int e = threadIdx.x; e < EP; e += blockDim.x) lds_a[e] = a[e];
   if (STAGE_LDS) {
-    const int total = C * EP / 8;
+    const int row_chunks = ts / 8;
+    const int total = C * row_chunks;
     const uint4* s4 = (const uint4*)src;
     uint4* d4 = (uint4*)tile;
-    for (int i = threadIdx.x; i < total; i += blockDim.x) d4[i] = s4[i];
+    for (int i = threadIdx.x; i < total; i += blockDim.x)
+      d4[i] = s4[(i / row_chunks) * (EP / 8) + i % row_chunks];
   }
   __syncthreads();
 
@@ -103,8 +114,8 @@ __global__ __launch_bounds__(256) void attention_fwd_kernel(
 
   // phase 1: scores — each 16-lane group owns one context per iteration
   for (int c = group; c < C; c += 16) {
-    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
-    float s = group_row_dot(row, lds_a, EP, lane16);
+    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * rstride;
+    float s = group_row_dot(row, lds_a, ts, lane16);
     s = group16_reduce_sum(s);
     if (lane16 == 0) {
       const float mask = starts[(long)b * C + c] > 0 ? 1.0f : 0.0f;
@@ -123,12 +134,14 @@ __global__ __launch_bounds__(256) void attention_fwd_kernel(
   const int c_hi = min(C, c_lo + span);
   for (int e0 = lane * 2; e0 < EP; e0 += WAVE * 2) {
     float acc0 = 0.f, acc1 = 0.f;
-    for (int c = c_lo; c < c_hi; ++c) {
-      const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
-      const bf16x2 v = *(const bf16x2*)(row + e0);
-      const float at = scores[c];
-      acc0 += at * bf2f(v[0]);
-      acc1 += at * bf2f(v[1]);
+    if (e0 < ts) {
+      for (int c = c_lo; c < c_hi; ++c) {
+        const bf16* row = (STAGE_LDS ? tile : src) + (long)c * rstride;
+        const bf16x2 v = *(const bf16x2*)(row + e0);
+        const float at = scores[c];
+        acc0 += at * bf2f(v[0]);
+        acc1 += at * bf2f(v[1]);
+      }
     }
     partials[wave * EP + e0] = acc0;
     partials[wave * EP + e0 + 1] = acc1;
@@ -137,6 +150,8 @@ __global__ __launch_bounds__(256) void attention_fwd_kernel(
   for (int e = threadIdx.x; e < EP; e += blockDim.x) {
     cv[(long)b * EP + e] = partials[e] + partials[EP + e] +
                            partials[2 * EP + e] + partials[3 * EP + e];
+  }
+  __syncthreads();  // smem reused by the next method
   }
 }
 
@@ -150,26 +165,30 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
     const bf16* __restrict__ ccv, const float* __restrict__ a,
     const int* __restrict__ starts, const float* __restrict__ attn,
     bf16* __restrict__ dccv, float* __restrict__ da, int B, int C, int EP,
-    int E, int has_dattn) {
-  const int b = blockIdx.x;
+    int E, int has_dattn, int ts) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* tile = (bf16*)smem;
-  float* lds_dcv = (float*)(smem + (STAGE_LDS ? (size_t)C * EP * 2 : 0));
+  float* lds_dcv = (float*)(smem + (STAGE_LDS ? (size_t)C * ts * 2 : 0));
   float* lds_a = lds_dcv + EP;
   float* ds = lds_a + EP;        // [C]
   float* red = ds + C;           // [64]
   float* partials = red + 64;    // [4][EP] (da)
+  const int rstride = STAGE_LDS ? ts : EP;
 
+  // persistent grid (see attention_fwd_kernel)
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
   const bf16* src = ccv + (long)b * C * EP;
   for (int e = threadIdx.x; e < EP; e += blockDim.x) {
     lds_dcv[e] = dcv[(long)b * EP + e];
     lds_a[e] = a[e];
   }
   if (STAGE_LDS) {
-    const int total = C * EP / 8;
+    const int row_chunks = ts / 8;
+    const int total = C * row_chunks;
     const uint4* s4 = (const uint4*)src;
     uint4* d4 = (uint4*)tile;
-    for (int i = threadIdx.x; i < total; i += blockDim.x) d4[i] = s4[i];
+    for (int i = threadIdx.x; i < total; i += blockDim.x)
+      d4[i] = s4[(i / row_chunks) * (EP / 8) + i % row_chunks];
   }
   __syncthreads();
 
@@ -181,8 +200,8 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   // phase A: g[c] (group-per-context) and block-reduced sum_g
   float part = 0.f;
   for (int c = group; c < C; c += 16) {
-    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
-    float g = group_row_dot(row, lds_dcv, EP, lane16);
+    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * rstride;
+    float g = group_row_dot(row, lds_dcv, ts, lane16);
     g = group16_reduce_sum(g);
     if (lane16 == 0) {
       if (has_dattn) g += dattn[(long)b * C + c];
@@ -215,12 +234,12 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   for (int c = group; c < C; c += 16) {
     const float at = attn[(long)b * C + c];
     const float dsc = ds[c];
-    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * EP;
+    const bf16* row = (STAGE_LDS ? tile : src) + (long)c * rstride;
     bf16* drow = dccv + ((long)b * C + c) * EP;
 #pragma unroll
     for (int i = 0; i < 3; ++i) {
       const int e0 = lane16 * 8 + 128 * i;
-      if (e0 < EP) {
+      if (e0 < ts) {
         bf16 v[8], o[8];
         *(uint4*)v = *(const uint4*)(row + e0);
 #pragma unroll
@@ -229,6 +248,10 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
           da_acc[i][j] += dsc * bf2f(v[j]);
         }
         *(uint4*)(drow + e0) = *(uint4*)o;
+      } else if (e0 < EP) {
+        // pad columns: dcv/a pad is zero, so the grad is exactly zero
+        const uint4 z = {0, 0, 0, 0};
+        *(uint4*)(drow + e0) = z;
       }
     }
   }
@@ -240,7 +263,7 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
 #pragma unroll
   for (int i = 0; i < 3; ++i) {
     const int e_base = lane16 * 8 + 128 * i;
-    if (e_base < EP) {
+    if (e_base < ts) {
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         atomicAdd(&partials[e_base + j], da_acc[i][j]);
@@ -249,12 +272,28 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
   __syncthreads();
   for (int e = threadIdx.x; e < EP; e += blockDim.x)
     da[(long)b * EP + e] = e < E ? partials[e] : 0.f;
+  __syncthreads();  // smem reused by the next method
+  }
+}
+
+// grid = min(B, occupancy * CUs): just enough resident blocks that the
+// persistent method loop balances across CUs with no tail round
+template <typename K>
+static int attn_grid(K kern, size_t smem, int B) {
+  static int ncu = 0;
+  if (ncu == 0)
+    hipDeviceGetAttribute(&ncu, hipDeviceAttributeMultiprocessorCount, 0);
+  int occ = 0;
+  hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ, kern, 256, smem);
+  if (occ < 1) occ = 1;
+  const long g = (long)occ * (ncu > 0 ? ncu : 256);
+  return (int)(g < B ? g : B);
 }
 
 extern "C" {
 
-static size_t attn_smem(int C, int EP, bool stage, bool bwd) {
-  size_t s = stage ? (size_t)C * EP * 2 : 0;
+static size_t attn_smem(int C, int EP, int ts, bool stage, bool bwd) {
+  size_t s = stage ? (size_t)C * ts * 2 : 0;
   s += (size_t)EP * 4;            // a (fwd) / dcv (bwd)
   if (bwd) s += (size_t)EP * 4;   // a (bwd extra)
   s += (size_t)C * 4;             // scores / ds
@@ -263,17 +302,25 @@ static size_t attn_smem(int C, int EP, bool stage, bool bwd) {
   return s;
 }
 
+static int attn_ts(int E, int EP) {
+  int ts = (E + 7) & ~7;
+  return ts > EP ? EP : ts;
+}
+
 void launch_attention_fwd(const void* ccv, const float* a, const int* starts,
                           float* cv, float* attn, int B, int C, int EP, int E,
                           hipStream_t stream) {
-  bool stage = attn_smem(C, EP, true, false) <= 160 * 1024 - 1024;
-  size_t smem = attn_smem(C, EP, stage, false);
+  const int ts = attn_ts(E, EP);
+  bool stage = attn_smem(C, EP, ts, true, false) <= 160 * 1024 - 1024;
+  size_t smem = attn_smem(C, EP, ts, stage, false);
   if (stage)
-    attention_fwd_kernel<true><<<B, 256, smem, stream>>>(
-        (const bf16*)ccv, a, starts, cv, attn, B, C, EP, E);
+    attention_fwd_kernel<true>
+        <<<attn_grid(attention_fwd_kernel<true>, smem, B), 256, smem,
+           stream>>>((const bf16*)ccv, a, starts, cv, attn, B, C, EP, E, ts);
   else
-    attention_fwd_kernel<false><<<B, 256, smem, stream>>>(
-        (const bf16*)ccv, a, starts, cv, attn, B, C, EP, E);
+    attention_fwd_kernel<false>
+        <<<attn_grid(attention_fwd_kernel<false>, smem, B), 256, smem,
+           stream>>>((const bf16*)ccv, a, starts, cv, attn, B, C, EP, E, ts);
 }
 
 void launch_attention_bwd(const float* dcv, const float* dattn,
@@ -281,16 +328,19 @@ void launch_attention_bwd(const float* dcv, const float* dattn,
                           const float* attn, void* dccv, float* da, int B,
                           int C, int EP, int E, int has_dattn,
                           hipStream_t stream) {
-  bool stage = attn_smem(C, EP, true, true) <= 160 * 1024 - 1024;
-  size_t smem = attn_smem(C, EP, stage, true);
+  const int ts = attn_ts(E, EP);
+  bool stage = attn_smem(C, EP, ts, true, true) <= 160 * 1024 - 1024;
+  size_t smem = attn_smem(C, EP, ts, stage, true);
   if (stage)
-    attention_bwd_kernel<true><<<B, 256, smem, stream>>>(
-        dcv, dattn, (const bf16*)ccv, a, starts, attn, (bf16*)dccv, da, B, C,
-        EP, E, has_dattn);
+    attention_bwd_kernel<true>
+        <<<attn_grid(attention_bwd_kernel<true>, smem, B), 256, smem,
+           stream>>>(dcv, dattn, (const bf16*)ccv, a, starts, attn,
+                     (bf16*)dccv, da, B, C, EP, E, has_dattn, ts);
   else
-    attention_bwd_kernel<false><<<B, 256, smem, stream>>>(
-        dcv, dattn, (const bf16*)ccv, a, starts, attn, (bf16*)dccv, da, B, C,
-        EP, E, has_dattn);
+    attention_bwd_kernel<false>
+        <<<attn_grid(attention_bwd_kernel<false>, smem, B), 256, smem,
+           stream>>>(dcv, dattn, (const bf16*)ccv, a, starts, attn,
+                     (bf16*)dccv, da, B, C, EP, E, has_dattn, ts);
 }
 
 }  // extern "C"
