@@ -37,6 +37,11 @@ _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
 _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
 # custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
+# combiner dgrad through the head-forward kernel — measured a wash vs
+# rocBLAS (1.350 vs 1.343 ms/step): the head kernel's 4-KB-per-wave
+# tiling doubles the wave count at this 131-MB output and per-wave
+# overhead eats the traffic win.  Kept as an opt-in experiment.
+_DGRAD2 = os.environ.get("C2V_DGRAD2", "0") == "1"
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
@@ -168,6 +173,25 @@ def _scratch_flags(tag: str, rows: int, device) -> torch.Tensor:
     return buf
 
 
+def _combiner_dgrad(dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """dx = dz @ w ([M, EP] @ [EP, KP], K = EP = 128) — a 131-MB
+    streaming write that hipBLASLt runs 3x off its floor (MT64x256x64,
+    91 us at top11).  With the weight transposed back to [KP, EP] this
+    is EXACTLY the head-forward kernel's shape (labels -> KP rows,
+    batch -> dz rows, zero bias, no loss stats), so head_fwd.hip is
+    reused as-is: contiguous fragments, LDS-staged dz tiles, packed
+    full-line nontemporal stores."""
+    M, EP = dz.shape
+    KP = w.shape[1]
+    if not (_DGRAD2 and dz.is_cuda and EP == 128):
+        return dz @ w
+    w2 = w.t().contiguous()  # [KP, EP], 82 KB at top11
+    zb = _scratch_f32("dgrad_zero_bias", (KP,), w.device)  # stays zero
+    dx = torch.empty(M, KP, dtype=torch.bfloat16, device=w.device)
+    ext().head_fwd(dz, w2, zb, dx, _NONE_T, _NONE_T)
+    return dx
+
+
 def _slab_sum(p: torch.Tensor) -> torch.Tensor:
     """Sum reduction-partial slabs [S, N] -> [N] f32 (block-per-column
     kernel; torch's generic reduce costs ~10 us per call at these tiny
@@ -245,7 +269,7 @@ class CombinerLNTanh(torch.autograd.Function):
                              device=w.device)
             ext().dgrad(dz, w2, dx)
         else:
-            dx = dz @ w
+            dx = _combiner_dgrad(dz, w)
         # wgrad: custom split-K MFMA kernel for the skinny big-K shape
         # (hipBLASLt is ~3.5x off there); partial slabs summed here.
         KP = x.shape[1]
@@ -313,8 +337,8 @@ class FusedGatherCombiner(torch.autograd.Function):
         else:
             x = GatherConcat.apply(starts, paths, ends, term_w, path_w)
             dw = (x.t() @ dz).t().contiguous()
-        # dX for the embedding scatter: one plain GEMM
-        dx = dz @ w
+        # dX for the embedding scatter
+        dx = _combiner_dgrad(dz, w)
         dterm, dpath = _scatter_embedding_grads(
             starts, paths, ends, dx, term_w.shape, path_w.shape,
             term_w.data_ptr(), path_w.data_ptr()
