@@ -61,6 +61,7 @@ void launch_head_wgrad(const void*, const void*, void*, long, long, int,
 void launch_head_dgrad(const void*, const void*, float*, long, long,
                        hipStream_t);
 void launch_transpose_w(const void*, void*, long, hipStream_t);
+void launch_dgrad2(const void*, const void*, void*, long, long, hipStream_t);
 void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
 void launch_slab_sum_f32(const float*, float*, int, int, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
@@ -411,6 +412,19 @@ void slab_sum_f32(torch::Tensor p, torch::Tensor out) {
                       cur_stream());
 }
 
+// dx = dz @ w2^T with w2 = [KP, EP=128] (the re-transposed combiner weight)
+void dgrad2(torch::Tensor dz, torch::Tensor w2, torch::Tensor dx) {
+  CHK_CUDA(dz); CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
+  CHK_CONTIG(w2); CHK_DT(w2, torch::kBFloat16);
+  CHK_CONTIG(dx); CHK_DT(dx, torch::kBFloat16);
+  const long M = dz.size(0), KP = w2.size(0);
+  TORCH_CHECK(dz.size(1) == 128 && w2.size(1) == 128, "dgrad2 EP");
+  TORCH_CHECK(dx.size(0) == M && dx.size(1) == KP && KP % 8 == 0,
+              "dgrad2 shapes");
+  launch_dgrad2(dz.data_ptr(), w2.data_ptr(), dx.data_ptr(), M, KP,
+                cur_stream());
+}
+
 void colsum_bf16(torch::Tensor x, torch::Tensor out) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_DT(out, torch::kFloat32);
@@ -472,6 +486,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
   m.def("transpose_w", &transpose_w);
+  m.def("dgrad2", &dgrad2);
   m.def("slab_sum_bf16", &slab_sum_bf16);
   m.def("slab_sum_f32", &slab_sum_f32);
   m.def("dgrad", &dgrad);

@@ -37,11 +37,8 @@ _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
 _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
 # custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
-# combiner dgrad through the head-forward kernel — measured a wash vs
-# rocBLAS (1.350 vs 1.343 ms/step): the head kernel's 4-KB-per-wave
-# tiling doubles the wave count at this 131-MB output and per-wave
-# overhead eats the traffic win.  Kept as an opt-in experiment.
-_DGRAD2 = os.environ.get("C2V_DGRAD2", "0") == "1"
+# combiner dgrad through dgrad2.hip (C2V_DGRAD2=0 -> rocBLAS)
+_DGRAD2 = os.environ.get("C2V_DGRAD2", "1") == "1"
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
@@ -176,19 +173,18 @@ def _scratch_flags(tag: str, rows: int, device) -> torch.Tensor:
 def _combiner_dgrad(dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """dx = dz @ w ([M, EP] @ [EP, KP], K = EP = 128) — a 131-MB
     streaming write that hipBLASLt runs 3x off its floor (MT64x256x64,
-    91 us at top11).  With the weight transposed back to [KP, EP] this
-    is EXACTLY the head-forward kernel's shape (labels -> KP rows,
-    batch -> dz rows, zero bias, no loss stats), so head_fwd.hip is
-    reused as-is: contiguous fragments, LDS-staged dz tiles, packed
-    full-line nontemporal stores."""
+    91 us at top11).  dgrad2.hip computes it transposed-orientation with
+    the weight re-transposed to [KP, EP]: both fragments contiguous,
+    dz tiles staged block-cooperatively, packed full-line nontemporal
+    stores; 128-row batch tiles keep per-wave overhead half of a naive
+    head_fwd reuse (which measured a wash — see PERF.md)."""
     M, EP = dz.shape
     KP = w.shape[1]
-    if not (_DGRAD2 and dz.is_cuda and EP == 128):
+    if not (_DGRAD2 and dz.is_cuda and EP == 128 and KP % 8 == 0):
         return dz @ w
     w2 = w.t().contiguous()  # [KP, EP], 82 KB at top11
-    zb = _scratch_f32("dgrad_zero_bias", (KP,), w.device)  # stays zero
     dx = torch.empty(M, KP, dtype=torch.bfloat16, device=w.device)
-    ext().head_fwd(dz, w2, zb, dx, _NONE_T, _NONE_T)
+    ext().dgrad2(dz, w2, dx)
     return dx
 
 

@@ -321,6 +321,19 @@ class TestHeadFwd:
         ref = dlogits.float() @ w.float()
         assert relerr(dcv, ref) < 2e-2
 
+    @pytest.mark.parametrize("M,KP", [(200, 320), (1000, 384), (96, 104)])
+    def test_dgrad2_matches_oracle(self, dev, M, KP):
+        from code2vec_amd.ops import ext
+
+        g = torch.Generator().manual_seed(23)
+        dz = (torch.randn(M, 128, generator=g) * 0.1).to(dev, torch.bfloat16)
+        w = (torch.randn(128, KP, generator=g) * 0.1).to(dev, torch.bfloat16)
+        dx = torch.full((M, KP), float("nan"), dtype=torch.bfloat16,
+                        device=dev)
+        ext().dgrad2(dz, w.t().contiguous(), dx)
+        ref = dz.float() @ w.float()
+        assert relerr(dx.float(), ref) < 2e-2
+
     def test_output_head_backward_custom_dcv(self, dev):
         """OutputHead.backward's custom split-K dcv path (L % 8 == 0,
         EP = 128) vs the fp32 oracle."""
