@@ -250,3 +250,27 @@ def test_native_parser_matches_python(tiny_corpus):
         assert a.source == b.source
         assert a.aliases == b.aliases
         assert np.array_equal(a.path_contexts, b.path_contexts)
+
+
+def test_gen_corpus_cli_output_parses(tmp_path):
+    """tools/gen_corpus.py writes the reference file format end to end:
+    the reader must parse its output with matching counts."""
+    import subprocess
+    import sys
+    import os
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = tmp_path / "gen"
+    res = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "gen_corpus.py"),
+         "--out", str(out), "--methods", "37", "--terminals", "60",
+         "--paths", "50", "--max_contexts", "9", "--seed", "3"],
+        capture_output=True, text=True, timeout=120)
+    assert res.returncode == 0, res.stderr
+    r = CorpusReader(str(out / "corpus.txt"), str(out / "path_idxs.txt"),
+                     str(out / "terminal_idxs.txt"))
+    assert len(r.items) == 37
+    assert all(len(it.path_contexts) >= 1 for it in r.items)
+    assert all(it.normalized_label for it in r.items)
+    # terminal vocab includes PAD + @question shift semantics
+    assert r.terminal_vocab.stoi["@question"] == 1
