@@ -11,7 +11,6 @@ import time
 from os import path as osp
 from typing import Optional
 
-import numpy as np
 import torch
 
 from concurrent.futures import ThreadPoolExecutor
@@ -19,7 +18,6 @@ from concurrent.futures import ThreadPoolExecutor
 from ..data.builder import DatasetBuilder
 from ..models.code2vec import (
     Code2VecHIP,
-    Code2VecTorch,
     reference_state_dict_torch,
 )
 from ..parallel.dist import DistContext, all_reduce_sum_list, all_reduce_sum_scalar

@@ -13,8 +13,6 @@
 // block-wide LDS patch (b64 writes, full 128-B line nontemporal stores).
 // KP/EP tails fall back to rocBLAS in functional.py (KP%8==0, EP==128).
 
-#include <cstdlib>
-
 #include "common.h"
 
 #define DG_LABS 256   // kp rows per block
