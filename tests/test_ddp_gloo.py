@@ -62,14 +62,15 @@ def _worker(rank, world, port, result_file):
 
 
 @pytest.mark.timeout(300)
-def test_ddp2_matches_single_process_large_batch(tmp_path):
-    """DP=2 with half batches + grad averaging == single process on the
-    full batch (same init, fp32 => near-bitwise)."""
+@pytest.mark.parametrize("world", [2, 4])
+def test_ddp2_matches_single_process_large_batch(tmp_path, world):
+    """DP=2/4 with batch shards + grad averaging == single process on
+    the full batch (same init, fp32 => near-bitwise)."""
     ctx = mp.get_context("spawn")
-    port = 29531
+    port = 29531 + world
     result_file = str(tmp_path / "params.pt")
-    procs = [ctx.Process(target=_worker, args=(r, 2, port, result_file))
-             for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, world, port, result_file))
+             for r in range(world)]
     for pr in procs:
         pr.start()
     for pr in procs:
@@ -90,7 +91,7 @@ def test_ddp2_matches_single_process_large_batch(tmp_path):
     optim = torch.optim.Adam(model.parameters(), lr=0.01)
     w = torch.ones(opt.label_count)
     rng = np.random.default_rng(99)
-    s, p, e, y = synthetic_batch(rng, 16, opt.max_path_length,
+    s, p, e, y = synthetic_batch(rng, 8 * world, opt.max_path_length,
                                  opt.terminal_count, opt.path_count,
                                  opt.label_count)
     s = torch.from_numpy(s); p = torch.from_numpy(p)
