@@ -274,3 +274,74 @@ def test_gen_corpus_cli_output_parses(tmp_path):
     assert all(it.normalized_label for it in r.items)
     # terminal vocab includes PAD + @question shift semantics
     assert r.terminal_vocab.stoi["@question"] == 1
+
+
+@pytest.mark.parametrize("variant", ["crlf", "no_trailing_blank",
+                                     "no_label", "empty_paths",
+                                     "extra_blanks", "spaces"])
+def test_native_parser_matches_python_edge_corpora(tmp_path, variant):
+    """C++ and Python parsers must agree on adversarial corpus shapes:
+    CRLF line endings, missing trailing blank line, records without
+    labels or paths, runs of blank lines, stray spaces."""
+    from code2vec_amd.data import reader as RD
+    from code2vec_amd.data.vocab import Vocab
+
+    if RD._native is None:
+        pytest.skip("native parser not built")
+
+    base = (
+        "#0\n"
+        "label:fooBar\n"
+        "class:A.java\n"
+        "doc: something to discard\n"
+        "paths:\n"
+        "1\t2\t3\n"
+        "4\t5\t6\n"
+        "vars:\n"
+        "counter\t@var_0\n"
+        "\n"
+        "#1\n"
+        "label:bazQux9\n"
+        "paths:\n"
+        "7\t8\t9\n"
+        "\n"
+    )
+    if variant == "crlf":
+        text = base.replace("\n", "\r\n")
+    elif variant == "no_trailing_blank":
+        text = base.rstrip("\n") + "\n"  # last record not blank-terminated
+    elif variant == "no_label":
+        text = "#0\npaths:\n1\t2\t3\n\n" + base
+    elif variant == "empty_paths":
+        text = "#5\nlabel:onlyVars\nvars:\nx\t@var_0\n\n" + base
+    elif variant == "extra_blanks":
+        text = base.replace("\n\n", "\n\n\n\n")
+    elif variant == "spaces":
+        text = base.replace("label:fooBar", "label:fooBar  ")
+
+    cp = tmp_path / "corpus.txt"
+    cp.write_text(text)
+
+    def parse(force_python):
+        r = CorpusReader.__new__(CorpusReader)
+        r.infer_method = True
+        r.infer_variable = False
+        r.label_vocab = Vocab()
+        r.items = []
+        if force_python:
+            r._load(str(cp))
+        else:
+            r._load_native(str(cp))
+        return r
+
+    rn, rp = parse(False), parse(True)
+    assert len(rn.items) == len(rp.items)
+    assert rn.label_vocab.stoi == rp.label_vocab.stoi
+    for a, b in zip(rn.items, rp.items):
+        assert a.id == b.id
+        assert a.label == b.label
+        assert a.normalized_label == b.normalized_label
+        assert a.source == b.source
+        assert a.aliases == b.aliases
+        assert np.array_equal(np.asarray(a.path_contexts),
+                              np.asarray(b.path_contexts))
