@@ -73,11 +73,12 @@ def test_cli_end_to_end(tmp_path, tiny_corpus):
     assert parts[1] in ("True", "False")
 
 
-def test_cli_variable_task(tmp_path, tiny_corpus):
-    out_dir = run_cli(
-        tmp_path, tiny_corpus,
-        extra=["--infer_method_name", "false", "--infer_variable_name", "true"],
-    )
+@pytest.mark.parametrize("shuffle", [False, True])
+def test_cli_variable_task(tmp_path, tiny_corpus, shuffle):
+    extra = ["--infer_method_name", "false", "--infer_variable_name", "true"]
+    if shuffle:
+        extra += ["--shuffle_variable_indexes", "true"]
+    out_dir = run_cli(tmp_path, tiny_corpus, extra=extra)
     assert (out_dir / "code.vec").exists()
 
 
