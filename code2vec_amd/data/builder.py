@@ -154,7 +154,8 @@ class DatasetBuilder:
 
     def _refresh(self, base: List[CodeItem], tag: str, epoch: int,
                  stream: int) -> EpochData:
-        shard_idx = self._shard_indices(len(base), epoch)
+        shard_idx = self._shard_indices(len(base), epoch,
+                                        equal_shards=(tag == "train"))
         if (
             _native is not None
             and self.reader.infer_method
@@ -165,14 +166,66 @@ class DatasetBuilder:
             return self._build_native(base, tag, f"{tag}{epoch % 2}",
                                       shard_idx, epoch, stream)
         items = [base[i] for i in shard_idx]
-        return self.build_data(items, self.option.max_path_length,
+        data = self.build_data(items, self.option.max_path_length,
                                epoch=epoch, stream=stream)
+        if tag == "train" and self.world_size > 1 and self.reader.infer_variable:
+            # variable task: rows per item vary, so equal ITEM counts do not
+            # give equal ROW (=> batch) counts across ranks.  Every rank can
+            # compute every rank's row count from the shared reader state, so
+            # truncate to the global min deterministically — no collective.
+            data = self._truncate_rows(data, base, epoch)
+        return data
 
-    def _shard_indices(self, n: int, epoch: int) -> np.ndarray:
+    def _rows_per_item(self, base: List[CodeItem]) -> np.ndarray:
+        key = ("rows", id(base))
+        cached = self._flat_cache.get(key)
+        if cached is None:
+            m = 1 if self.reader.infer_method else 0
+            cached = np.fromiter(
+                (m + len(_filter_variable_aliases(it.aliases)) for it in base),
+                count=len(base), dtype=np.int64,
+            )
+            self._flat_cache[key] = cached
+        return cached
+
+    def _truncate_rows(self, data: EpochData, base: List[CodeItem],
+                       epoch: int) -> EpochData:
+        rows = self._rows_per_item(base)
+        rng = np.random.default_rng([self.seed, epoch, 0x5A5A])
+        order = rng.permutation(len(base))
+        per = len(base) // self.world_size
+        min_rows = min(
+            int(rows[order[r :: self.world_size][:per]].sum())
+            for r in range(self.world_size)
+        )
+        if len(data) <= min_rows:
+            return data
+        return EpochData(
+            ids=data.ids[:min_rows],
+            starts=data.starts[:min_rows],
+            paths=data.paths[:min_rows],
+            ends=data.ends[:min_rows],
+            labels=data.labels[:min_rows],
+        )
+
+    def _shard_indices(self, n: int, epoch: int,
+                       equal_shards: bool = False) -> np.ndarray:
+        """Per-rank item indexes for one epoch.
+
+        Train shards are truncated to a common size floor(n/world): every
+        rank then runs the SAME number of batches per epoch, so the
+        per-step gradient all-reduces stay in lockstep (a rank with one
+        extra batch would fire collectives no peer matches — deadlock).
+        The dropped tail (< world items) rotates with the epoch-seeded
+        permutation, so all items are still seen across epochs.  Test
+        shards keep every item (eval has no in-loop collectives)."""
         if self.world_size <= 1:
             return np.arange(n)
         rng = np.random.default_rng([self.seed, epoch, 0x5A5A])
         order = rng.permutation(n)
+        if equal_shards:
+            per = n // self.world_size
+            return order[self.rank :: self.world_size][:per]
         return order[self.rank :: self.world_size]
 
     # ------------------------------------------------------------------

@@ -99,9 +99,47 @@ class CorpusReader:
             self._load_native(corpus_path)
         else:
             self._load(corpus_path)
+        self._validate_indices()
 
         logger.info("label vocab size: %d", len(self.label_vocab))
         logger.info("corpus: %d", len(self.items))
+
+    def _validate_indices(self) -> None:
+        """Fail loudly on out-of-range corpus indices at load time.
+
+        The replicated vocab quirk (duplicate names in an index file are
+        skipped, so a file's max index can exceed len(vocab)-1) makes
+        out-of-range indices reachable with real data.  The reference
+        fails in nn.Embedding (table rows == len(vocab)); our HIP gather
+        must never see such an index, so we validate once on the host
+        here instead of bounds-checking in kernels."""
+        t_hi = len(self.terminal_vocab)
+        p_hi = len(self.path_vocab)
+        flat = getattr(self, "_flat_contexts", None)
+        if flat is not None and flat.shape[0] > 0:
+            m = flat.max(axis=0)  # native path: one pass over the flat array
+            s_max, p_max, e_max = int(m[0]), int(m[1]), int(m[2])
+            lo = int(flat.min())
+        else:
+            s_max = p_max = e_max = -1
+            lo = 0
+            for it in self.items:
+                pcs = it.path_contexts
+                if pcs.shape[0] == 0:
+                    continue
+                m = pcs.max(axis=0)
+                s_max = max(s_max, int(m[0]))
+                p_max = max(p_max, int(m[1]))
+                e_max = max(e_max, int(m[2]))
+                lo = min(lo, int(pcs.min()))
+        if lo < 0 or max(s_max, e_max) >= t_hi or p_max >= p_hi:
+            raise ValueError(
+                "corpus contains out-of-range indices: "
+                f"start/end max {max(s_max, e_max)} (terminal vocab {t_hi} "
+                f"rows incl. @question shift), path max {p_max} (path vocab "
+                f"{p_hi} rows), min {lo} (must be >= 0). The corpus does "
+                "not match the given index files."
+            )
 
     def _load_native(self, corpus_path: str) -> None:
         """Native parse (data/csrc/epoch_builder.cpp::parse_corpus) +
@@ -109,6 +147,7 @@ class CorpusReader:
         reference); path_contexts become zero-copy views of one flat
         array."""
         res = _native.parse_corpus(corpus_path)
+        self._flat_contexts = res["contexts"].numpy()
         ids = res["ids"].numpy()
         offsets = res["offsets"].numpy()
         voffsets = res["var_offsets"].numpy()

@@ -45,14 +45,30 @@ class BatchIterator:
             [i if i is not None else -1 for i in data.ids], dtype=torch.int64
         )
         if self.prefetch:
-            def _pin(t):
-                return t if t.is_pinned() else t.pin_memory()
-
-            self.starts = _pin(self.starts)
-            self.paths = _pin(self.paths)
-            self.ends = _pin(self.ends)
-            self.labels = _pin(self.labels)
             self._copy_stream = torch.cuda.Stream(self.device)
+            # Double-buffered PINNED staging: a fancy-indexed gather
+            # (self.starts[idx]) materializes a *pageable* tensor, which
+            # silently degrades the non_blocking H2D copy to a staged
+            # synchronous one.  Instead each batch is gathered into one of
+            # two persistent pinned buffers and copied from there; a
+            # per-slot event keeps the host from overwriting a slot whose
+            # copy is still in flight.
+            C = self.starts.shape[1]
+            self._stage = [
+                {
+                    "starts": torch.empty(batch_size, C, dtype=torch.int32,
+                                          pin_memory=True),
+                    "paths": torch.empty(batch_size, C, dtype=torch.int32,
+                                         pin_memory=True),
+                    "ends": torch.empty(batch_size, C, dtype=torch.int32,
+                                        pin_memory=True),
+                    "label": torch.empty(batch_size, dtype=torch.int64,
+                                         pin_memory=True),
+                }
+                for _ in range(2)
+            ]
+            self._stage_ev = [torch.cuda.Event(), torch.cuda.Event()]
+            self._stage_used = [False, False]
 
     def __len__(self) -> int:
         n = self.starts.shape[0]
@@ -86,21 +102,35 @@ class BatchIterator:
                 yield host_batch(lo, min(lo + bs, n))
             return
 
-        # double-buffered async H2D on a copy stream
-        def to_device(hb: dict):
-            ev = torch.cuda.Event()
+        # double-buffered async H2D on a copy stream, staged through the
+        # persistent pinned buffers (a pinned source is what makes the
+        # non_blocking copy actually asynchronous)
+        def stage_and_copy(slot: int, lo: int, hi: int):
+            if self._stage_used[slot]:
+                self._stage_ev[slot].synchronize()  # copy out of this slot done
+            self._stage_used[slot] = True
+            idx = order[lo:hi]
+            k = hi - lo
+            st = self._stage[slot]
+            torch.index_select(self.starts, 0, idx, out=st["starts"][:k])
+            torch.index_select(self.paths, 0, idx, out=st["paths"][:k])
+            torch.index_select(self.ends, 0, idx, out=st["ends"][:k])
+            torch.index_select(self.labels, 0, idx, out=st["label"][:k])
+            ev = self._stage_ev[slot]
             with torch.cuda.stream(self._copy_stream):
                 db = {
-                    k: (v.to(self.device, non_blocking=True) if k != "id" else v)
-                    for k, v in hb.items()
+                    "id": self.ids[idx],
+                    "starts": st["starts"][:k].to(self.device, non_blocking=True),
+                    "paths": st["paths"][:k].to(self.device, non_blocking=True),
+                    "ends": st["ends"][:k].to(self.device, non_blocking=True),
+                    "label": st["label"][:k].to(self.device, non_blocking=True),
                 }
                 ev.record(self._copy_stream)
             return db, ev
 
         pending = None
-        for lo in range(0, n, bs):
-            hb = host_batch(lo, min(lo + bs, n))
-            nxt = to_device(hb)
+        for i, lo in enumerate(range(0, n, bs)):
+            nxt = stage_and_copy(i % 2, lo, min(lo + bs, n))
             if pending is not None:
                 db, ev = pending
                 torch.cuda.current_stream(self.device).wait_event(ev)

@@ -220,13 +220,24 @@ class Trainer:
         return 1.0 - f1
 
     # ------------------------------------------------------------------
-    def _make_loader(self, train: bool, epoch: int) -> BatchIterator:
+    def _make_loader(self, train: bool, epoch: int,
+                     keep_pending: bool = False) -> BatchIterator:
         if train:
-            if self._next_train is not None and self._next_train[0] == epoch:
-                data = self._next_train[1].result()
+            pending = self._next_train
+            if pending is not None and pending[0] == epoch:
+                data = pending[1].result()
+                self._next_train = None
             else:
+                if pending is not None and not keep_pending:
+                    # await before replacing: the orphaned background build
+                    # writes the same (tag, parity) pinned pool a synchronous
+                    # rebuild of pending[0] would also write
+                    pending[1].result()
+                    self._next_train = None
+                # keep_pending: the export path rebuilds THIS epoch (parity
+                # epoch%2) while the pending build writes parity (epoch+1)%2
+                # — disjoint pools, safe to run concurrently
                 data = self.builder.refresh_train_dataset(epoch)
-            self._next_train = None
         else:
             data = self.builder.refresh_test_dataset(epoch)
         return BatchIterator(
@@ -285,7 +296,7 @@ class Trainer:
         all-reduced; exact-match labels are gathered to rank 0."""
         model = self.model
         model.eval()
-        test_loss = 0.0
+        loss_acc = torch.zeros((), dtype=torch.float64, device=self.device)
         expected = []
         actual = []
         with torch.no_grad():
@@ -296,12 +307,15 @@ class Trainer:
                 label = batch["label"].to(self.device)
                 outputs, _, _ = model(starts, paths, ends, label)
                 loss = model.loss(outputs, label, self.class_weight)
-                test_loss += float(loss.item())
-                _, preds = torch.max(outputs.float(), dim=1)
-                expected.extend(label.cpu().tolist())
-                actual.extend(preds.cpu().tolist())
+                # deferred accumulation — no per-batch .item()/.cpu() sync
+                loss_acc += loss.detach().double()
+                expected.append(label)
+                actual.append(torch.argmax(outputs.float(), dim=1))
 
-        test_loss = all_reduce_sum_scalar(test_loss, self.ctx)
+        if expected:
+            expected = torch.cat(expected).cpu().tolist()
+            actual = torch.cat(actual).cpu().tolist()
+        test_loss = all_reduce_sum_scalar(float(loss_acc.item()), self.ctx)
         method = self.cfg.eval_method
         if method == "subtoken":
             m, ec, ac = M.subtoken_match_counts(expected, actual, self.reader.label_vocab)
@@ -345,7 +359,8 @@ class Trainer:
         write_vector_header(
             cfg.vectors_path, len(self.reader.items), self.option.encode_size
         )
-        train_loader = self._make_loader(train=True, epoch=epoch)
+        train_loader = self._make_loader(train=True, epoch=epoch,
+                                         keep_pending=True)
         test_loader = self._make_loader(train=False, epoch=epoch)
         # hipGraph-captured forward for the batched export (config 5)
         from .infer import graphed_export_forward

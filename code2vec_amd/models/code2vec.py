@@ -10,7 +10,8 @@ The HIP backend stores parameters in MI355X-native padded layouts:
 - combiner weight bf16 [KP, EP] (K-major B-operand layout for the MFMA
   GEMM; KP = 2*TS+PS, EP = round_up(E, 32)),
 - output weight bf16 [L, EP]; LN/attention/bias params fp32.
-Pad regions are zero and stay zero through training (tests/test_padding.py).
+Pad regions are zero and stay zero through training (padding checks in
+tests/test_model_gpu.py).
 """
 
 from __future__ import annotations

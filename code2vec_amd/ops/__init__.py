@@ -12,7 +12,7 @@ Layout conventions (all HIP kernels assume these):
 - the combiner input is packed [B*C, KP] with segment offsets
   [0, TS, TS+PS] for (start-, path-, end-embeddings); KP = 2*TS + PS,
 - pad regions are identically zero in parameters, activations and grads
-  (asserted by tests/test_padding.py).
+  (asserted by the padding checks in tests/test_model_gpu.py).
 """
 
 from __future__ import annotations

@@ -109,7 +109,7 @@ def test_ddp2_matches_single_process_large_batch(tmp_path, world):
         assert torch.allclose(p_ddp, p_ref, atol=1e-5), name
 
 
-def _worker_trainer(rank, world, port, tmpdir, result_file):
+def _worker_trainer(rank, world, port, tmpdir, result_file, n_methods=40):
     """Full Trainer epoch under gloo world=2 (sharded data, reduced metrics)."""
     import torch.distributed as dist
 
@@ -128,7 +128,7 @@ def _worker_trainer(rank, world, port, tmpdir, result_file):
 
     files = write_synthetic_corpus(
         os.path.join(tmpdir, "data"),
-        SyntheticSpec(n_methods=40, n_terminals=60, n_paths=50,
+        SyntheticSpec(n_methods=n_methods, n_terminals=60, n_paths=50,
                       max_contexts=16, seed=5),
     )
     reader = CorpusReader(files["corpus_path"], files["path_idx_path"],
@@ -166,6 +166,27 @@ def test_trainer_world2_runs(tmp_path):
     procs = [
         ctx.Process(target=_worker_trainer,
                     args=(r, 2, 29532, str(tmp_path), result_file))
+        for r in range(2)
+    ]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    obj = float(open(result_file).read())
+    assert 0.0 <= obj <= 1.0
+
+
+@pytest.mark.timeout(300)
+def test_trainer_world2_uneven_dataset(tmp_path):
+    """n_methods=41 -> 33 train items, not divisible by world*batch: without
+    equal-size shards one rank runs an extra step and its collectives
+    deadlock (the ADVICE.md stride-sharding finding).  Must complete."""
+    ctx = mp.get_context("spawn")
+    result_file = str(tmp_path / "obj.txt")
+    procs = [
+        ctx.Process(target=_worker_trainer,
+                    args=(r, 2, 29537, str(tmp_path), result_file, 41))
         for r in range(2)
     ]
     for pr in procs:

@@ -110,9 +110,23 @@ def test_builder_rank_sharding(tiny_corpus):
         d = b.refresh_train_dataset(epoch=0)
         shards.append(set(d.ids))
     assert shards[0].isdisjoint(shards[1])
+    # equal shard sizes => equal per-rank step counts (collective lockstep);
+    # the dropped tail is < world_size items
+    assert len(shards[0]) == len(shards[1])
     b_all = DatasetBuilder(r, opt, seed=5)
     d_all = b_all.refresh_train_dataset(epoch=0)
-    assert shards[0] | shards[1] == set(d_all.ids)
+    all_ids = set(d_all.ids)
+    union = shards[0] | shards[1]
+    assert union <= all_ids
+    assert len(all_ids) - len(union) < 2
+    # the epoch-seeded permutation rotates which items land in the dropped
+    # tail, so across epochs everything is seen
+    seen = set(union)
+    for epoch in range(1, 6):
+        for rank in range(2):
+            b = DatasetBuilder(r, opt, seed=5, rank=rank, world_size=2)
+            seen |= set(b.refresh_train_dataset(epoch=epoch).ids)
+    assert seen == all_ids
 
 
 def test_builder_variable_task(tiny_corpus):
@@ -345,3 +359,36 @@ def test_native_parser_matches_python_edge_corpora(tmp_path, variant):
         assert a.aliases == b.aliases
         assert np.array_equal(np.asarray(a.path_contexts),
                               np.asarray(b.path_contexts))
+
+
+def test_reader_rejects_out_of_range_indices(tmp_path, tiny_corpus):
+    """Out-of-range corpus indices fail loudly at load time (the HIP
+    gather never bounds-checks; the reference fails in nn.Embedding)."""
+    bad = tmp_path / "bad_corpus.txt"
+    with open(tiny_corpus["corpus_path"]) as f:
+        text = f.read()
+    # append a record whose path index is far beyond the path vocab
+    text += "\n#9999\nlabel:badMethod\npaths:\n2\t999999\t3\n\n"
+    bad.write_text(text)
+    with pytest.raises(ValueError, match="out-of-range"):
+        CorpusReader(str(bad), tiny_corpus["path_idx_path"],
+                     tiny_corpus["terminal_idx_path"])
+    # and a bad terminal index (after the +1 question shift)
+    bad2 = tmp_path / "bad_corpus2.txt"
+    bad2.write_text(text.replace("2\t999999\t3", "888888\t1\t3"))
+    with pytest.raises(ValueError, match="out-of-range"):
+        CorpusReader(str(bad2), tiny_corpus["path_idx_path"],
+                     tiny_corpus["terminal_idx_path"])
+
+
+def test_variable_task_equal_rows_across_ranks(tiny_corpus):
+    """With the variable task on, per-rank ROW counts (not just item
+    counts) must be equal, or DP ranks run different step counts."""
+    r = make_reader(tiny_corpus, infer_method=True, infer_variable=True)
+    opt = make_option(r)
+    for epoch in range(3):
+        sizes = []
+        for rank in range(2):
+            b = DatasetBuilder(r, opt, seed=11, rank=rank, world_size=2)
+            sizes.append(len(b.refresh_train_dataset(epoch=epoch)))
+        assert sizes[0] == sizes[1], sizes
