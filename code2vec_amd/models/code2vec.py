@@ -215,12 +215,33 @@ class Code2VecHIP(nn.Module):
                 self.cos_m, self.sin_m, opt.inverse_temp,
             )
         else:
-            outputs = Fn.OutputHead.apply(
-                cv.to(torch.bfloat16), self.output_weight, self.output_bias
-            )
+            cvb = cv.to(torch.bfloat16)
+            if Fn.fused_head_loss_supported(cvb, self.output_weight,
+                                            self.training):
+                # fused head+loss path: logits are computed WITHOUT autograd
+                # tracking; loss() builds the single FusedHeadLoss node whose
+                # backward recomputes G (dlogits never materialized).  The
+                # returned outputs tensor itself carries no grad — training
+                # flows through model.loss(), which is the only consumer
+                # (reference main.py:172-174).
+                outputs = Fn.head_logits_with_stats(
+                    cvb.detach(), self.output_weight.detach(),
+                    self.output_bias.detach())
+                outputs._c2v_fused_head = (cvb, self.output_weight,
+                                           self.output_bias)
+            else:
+                outputs = Fn.OutputHead.apply(
+                    cvb, self.output_weight, self.output_bias
+                )
         return outputs, cv[:, : self.E], attn
 
     def loss(self, outputs, label, class_weight):
+        fused = getattr(outputs, "_c2v_fused_head", None)
+        if fused is not None:
+            cvb, w, bias = fused
+            del outputs._c2v_fused_head
+            return Fn.FusedHeadLoss.apply(
+                outputs.contiguous(), cvb, w, bias, label, class_weight)
         return Fn.FusedLogSoftmaxNLL.apply(
             outputs.contiguous(), label, class_weight)
 

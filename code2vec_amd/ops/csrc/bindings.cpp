@@ -65,6 +65,12 @@ void launch_dgrad2(const void*, const void*, void*, long, long, hipStream_t);
 void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
 void launch_slab_sum_f32(const float*, float*, int, int, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
+void launch_head_bwd_dw(const void*, const void*, const float*, const long*,
+                        const float*, const float*, const float*, void*,
+                        float*, long, long, hipStream_t);
+void launch_head_bwd_dcv(const void*, const void*, const float*, const long*,
+                         const float*, const float*, const float*, float*,
+                         long, long, int, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
                      float*, long, long, int, hipStream_t);
 void launch_lsm_finalize(const void*, const float*, const float*, const long*,
@@ -382,6 +388,56 @@ void head_dgrad(torch::Tensor dlogits, torch::Tensor wt,
                     partials.data_ptr<float>(), B, L, cur_stream());
 }
 
+// Fused head+loss backward kernel 1: dw[L, 128] bf16 + dbias[L] f32 from
+// (logits, lse) with G recomputed in-kernel; cvt = cv transposed [128, B].
+void head_bwd_dw(torch::Tensor logits, torch::Tensor cvt, torch::Tensor lse,
+                 torch::Tensor label, torch::Tensor weight, torch::Tensor acc,
+                 torch::Tensor gscale, torch::Tensor dw, torch::Tensor dbias) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_CONTIG(cvt); CHK_DT(cvt, torch::kBFloat16);
+  CHK_DT(lse, torch::kFloat32); CHK_DT(label, torch::kInt64);
+  CHK_CONTIG(dw); CHK_DT(dw, torch::kBFloat16);
+  CHK_CONTIG(dbias); CHK_DT(dbias, torch::kFloat32);
+  const long B = logits.size(0), L = logits.size(1);
+  TORCH_CHECK(cvt.size(0) == 128 && cvt.size(1) == B, "cvt must be [128, B]");
+  TORCH_CHECK(L % 8 == 0 && B % 8 == 0, "head_bwd_dw shape gates");
+  TORCH_CHECK(dw.size(0) == L && dw.size(1) == 128 && dbias.numel() == L,
+              "head_bwd_dw output shapes");
+  const float* wp = weight.defined() && weight.numel() > 0
+                        ? weight.data_ptr<float>()
+                        : nullptr;
+  launch_head_bwd_dw(logits.data_ptr(), cvt.data_ptr(),
+                     lse.data_ptr<float>(), label.data_ptr<long>(), wp,
+                     acc.data_ptr<float>(), gscale.data_ptr<float>(),
+                     dw.data_ptr(), dbias.data_ptr<float>(), B, L,
+                     cur_stream());
+}
+
+// Fused head+loss backward kernel 2: dcv split-K partials
+// [ceil(L/chunk), B, 128] f32 with G recomputed; wt = W transposed [128, L].
+void head_bwd_dcv(torch::Tensor logits, torch::Tensor wt, torch::Tensor lse,
+                  torch::Tensor label, torch::Tensor weight,
+                  torch::Tensor acc, torch::Tensor gscale,
+                  torch::Tensor partials, long chunk) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
+  CHK_DT(lse, torch::kFloat32); CHK_DT(label, torch::kInt64);
+  CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
+  const long B = logits.size(0), L = logits.size(1);
+  TORCH_CHECK(wt.size(0) == 128 && wt.size(1) == L, "wt must be [128, L]");
+  TORCH_CHECK(L % 8 == 0 && chunk % 128 == 0, "head_bwd_dcv shape gates");
+  TORCH_CHECK(partials.numel() == (L + chunk - 1) / chunk * B * 128,
+              "head_bwd_dcv partials shape");
+  const float* wp = weight.defined() && weight.numel() > 0
+                        ? weight.data_ptr<float>()
+                        : nullptr;
+  launch_head_bwd_dcv(logits.data_ptr(), wt.data_ptr(),
+                      lse.data_ptr<float>(), label.data_ptr<long>(), wp,
+                      acc.data_ptr<float>(), gscale.data_ptr<float>(),
+                      partials.data_ptr<float>(), B, L, (int)chunk,
+                      cur_stream());
+}
+
 void transpose_w(torch::Tensor w, torch::Tensor wt) {
   CHK_CUDA(w); CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
   CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
@@ -485,6 +541,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
+  m.def("head_bwd_dw", &head_bwd_dw);
+  m.def("head_bwd_dcv", &head_bwd_dcv);
   m.def("transpose_w", &transpose_w);
   m.def("dgrad2", &dgrad2);
   m.def("slab_sum_bf16", &slab_sum_bf16);

@@ -39,6 +39,10 @@ _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
 # combiner dgrad through dgrad2.hip (C2V_DGRAD2=0 -> rocBLAS)
 _DGRAD2 = os.environ.get("C2V_DGRAD2", "1") == "1"
+# fused head+loss backward (recompute-G; dlogits never materialized).
+# C2V_FUSED_HEAD=0 falls back to the unfused OutputHead+FusedLogSoftmaxNLL
+# chain (kept for A/B and as the general-shape path).
+FUSED_HEAD = os.environ.get("C2V_FUSED_HEAD", "1") == "1"
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
 # Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
@@ -486,6 +490,96 @@ class FusedLogSoftmaxNLL(torch.autograd.Function):
             logits, label, weight, lse, acc, dloss.contiguous().float(), dlogits
         )
         return dlogits, None, None
+
+
+def head_logits_with_stats(cv_bf16, w, bias):
+    """Forward logits for the FUSED head+loss path (no autograd tracking —
+    gradients flow through FusedHeadLoss instead).  Uses the custom MFMA
+    head_fwd (+ online-softmax stats epilogue) when gated on, else the
+    hipBLASLt linear; either way the result carries what the fused loss
+    needs stashed as tensor attributes."""
+    B, EP = cv_bf16.shape
+    L = w.shape[0]
+    with torch.no_grad():
+        if _HEAD_FWD and EP % 32 == 0 and L <= _HEAD_FWD_MAXL:
+            logits = torch.empty(B, L, dtype=torch.bfloat16,
+                                 device=cv_bf16.device)
+            gx = (L + 255) // 256
+            pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
+            ps = torch.empty_like(pm)
+            ext().head_fwd(cv_bf16, w, bias.float(), logits, pm, ps)
+            logits._c2v_lsm_partials = (pm, ps)
+        else:
+            logits = torch.nn.functional.linear(
+                cv_bf16, w, bias.to(torch.bfloat16))
+    return logits
+
+
+class FusedHeadLoss(torch.autograd.Function):
+    """K10+K12 collapsed into one autograd node (recompute-G backward).
+
+    Forward: merges the head epilogue's online-softmax partials into
+    (lse, acc) — the [B, L] logits are never re-read in full when the
+    custom head forward ran.  Backward: dW/dbias (head_bwd_dw) and dcv
+    (head_bwd_dcv) recompute G[b,l] = coef_b*(exp(logit-lse_b) - [l==y_b])
+    from logits+lse in registers; the dlogits tensor of the unfused chain
+    (a 61 MB write + 3x 61 MB reads at top11, 534 MB x4 at java-large) is
+    never materialized.  Reference math: main.py:251-264 composed with
+    model/model.py:83.
+
+    ``logits`` is a NON-differentiable input (computed under no_grad by
+    head_logits_with_stats); gradients flow to cv/w/bias directly.
+    """
+
+    @staticmethod
+    def forward(ctx, logits, cv_bf16, w, bias, label, weight):
+        B, L = logits.shape
+        lse = torch.empty(B, dtype=torch.float32, device=logits.device)
+        acc = torch.zeros(2, dtype=torch.float32, device=logits.device)
+        partials = getattr(logits, "_c2v_lsm_partials", None)
+        if partials is not None:
+            pm, ps = partials
+            del logits._c2v_lsm_partials
+            ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight,
+                                          lse, acc)
+        else:
+            ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc)
+        loss = acc[0] / acc[1]
+        ctx.save_for_backward(logits, cv_bf16, w, label, weight, lse, acc)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, cv, w, label, weight, lse, acc = ctx.saved_tensors
+        B, L = logits.shape
+        dev = w.device
+        g = dloss.contiguous().float().view(1)
+        # dW + dbias (label-major): cv transposed once (256 KB, L2-resident)
+        cvt = _scratch_bf16("head_cvt", (128, B), dev)
+        ext().transpose_w(cv.contiguous(), cvt)
+        dw = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
+        dbias = torch.empty(L, dtype=torch.float32, device=dev)
+        ext().head_bwd_dw(logits, cvt, lse, label, weight, acc, g, dw, dbias)
+        # dcv (batch-major split-K): label chunk sized so the fp32 partial
+        # slabs stay ~30 MB at any L
+        chunk = 512 if L <= 65536 else 4096
+        split = (L + chunk - 1) // chunk
+        wt = _scratch_bf16("head_wt", (128, L), dev)
+        ext().transpose_w(w, wt)
+        partials = _scratch_f32("head_fused_dcv", (split, B, 128), dev)
+        ext().head_bwd_dcv(logits, wt, lse, label, weight, acc, g,
+                           partials, chunk)
+        dcv = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
+        ext().slab_sum_bf16(partials, dcv)
+        return None, dcv, dw, dbias, None, None
+
+
+def fused_head_loss_supported(cv_bf16, w, training: bool) -> bool:
+    """Shape/mode gates for the fused head+loss path."""
+    B, EP = cv_bf16.shape
+    L = w.shape[0]
+    return (FUSED_HEAD and training and torch.is_grad_enabled()
+            and cv_bf16.is_cuda and EP == 128 and L % 8 == 0 and B % 8 == 0)
 
 
 def adam_step(

@@ -223,6 +223,29 @@ class TestAttention:
         assert torch.allclose(attn, torch.full_like(attn, 1.0 / attn.shape[1]),
                               atol=1e-5)
 
+    def test_large_c_non_staged_fallback(self, dev):
+        """C = 1000: the tile no longer fits the LDS-staged path, so the
+        non-staged large-C fallback template runs (attention.hip) — the
+        long-context tunable of SURVEY §5.7.  Forward AND backward."""
+        from code2vec_amd.ops.functional import AttentionPool
+
+        ccv, a, starts, E, EP = self._setup(dev, B=6, C=1000, seed=9)
+        ch = ccv.detach().clone().requires_grad_(True)
+        ah = a.detach().clone().requires_grad_(True)
+        cv, attn = AttentionPool.apply(ch, ah, starts, E)
+        mask = (starts > 0).float()
+        cr = ccv.float().detach().requires_grad_(True)
+        ar = a.detach().clone().requires_grad_(True)
+        cv_ref, attn_ref = R.attention_code_vector(cr, ar, mask)
+        assert relerr(attn, attn_ref) < 2e-2
+        assert relerr(cv[:, :E], cv_ref[:, :E]) < 2e-2
+        dcv = torch.randn_like(cv) * 0.1
+        dcv[:, E:] = 0
+        cv.backward(dcv)
+        cv_ref.backward(dcv)
+        assert relerr(ch.grad.float(), cr.grad.to(torch.bfloat16).float()) < 4e-2
+        assert relerr(ah.grad[:E], ar.grad[:E]) < 4e-2
+
     def test_bwd(self, dev):
         from code2vec_amd.ops.functional import AttentionPool
 
@@ -506,3 +529,114 @@ class TestScatterProperty:
         err_p = (dpath.float() - ref_p).norm() / (ref_p.norm() + 1e-9)
         assert float(err_t) < 2e-2, float(err_t)
         assert float(err_p) < 2e-2, float(err_p)
+
+
+# ---------------------------------------------------------------------------
+class TestFusedHeadLoss:
+    """Fused head+loss backward (recompute-G; head_bwd.hip) vs the fp32
+    oracle and vs the unfused OutputHead+FusedLogSoftmaxNLL chain."""
+
+    def _setup(self, dev, B, L, EP=128, seed=23):
+        g = torch.Generator().manual_seed(seed)
+        cv = (torch.randn(B, EP, generator=g) * 0.5).to(dev, torch.bfloat16)
+        w = (torch.randn(L, EP, generator=g) * 0.1).to(dev, torch.bfloat16)
+        bias = torch.randn(L, generator=g).to(dev)
+        label = torch.randint(0, L, (B,), generator=g).to(dev)
+        weight = (torch.rand(L, generator=g) + 0.5).to(dev)
+        return cv, w, bias, label, weight
+
+    def _run_fused(self, cv, w, bias, label, weight, scale=1.0):
+        from code2vec_amd.ops import functional as Fn
+
+        cvh = cv.clone().requires_grad_(True)
+        wh = w.clone().requires_grad_(True)
+        bh = bias.clone().requires_grad_(True)
+        logits = Fn.head_logits_with_stats(cvh.detach(), wh.detach(),
+                                           bh.detach())
+        loss = Fn.FusedHeadLoss.apply(logits, cvh, wh, bh, label, weight)
+        (loss * scale).backward()
+        return loss, cvh.grad, wh.grad, bh.grad
+
+    @pytest.mark.parametrize("B,L", [(96, 7320), (64, 1992), (40, 4096)])
+    def test_matches_fp32_oracle(self, dev, B, L):
+        cv, w, bias, label, weight = self._setup(dev, B, L)
+        loss, dcv, dw, dbias = self._run_fused(cv, w, bias, label, weight)
+
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        ref = R.logsoftmax_nll(cvr @ wr.t() + br, label, weight)
+        ref.backward()
+        assert abs(float(loss) - float(ref)) / abs(float(ref)) < 1e-2
+        assert relerr(dcv.float(), cvr.grad) < 3e-2
+        assert relerr(dw.float(), wr.grad) < 3e-2
+        assert relerr(dbias.float(), br.grad) < 3e-2
+
+    def test_large_vocab_library_fwd_path(self, dev):
+        """L > C2V_HEAD_FWD_MAXL: forward takes the hipBLASLt path (no
+        stats partials -> full lsm fwd) and backward the 4096-label-chunk
+        split; still must match the oracle."""
+        B, L = 32, 70016
+        cv, w, bias, label, weight = self._setup(dev, B, L, seed=31)
+        loss, dcv, dw, dbias = self._run_fused(cv, w, bias, label, weight)
+
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        ref = R.logsoftmax_nll(cvr @ wr.t() + br, label, weight)
+        ref.backward()
+        assert abs(float(loss) - float(ref)) / abs(float(ref)) < 1e-2
+        assert relerr(dcv.float(), cvr.grad) < 3e-2
+        assert relerr(dw.float(), wr.grad) < 3e-2
+        assert relerr(dbias.float(), br.grad) < 3e-2
+
+    def test_matches_unfused_chain(self, dev):
+        """Fused backward vs the unfused OutputHead+FusedLogSoftmaxNLL
+        chain on identical inputs (both bf16 paths -> tight tolerance);
+        non-1 upstream gradient exercises the gscale plumbing."""
+        from code2vec_amd.ops.functional import (FusedLogSoftmaxNLL,
+                                                 OutputHead)
+
+        B, L = 64, 2048
+        cv, w, bias, label, weight = self._setup(dev, B, L, seed=41)
+        loss_f, dcv_f, dw_f, db_f = self._run_fused(
+            cv, w, bias, label, weight, scale=2.5)
+
+        cvh = cv.clone().requires_grad_(True)
+        wh = w.clone().requires_grad_(True)
+        bh = bias.clone().requires_grad_(True)
+        loss_u = FusedLogSoftmaxNLL.apply(
+            OutputHead.apply(cvh, wh, bh), label, weight)
+        (loss_u * 2.5).backward()
+        assert abs(float(loss_f) - float(loss_u)) <= 1e-5 * abs(float(loss_u))
+        assert relerr(dcv_f.float(), cvh.grad.float()) < 1e-2
+        assert relerr(dw_f.float(), wh.grad.float()) < 1e-2
+        assert relerr(db_f.float(), bh.grad.float()) < 1e-2
+
+    def test_model_takes_fused_path_in_training(self, dev):
+        """Code2VecHIP training forward stashes the fused-head context and
+        loss() consumes it (the hot path actually runs these kernels)."""
+        from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+        from code2vec_amd.utils.options import Option
+
+        opt = Option(terminal_count=500, path_count=400, label_count=2000,
+                     max_path_length=16, terminal_embed_size=100,
+                     path_embed_size=100, encode_size=100, dropout_prob=0.0)
+        g = torch.Generator().manual_seed(3)
+        m = Code2VecHIP(opt, init_logical_params(opt, g), device=dev).train()
+        s = torch.randint(1, 500, (16, 16), dtype=torch.int32, device=dev)
+        p = torch.randint(1, 400, (16, 16), dtype=torch.int32, device=dev)
+        e = torch.randint(1, 500, (16, 16), dtype=torch.int32, device=dev)
+        y = torch.randint(0, 2000, (16,), device=dev)
+        out, _, _ = m(s, p, e, y)
+        assert hasattr(out, "_c2v_fused_head")
+        loss = m.loss(out, y, torch.ones(2000, device=dev))
+        loss.backward()
+        assert m.output_weight.grad is not None
+        assert m.output_bias.grad is not None
+        assert m.terminal_embedding.grad is not None
+        # eval path: no stash, plain logits
+        m.eval()
+        with torch.no_grad():
+            out_e, _, _ = m(s, p, e, y)
+        assert not hasattr(out_e, "_c2v_fused_head")
