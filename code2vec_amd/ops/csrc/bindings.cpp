@@ -65,11 +65,13 @@ void launch_dgrad2(const void*, const void*, void*, long, long, hipStream_t);
 void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
 void launch_slab_sum_f32(const float*, float*, int, int, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
-void launch_head_bwd_dw(const void*, const void*, const float*, const long*,
-                        const float*, const float*, const float*, void*,
+void launch_head_bwd_prep(const long*, const float*, const float*,
+                          const float*, const float*, float*, long,
+                          hipStream_t);
+void launch_swizzle_cv(const void*, void*, long, hipStream_t);
+void launch_head_bwd_dw(const void*, const void*, const float*, void*,
                         float*, long, long, hipStream_t);
-void launch_head_bwd_dcv(const void*, const void*, const float*, const long*,
-                         const float*, const float*, const float*, float*,
+void launch_head_bwd_dcv(const void*, const void*, const float*, float*,
                          long, long, int, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
                      float*, long, long, int, hipStream_t);
@@ -388,52 +390,72 @@ void head_dgrad(torch::Tensor dlogits, torch::Tensor wt,
                     partials.data_ptr<float>(), B, L, cur_stream());
 }
 
-// Fused head+loss backward kernel 1: dw[L, 128] bf16 + dbias[L] f32 from
-// (logits, lse) with G recomputed in-kernel; cvt = cv transposed [128, B].
-void head_bwd_dw(torch::Tensor logits, torch::Tensor cvt, torch::Tensor lse,
-                 torch::Tensor label, torch::Tensor weight, torch::Tensor acc,
-                 torch::Tensor gscale, torch::Tensor dw, torch::Tensor dbias) {
-  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
-  CHK_CONTIG(cvt); CHK_DT(cvt, torch::kBFloat16);
-  CHK_DT(lse, torch::kFloat32); CHK_DT(label, torch::kInt64);
-  CHK_CONTIG(dw); CHK_DT(dw, torch::kBFloat16);
-  CHK_CONTIG(dbias); CHK_DT(dbias, torch::kFloat32);
-  const long B = logits.size(0), L = logits.size(1);
-  TORCH_CHECK(cvt.size(0) == 128 && cvt.size(1) == B, "cvt must be [128, B]");
-  TORCH_CHECK(L % 8 == 0 && B % 8 == 0, "head_bwd_dw shape gates");
-  TORCH_CHECK(dw.size(0) == L && dw.size(1) == 128 && dbias.numel() == L,
-              "head_bwd_dw output shapes");
+// Fused head+loss backward prep: coef_lse[B, 4] = (coef_b, lse_b, y_b, 0).
+void head_bwd_prep(torch::Tensor label, torch::Tensor weight,
+                   torch::Tensor acc, torch::Tensor gscale, torch::Tensor lse,
+                   torch::Tensor coef_lse) {
+  CHK_CUDA(lse); CHK_DT(lse, torch::kFloat32); CHK_DT(label, torch::kInt64);
+  CHK_CONTIG(coef_lse); CHK_DT(coef_lse, torch::kFloat32);
+  const long B = label.numel();
+  TORCH_CHECK(coef_lse.numel() == 4 * B, "coef_lse must be [B, 4]");
   const float* wp = weight.defined() && weight.numel() > 0
                         ? weight.data_ptr<float>()
                         : nullptr;
-  launch_head_bwd_dw(logits.data_ptr(), cvt.data_ptr(),
-                     lse.data_ptr<float>(), label.data_ptr<long>(), wp,
-                     acc.data_ptr<float>(), gscale.data_ptr<float>(),
-                     dw.data_ptr(), dbias.data_ptr<float>(), B, L,
-                     cur_stream());
+  launch_head_bwd_prep(label.data_ptr<long>(), wp, acc.data_ptr<float>(),
+                       gscale.data_ptr<float>(), lse.data_ptr<float>(),
+                       coef_lse.data_ptr<float>(), B, cur_stream());
+}
+
+// cv [B, 128] -> B-fragment image [ceil(B/64)*2, 8, 64, 8].
+void swizzle_cv(torch::Tensor cv, torch::Tensor cvimg) {
+  CHK_CUDA(cv); CHK_CONTIG(cv); CHK_DT(cv, torch::kBFloat16);
+  CHK_CONTIG(cvimg); CHK_DT(cvimg, torch::kBFloat16);
+  const long B = cv.size(0);
+  TORCH_CHECK(cv.size(1) == 128, "cv must be [B, 128]");
+  TORCH_CHECK(cvimg.numel() == (B + 63) / 64 * 2 * 4096, "cvimg shape");
+  launch_swizzle_cv(cv.data_ptr(), cvimg.data_ptr(), B, cur_stream());
+}
+
+// Fused head+loss backward kernel 1: dw[L, 128] bf16 + dbias[L] f32 from
+// (logits, coef_lse) with G recomputed in-kernel; cvimg = swizzle_cv(cv).
+void head_bwd_dw(torch::Tensor logits, torch::Tensor cvimg,
+                 torch::Tensor coef_lse, torch::Tensor dw,
+                 torch::Tensor dbias) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_CONTIG(cvimg); CHK_DT(cvimg, torch::kBFloat16);
+  CHK_CONTIG(coef_lse); CHK_DT(coef_lse, torch::kFloat32);
+  CHK_CONTIG(dw); CHK_DT(dw, torch::kBFloat16);
+  CHK_CONTIG(dbias); CHK_DT(dbias, torch::kFloat32);
+  const long B = logits.size(0), L = logits.size(1);
+  TORCH_CHECK(cvimg.numel() == (B + 63) / 64 * 2 * 4096, "cvimg shape");
+  TORCH_CHECK(L % 8 == 0 && B % 8 == 0, "head_bwd_dw shape gates");
+  TORCH_CHECK(L < (1L << 24), "label index carried as f32 needs L < 2^24");
+  TORCH_CHECK(coef_lse.numel() == 4 * B, "coef_lse must be [B, 4]");
+  TORCH_CHECK(dw.size(0) == L && dw.size(1) == 128 && dbias.numel() == L,
+              "head_bwd_dw output shapes");
+  launch_head_bwd_dw(logits.data_ptr(), cvimg.data_ptr(),
+                     coef_lse.data_ptr<float>(), dw.data_ptr(),
+                     dbias.data_ptr<float>(), B, L, cur_stream());
 }
 
 // Fused head+loss backward kernel 2: dcv split-K partials
-// [ceil(L/chunk), B, 128] f32 with G recomputed; wt = W transposed [128, L].
-void head_bwd_dcv(torch::Tensor logits, torch::Tensor wt, torch::Tensor lse,
-                  torch::Tensor label, torch::Tensor weight,
-                  torch::Tensor acc, torch::Tensor gscale,
-                  torch::Tensor partials, long chunk) {
+// [ceil(L/chunk), B, 128] f32 with G recomputed; wt = W^T [128, L].
+void head_bwd_dcv(torch::Tensor logits, torch::Tensor wt,
+                  torch::Tensor coef_lse, torch::Tensor partials,
+                  long chunk) {
   CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
   CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
-  CHK_DT(lse, torch::kFloat32); CHK_DT(label, torch::kInt64);
+  CHK_CONTIG(coef_lse); CHK_DT(coef_lse, torch::kFloat32);
   CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
   const long B = logits.size(0), L = logits.size(1);
   TORCH_CHECK(wt.size(0) == 128 && wt.size(1) == L, "wt must be [128, L]");
   TORCH_CHECK(L % 8 == 0 && chunk % 128 == 0, "head_bwd_dcv shape gates");
+  TORCH_CHECK(L < (1L << 24), "label index carried as f32 needs L < 2^24");
+  TORCH_CHECK(coef_lse.numel() == 4 * B, "coef_lse must be [B, 4]");
   TORCH_CHECK(partials.numel() == (L + chunk - 1) / chunk * B * 128,
               "head_bwd_dcv partials shape");
-  const float* wp = weight.defined() && weight.numel() > 0
-                        ? weight.data_ptr<float>()
-                        : nullptr;
   launch_head_bwd_dcv(logits.data_ptr(), wt.data_ptr(),
-                      lse.data_ptr<float>(), label.data_ptr<long>(), wp,
-                      acc.data_ptr<float>(), gscale.data_ptr<float>(),
+                      coef_lse.data_ptr<float>(),
                       partials.data_ptr<float>(), B, L, (int)chunk,
                       cur_stream());
 }
@@ -541,6 +563,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
+  m.def("head_bwd_prep", &head_bwd_prep);
+  m.def("swizzle_cv", &swizzle_cv);
   m.def("head_bwd_dw", &head_bwd_dw);
   m.def("head_bwd_dcv", &head_bwd_dcv);
   m.def("transpose_w", &transpose_w);

@@ -9,6 +9,11 @@
 // model/model.py:83; G is the gradient of the weighted-NLL log-softmax
 // composed with the linear head).
 //
+// A tiny prep kernel packs per-row (coef_b, lse_b, y_b) into one f32x4 so
+// the G recompute costs ONE 16-B load per staged logits chunk instead of
+// three scattered scalar loads (label i64 + weight gather + lse) per
+// thread per stage.  y is carried as a float: exact for L < 2^24.
+//
 // Kernel 1 (label-major): dW[L, EP] = G^T @ cv and dbias[L] = colsum(G).
 //   Block = 64 labels x EP(=128), K-loop over ALL batch rows in 64-row
 //   stages.  G tiles are computed at stage time from coalesced logits row
@@ -16,10 +21,13 @@
 //   wgrad.hip padded-image recipe — G^T fragment k-runs walk the batch
 //   dim, which is logits' slow axis).  cv is pre-transposed once
 //   ([EP, B], 256 KB — L2-resident) so B-fragment k-runs are contiguous
-//   global loads.  dbias falls out of the same pass: each thread keeps
-//   8 label-column partial sums of its staged G values, LDS-reduced at
-//   the end; every label belongs to exactly ONE block, so the store is
-//   a plain f32 write (no atomics).
+//   global loads.  All staging values live in named vector registers —
+//   NO lambdas, NO address-taken arrays (hipcc demotes those to scratch
+//   inside K-loops; PERF.md pathology #0 — the first version of this
+//   kernel measured 81 us instead of ~25 for exactly that reason).
+//   dbias falls out of the same pass: each thread keeps 8 label-column
+//   partial sums of its staged G values, LDS-reduced at the end; every
+//   label belongs to exactly ONE block, so the store is a plain f32 write.
 //
 // Kernel 2 (batch-major): dcv[B, EP] = G @ W — head_dgrad.hip's proven
 //   split-K structure with the A fragment recomputed from contiguous
@@ -28,30 +36,56 @@
 //   4096 at java-large scale so the fp32 partial-slab traffic stays ~33 MB
 //   instead of 268 MB.
 
+#include <cstdlib>
+
 #include "common.h"
 
 #define HB_LB 64    // labels per dW block
 #define HB_ROWS 64  // batch rows staged per K-iteration
 #define HB_NSTR (64 * 8 + 8)  // padded A-image stride (wgrad recipe)
 
+// coef_lse[b] = (coef_b, lse_b, (float)y_b, 0)
+__global__ __launch_bounds__(256) void head_bwd_prep_kernel(
+    const long* __restrict__ label, const float* __restrict__ weight,
+    const float* __restrict__ acc_ws, const float* __restrict__ gscale,
+    const float* __restrict__ lse, float* __restrict__ coef_lse, long B) {
+  const long b = (long)blockIdx.x * 256 + threadIdx.x;
+  if (b >= B) return;
+  const long y = label[b];
+  const float coef = gscale[0] * (weight ? weight[y] : 1.f) / acc_ws[1];
+  f32x4 v = {coef, lse[b], (float)y, 0.f};
+  *(f32x4*)(coef_lse + 4 * b) = v;
+}
+
+// variant: phase-isolation bitmask (C2V_HBDW_VARIANT, perf diagnosis):
+//   1 = skip dW/dbias global stores, 2 = skip cv (B-operand) staging loads,
+//   4 = skip logits staging loads, 8 = replace expf with identity,
+//   16 = replace the MFMA chain with a cheap accumulate
+//
+// The B operand (cv) is PRE-SWIZZLED ONCE into MFMA fragment-image
+// layout in global memory (swizzle_cv_kernel, 256 KB at B=1024): each
+// B-fragment read is then a fully contiguous 1-KB wave read that streams
+// from L2.  Two rejected designs, both variant-isolated on hardware:
+// (a) per-lane loads from a pre-transposed cvT — 16 lanes hit
+// 2-KB-strided addresses, camping on a few memory channels (~37 of
+// 69 us); (b) staging cv rows through LDS fragment images per stage —
+// 16 extra ds_write_b16 per thread per stage on the CU-shared LDS pipe
+// (~33 us of skeleton).
 __global__ __launch_bounds__(512) void head_bwd_dw_kernel(
-    const bf16* __restrict__ logits, const bf16* __restrict__ cvt,
-    const float* __restrict__ lse, const long* __restrict__ label,
-    const float* __restrict__ weight, const float* __restrict__ acc_ws,
-    const float* __restrict__ gscale, bf16* __restrict__ dw,
-    float* __restrict__ dbias, long B, long L) {
+    const bf16* __restrict__ logits, const bf16* __restrict__ cvimg,
+    const float* __restrict__ coef_lse, bf16* __restrict__ dw,
+    float* __restrict__ dbias, long B, long L, int variant) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const long l0 = (long)blockIdx.x * HB_LB;
-  // staging role: one bf16x8 logits chunk per thread per stage
+  // staging role A (logits->G): one bf16x8 chunk per thread per stage
   const int krow = threadIdx.x >> 3;        // 0..63: batch row within stage
   const int col8 = (threadIdx.x & 7) * 8;   // 0..56: label chunk start
-  const float inv_ws = gscale[0] / acc_ws[1];
 
   // A-fragment images (double-buffered) reused as the dbias reduction
   // scratch after the K-loop (lives separated by barriers)
   __shared__ union {
-    bf16 img[2][2][4][HB_NSTR];  // [buf][ksub][label tile][fragment image]
+    bf16 img[2][2][4][HB_NSTR];  // [buf][ksub][label tile][image]
     float red[HB_ROWS][HB_LB + 1];
   } sm;
   __shared__ float red2[8][HB_LB + 1];
@@ -61,89 +95,124 @@ __global__ __launch_bounds__(512) void head_bwd_dw_kernel(
   const int ksub = krow >> 5;       // which 32-k MFMA step of the stage
   const int kgrp = (krow >> 3) & 3; // 16-lane fragment group
   const int jslot = krow & 7;
+  const long lc = l0 + col8;
+  const bool lok = lc < L;          // L % 8 == 0: chunk fully in bounds
 
-  float db[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) db[j] = 0.f;
-
-  bf16 gv[8];
-  auto stage_load = [&](long mb) {
-    const long row = mb + krow;
-    const long lc = l0 + col8;
-    if (row < B && lc < L) {  // L % 8 == 0: chunk fully in bounds
-      const long y = label[row];
-      const float coef = inv_ws * (weight ? weight[y] : 1.f);
-      const float lseb = lse[row];
-      bf16 v[8];
-      *(uint4*)v = *(const uint4*)(logits + row * L + lc);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = coef * __expf(bf2f(v[j]) - lseb);
-        if (lc + j == y) g -= coef;
-        const bf16 gb = f2bf(g);
-        gv[j] = gb;
-        db[j] += bf2f(gb);  // dbias from the rounded value: bitwise parity
-                            // with colsum over a stored bf16 dlogits
-      }
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) gv[j] = f2bf(0.f);
-    }
-  };
-  auto stage_write = [&](int buf) {
-    bf16* dst = sm.img[buf][ksub][n];
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      dst[(base_l + j + kgrp * 16) * 8 + jslot] = gv[j];
-  };
-
-  f32x4 acc[4];
-#pragma unroll
-  for (int nt = 0; nt < 4; ++nt) acc[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  f32x4 db0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 db1 = {0.f, 0.f, 0.f, 0.f};
   const bf16x8 zero8 = {};
+
+  // named staging registers (macro-expanded; no address-taken arrays)
+  bf16x8 lv;          // staged logits chunk
+  f32x4 meta;         // (coef, lse, yf, -)
+  bool rok;
+
+#define DW_LOAD(mb)                                                        \
+  do {                                                                     \
+    const long row_ = (mb) + krow;                                         \
+    rok = row_ < B && lok && !(variant & 4);                               \
+    if (rok) {                                                             \
+      meta = *(const f32x4*)(coef_lse + 4 * row_);                         \
+      lv = *(const bf16x8*)(logits + row_ * L + lc);                       \
+    }                                                                      \
+  } while (0)
+
+#define DW_WRITE(buf)                                                      \
+  do {                                                                     \
+    bf16x8 gv = zero8;                                                     \
+    if (rok) {                                                             \
+      const float coef_ = meta[0];                                         \
+      const float lse_ = meta[1];                                          \
+      const long y_ = (long)meta[2];                                       \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
+        const float x_ = bf2f(lv[j]) - lse_;                               \
+        float g_ = coef_ * ((variant & 8) ? x_ : __expf(x_));              \
+        if (lc + j == y_) g_ -= coef_;                                     \
+        const bf16 gb_ = f2bf(g_);                                         \
+        gv[j] = gb_;                                                       \
+        if (j < 4) db0[j & 3] += bf2f(gb_);                                \
+        else db1[j & 3] += bf2f(gb_);                                      \
+      }                                                                    \
+    }                                                                      \
+    bf16* dst_ = sm.img[buf][ksub][n];                                     \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j)                          \
+        dst_[(base_l + j + kgrp * 16) * 8 + jslot] = gv[j];                \
+  } while (0)
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc3 = {0.f, 0.f, 0.f, 0.f};
 
   const int lt = wave >> 1;          // this wave's label tile
   const int nh = (wave & 1) * 64;    // this wave's EP half
+  const int ng = (wave & 1) * 4;     // first B-image tile of this wave
+  // cvimg layout: [k-chunk][8 nt][64 lanes][8] — per-fragment reads are
+  // contiguous 1-KB wave reads
+  const bf16* cbase = cvimg + (long)lane * 8;
 
-  stage_load(0);
-  stage_write(0);
+  DW_LOAD(0);
+  DW_WRITE(0);
   __syncthreads();
   int buf = 0;
   for (long mb = 0; mb < B; mb += HB_ROWS) {
-    if (mb + HB_ROWS < B) stage_load(mb + HB_ROWS);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&sm.img[buf][ks][lt][lane * 8];
-      const long k0 = mb + ks * 32 + (lane >> 4) * 8;
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        const long colv = nh + nt * 16 + (lane & 15);
-        const bf16x8 b =
-            (k0 < B) ? *(const bf16x8*)(cvt + colv * B + k0) : zero8;
-        acc[nt] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
-      }
+    const bool more = mb + HB_ROWS < B;
+    if (more) DW_LOAD(mb + HB_ROWS);
+    // B fragments stream from the global cv image (contiguous per wave);
+    // A fragments from the LDS G image
+    const bool bl = !(variant & 2);
+    const long kc0 = (mb >> 5) * 8 * 512;       // this stage's first chunk
+    const bf16x8 b00 = bl ? *(const bf16x8*)(cbase + kc0 + (ng + 0) * 512) : zero8;
+    const bf16x8 b01 = bl ? *(const bf16x8*)(cbase + kc0 + (ng + 1) * 512) : zero8;
+    const bf16x8 b02 = bl ? *(const bf16x8*)(cbase + kc0 + (ng + 2) * 512) : zero8;
+    const bf16x8 b03 = bl ? *(const bf16x8*)(cbase + kc0 + (ng + 3) * 512) : zero8;
+    const bf16x8 b10 = bl ? *(const bf16x8*)(cbase + kc0 + (8 + ng + 0) * 512) : zero8;
+    const bf16x8 b11 = bl ? *(const bf16x8*)(cbase + kc0 + (8 + ng + 1) * 512) : zero8;
+    const bf16x8 b12 = bl ? *(const bf16x8*)(cbase + kc0 + (8 + ng + 2) * 512) : zero8;
+    const bf16x8 b13 = bl ? *(const bf16x8*)(cbase + kc0 + (8 + ng + 3) * 512) : zero8;
+    const bf16x8 a0 = *(const bf16x8*)&sm.img[buf][0][lt][lane * 8];
+    const bf16x8 a1 = *(const bf16x8*)&sm.img[buf][1][lt][lane * 8];
+    if (!(variant & 16)) {
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b00, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b01, acc1, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b02, acc2, 0, 0, 0);
+      acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b03, acc3, 0, 0, 0);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b10, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b11, acc1, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b12, acc2, 0, 0, 0);
+      acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b13, acc3, 0, 0, 0);
+    } else {
+      acc0[0] += bf2f(b00[0]) + bf2f(a0[0]);
+      acc1[0] += bf2f(b01[0]) + bf2f(a1[0]);
+      acc2[0] += bf2f(b02[0]) + bf2f(b12[0]);
+      acc3[0] += bf2f(b03[0]) + bf2f(b10[0]) + bf2f(b11[0]) + bf2f(b13[0]);
     }
-    if (mb + HB_ROWS < B) stage_write(buf ^ 1);
+    if (more) DW_WRITE(buf ^ 1);
     buf ^= 1;
     __syncthreads();
   }
+#undef DW_LOAD
+#undef DW_WRITE
 
   // dW: each wave owns [16 labels x 64 EP] of the block's [64, 128] tile
-#pragma unroll
-  for (int nt = 0; nt < 4; ++nt) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const long lrow = l0 + lt * 16 + (lane >> 4) * 4 + r;
-      if (lrow < L)
-        dw[lrow * 128 + nh + nt * 16 + (lane & 15)] = f2bf(acc[nt][r]);
-    }
+#define DW_STORE(nt, accv)                                                 \
+  _Pragma("unroll") for (int r = 0; r < 4; ++r) {                          \
+    const long lrow_ = l0 + lt * 16 + (lane >> 4) * 4 + r;                 \
+    if (lrow_ < L && !(variant & 1))                                       \
+      dw[lrow_ * 128 + nh + nt * 16 + (lane & 15)] = f2bf(accv[r]);        \
   }
+  DW_STORE(0, acc0)
+  DW_STORE(1, acc1)
+  DW_STORE(2, acc2)
+  DW_STORE(3, acc3)
+#undef DW_STORE
 
   // dbias: reduce the per-thread 8-label partials.  Write layout
   // red[krow][label]: bank-conflict-free both ways (stride 65).
 #pragma unroll
-  for (int j = 0; j < 8; ++j) sm.red[krow][col8 + j] = db[j];
+  for (int j = 0; j < 4; ++j) sm.red[krow][col8 + j] = db0[j];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) sm.red[krow][col8 + 4 + j] = db1[j];
   __syncthreads();
   {
     const int lbl = threadIdx.x & 63;
@@ -154,7 +223,7 @@ __global__ __launch_bounds__(512) void head_bwd_dw_kernel(
     red2[threadIdx.x >> 6][lbl] = p;
   }
   __syncthreads();
-  if (threadIdx.x < 64 && l0 + threadIdx.x < L) {
+  if (threadIdx.x < 64 && l0 + threadIdx.x < L && !(variant & 1)) {
     float s = 0.f;
 #pragma unroll
     for (int q = 0; q < 8; ++q) s += red2[q][threadIdx.x];
@@ -162,12 +231,40 @@ __global__ __launch_bounds__(512) void head_bwd_dw_kernel(
   }
 }
 
+// cv [B, 128] row-major -> MFMA B-fragment image
+// [ceil(B/32)][8 nt][64 lanes][8]: entry [kc][nt][l][j] =
+// cv[kc*32 + (l>>4)*8 + j][nt*16 + (l&15)] (zero-padded past B).  One
+// 32-row tile per block staged through LDS; output writes are contiguous
+// 16 B per thread.
+__global__ __launch_bounds__(256) void swizzle_cv_kernel(
+    const bf16* __restrict__ cv, bf16* __restrict__ cvimg, long B) {
+  const long kc = blockIdx.x;
+  __shared__ bf16 tile[32][132];
+  const bf16x8 zero8 = {};
+  for (int t = threadIdx.x; t < 32 * 16; t += 256) {
+    const int r = t >> 4;
+    const int c = t & 15;
+    const long row = kc * 32 + r;
+    const bf16x8 v =
+        (row < B) ? *(const bf16x8*)(cv + row * 128 + c * 8) : zero8;
+    *(bf16x8*)&tile[r][c * 8] = v;
+  }
+  __syncthreads();
+  // 512 outputs of 16 B per tile: (nt, lane) pairs, 2 per thread
+  for (int t = threadIdx.x; t < 512; t += 256) {
+    const int nt = t >> 6;
+    const int l = t & 63;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tile[(l >> 4) * 8 + j][nt * 16 + (l & 15)];
+    *(bf16x8*)(cvimg + ((kc * 8 + nt) * 64 + (long)l) * 8) = v;
+  }
+}
+
 // dcv split-K partials: head_dgrad.hip's kernel with A = G recomputed.
 __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
     const bf16* __restrict__ logits, const bf16* __restrict__ wt,
-    const float* __restrict__ lse, const long* __restrict__ label,
-    const float* __restrict__ weight, const float* __restrict__ acc_ws,
-    const float* __restrict__ gscale, float* __restrict__ partials, long B,
+    const float* __restrict__ coef_lse, float* __restrict__ partials, long B,
     long L, int chunk, int GYB) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -185,9 +282,10 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
   float coef = 0.f, lseb = 0.f;
   long y = -1;
   if (aok) {
-    y = label[row];
-    coef = gscale[0] * (weight ? weight[y] : 1.f) / acc_ws[1];
-    lseb = lse[row];
+    const f32x4 meta = *(const f32x4*)(coef_lse + 4 * row);
+    coef = meta[0];
+    lseb = meta[1];
+    y = (long)meta[2];
   }
   const bf16* ap = logits + row * L + kj;
 
@@ -215,16 +313,13 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
       const long lc = l0 + kk * 32 + kj;
       bf16x8 a = zero8;
       if (aok && lc < L) {
-        bf16 v[8];
-        *(uint4*)v = *(const uint4*)(ap + l0 + kk * 32);
-        bf16 g[8];
+        const bf16x8 v = *(const bf16x8*)(ap + l0 + kk * 32);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float gg = coef * __expf(bf2f(v[j]) - lseb);
           if (lc + j == y) gg -= coef;
-          g[j] = f2bf(gg);
+          a[j] = f2bf(gg);
         }
-        a = *(const bf16x8*)g;
       }
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
@@ -251,26 +346,41 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
 
 extern "C" {
 
-void launch_head_bwd_dw(const void* logits, const void* cvt, const float* lse,
-                        const long* label, const float* weight,
-                        const float* acc_ws, const float* gscale, void* dw,
-                        float* dbias, long B, long L, hipStream_t stream) {
-  const int grid = (int)((L + HB_LB - 1) / HB_LB);
-  head_bwd_dw_kernel<<<grid, 512, 0, stream>>>(
-      (const bf16*)logits, (const bf16*)cvt, lse, label, weight, acc_ws,
-      gscale, (bf16*)dw, dbias, B, L);
+void launch_head_bwd_prep(const long* label, const float* weight,
+                          const float* acc_ws, const float* gscale,
+                          const float* lse, float* coef_lse, long B,
+                          hipStream_t stream) {
+  head_bwd_prep_kernel<<<(B + 255) / 256, 256, 0, stream>>>(
+      label, weight, acc_ws, gscale, lse, coef_lse, B);
 }
 
-void launch_head_bwd_dcv(const void* logits, const void* wt, const float* lse,
-                         const long* label, const float* weight,
-                         const float* acc_ws, const float* gscale,
-                         float* partials, long B, long L, int chunk,
-                         hipStream_t stream) {
+void launch_swizzle_cv(const void* cv, void* cvimg, long B,
+                       hipStream_t stream) {
+  // chunk count rounded to the dW kernel's 64-row stage granularity so
+  // tail-stage fragment reads always hit written (zero-padded) chunks
+  swizzle_cv_kernel<<<((B + 63) / 64) * 2, 256, 0, stream>>>(
+      (const bf16*)cv, (bf16*)cvimg, B);
+}
+
+void launch_head_bwd_dw(const void* logits, const void* cvimg,
+                        const float* coef_lse, void* dw, float* dbias,
+                        long B, long L, hipStream_t stream) {
+  const int grid = (int)((L + HB_LB - 1) / HB_LB);
+  const char* ve = getenv("C2V_HBDW_VARIANT");
+  const int variant = ve ? atoi(ve) : 0;
+  head_bwd_dw_kernel<<<grid, 512, 0, stream>>>(
+      (const bf16*)logits, (const bf16*)cvimg, coef_lse, (bf16*)dw, dbias, B,
+      L, variant);
+}
+
+void launch_head_bwd_dcv(const void* logits, const void* wt,
+                         const float* coef_lse, float* partials, long B,
+                         long L, int chunk, hipStream_t stream) {
   const int GYB = (int)((B + 127) / 128);
   const int split = (int)((L + chunk - 1) / chunk);
   head_bwd_dcv_kernel<<<GYB * split, 512, 0, stream>>>(
-      (const bf16*)logits, (const bf16*)wt, lse, label, weight, acc_ws,
-      gscale, partials, B, L, chunk, GYB);
+      (const bf16*)logits, (const bf16*)wt, coef_lse, partials, B, L, chunk,
+      GYB);
 }
 
 }  // extern "C"

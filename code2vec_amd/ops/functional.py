@@ -554,21 +554,28 @@ class FusedHeadLoss(torch.autograd.Function):
         B, L = logits.shape
         dev = w.device
         g = dloss.contiguous().float().view(1)
-        # dW + dbias (label-major): cv transposed once (256 KB, L2-resident)
-        cvt = _scratch_bf16("head_cvt", (128, B), dev)
-        ext().transpose_w(cv.contiguous(), cvt)
+        # per-row (coef, lse, y) packed once: ONE 16-B load per staged
+        # logits chunk in the kernels below
+        coef_lse = _scratch_f32("head_coef_lse", (B, 4), dev)
+        ext().head_bwd_prep(label, weight, acc, g, lse, coef_lse)
+        # dW + dbias (label-major): cv pre-swizzled once (256 KB) into the
+        # MFMA fragment-image layout so every B-fragment read in the dW
+        # kernel is a contiguous 1-KB wave read from L2
+        nchunk = (B + 63) // 64 * 2
+        cvimg = _scratch_bf16("head_cvimg", (nchunk, 8, 64, 8), dev)
+        ext().swizzle_cv(cv.contiguous(), cvimg)
         dw = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
         dbias = torch.empty(L, dtype=torch.float32, device=dev)
-        ext().head_bwd_dw(logits, cvt, lse, label, weight, acc, g, dw, dbias)
+        ext().head_bwd_dw(logits, cvimg, coef_lse, dw, dbias)
         # dcv (batch-major split-K): label chunk sized so the fp32 partial
         # slabs stay ~30 MB at any L
-        chunk = 512 if L <= 65536 else 4096
+        chunk = int(os.environ.get("C2V_HB_DCV_CHUNK", "0")) or (
+            512 if L <= 65536 else 4096)
         split = (L + chunk - 1) // chunk
         wt = _scratch_bf16("head_wt", (128, L), dev)
         ext().transpose_w(w, wt)
         partials = _scratch_f32("head_fused_dcv", (split, B, 128), dev)
-        ext().head_bwd_dcv(logits, wt, lse, label, weight, acc, g,
-                           partials, chunk)
+        ext().head_bwd_dcv(logits, wt, coef_lse, partials, chunk)
         dcv = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
         ext().slab_sum_bf16(partials, dcv)
         return None, dcv, dw, dbias, None, None
@@ -579,7 +586,8 @@ def fused_head_loss_supported(cv_bf16, w, training: bool) -> bool:
     B, EP = cv_bf16.shape
     L = w.shape[0]
     return (FUSED_HEAD and training and torch.is_grad_enabled()
-            and cv_bf16.is_cuda and EP == 128 and L % 8 == 0 and B % 8 == 0)
+            and cv_bf16.is_cuda and EP == 128 and L % 8 == 0 and B % 8 == 0
+            and L < (1 << 24))  # y carried as f32 in the G recompute
 
 
 def adam_step(

@@ -116,6 +116,38 @@ def main():
         from code2vec_amd.ops import ext
         results["wgrad"] = timeit(lambda: ext().wgrad(x, dz, partials))
 
+    if "head_bwd" in ops:
+        from code2vec_amd.ops import ext
+        cv, _ = Fn.AttentionPool.apply(ccv, a, starts, E)
+        cvb = cv.to(torch.bfloat16).contiguous()
+        bias = torch.zeros(L, device=dev)
+        logits = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+        gx = (L + 255) // 256
+        pm = torch.empty(gx, B, dtype=torch.float32, device=dev)
+        ps = torch.empty_like(pm)
+        ext().head_fwd(cvb, wout, bias, logits, pm, ps)
+        lse = torch.empty(B, dtype=torch.float32, device=dev)
+        acc = torch.zeros(2, dtype=torch.float32, device=dev)
+        ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc)
+        g1 = torch.ones(1, device=dev)
+        coef_lse = torch.zeros(B, 4, device=dev)
+        ext().head_bwd_prep(label, weight, acc, g1, lse, coef_lse)
+        nchunk = (B + 63) // 64 * 2
+        cvimg = torch.empty(nchunk, 8, 64, 8, dtype=torch.bfloat16, device=dev)
+        ext().swizzle_cv(cvb, cvimg)
+        dw = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
+        dbias = torch.empty(L, dtype=torch.float32, device=dev)
+        results["swizzle_cv"] = timeit(lambda: ext().swizzle_cv(cvb, cvimg))
+        results["head_bwd_dw"] = timeit(
+            lambda: ext().head_bwd_dw(logits, cvimg, coef_lse, dw, dbias))
+        chunk = 512 if L <= 65536 else 4096
+        split = (L + chunk - 1) // chunk
+        partials = torch.empty(split, B, 128, dtype=torch.float32, device=dev)
+        wt = torch.empty(128, L, dtype=torch.bfloat16, device=dev)
+        ext().transpose_w(wout, wt)
+        results["head_bwd_dcv"] = timeit(
+            lambda: ext().head_bwd_dcv(logits, wt, coef_lse, partials, chunk))
+
     for k, vv in results.items():
         print(f"{k:24s} {vv:10.1f} us")
 
