@@ -37,6 +37,7 @@ def main():
     dev = torch.device("cuda:0")
     B, C = 1024, 200
     T, P, L = 360632, 342846, 30000
+    L = int(os.environ.get("C2V_KBENCH_L", L))  # e.g. 261000 (java-large)
     dt = dp = E = 100
     TS, PS, EP = round_up(dt), round_up(dp), round_up(E)
     KP = 2 * TS + PS
@@ -115,6 +116,35 @@ def main():
         partials = torch.empty(256, KP, EP, dtype=torch.float32, device=dev)
         from code2vec_amd.ops import ext
         results["wgrad"] = timeit(lambda: ext().wgrad(x, dz, partials))
+
+    if "head_fwd" in ops:
+        from code2vec_amd.ops import ext
+        cv, _ = Fn.AttentionPool.apply(ccv, a, starts, E)
+        cvb = cv.to(torch.bfloat16).contiguous()
+        bias = torch.zeros(L, device=dev)
+        logits = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+        gx = (L + 255) // 256
+        pm = torch.empty(gx, B, dtype=torch.float32, device=dev)
+        ps = torch.empty_like(pm)
+        nil = torch.Tensor()
+        results["head_fwd(no stats)"] = timeit(
+            lambda: ext().head_fwd(cvb, wout, bias, logits, nil, nil),
+            iters=10)
+        results["head_fwd(stats)"] = timeit(
+            lambda: ext().head_fwd(cvb, wout, bias, logits, pm, ps),
+            iters=10)
+        lse = torch.empty(B, dtype=torch.float32, device=dev)
+        acc = torch.zeros(2, dtype=torch.float32, device=dev)
+        results["lsm_finalize"] = timeit(
+            lambda: ext().logsoftmax_nll_finalize(logits, pm, ps, label,
+                                                  weight, lse, acc),
+            iters=10)
+        results["lsm_full_fwd"] = timeit(
+            lambda: ext().logsoftmax_nll_fwd(logits, label, weight, lse,
+                                             acc), iters=10)
+        bias_h = bias.to(torch.bfloat16)
+        results["linear(hipblaslt)"] = timeit(
+            lambda: torch.nn.functional.linear(cvb, wout, bias_h), iters=10)
 
     if "head_bwd" in ops:
         from code2vec_amd.ops import ext
