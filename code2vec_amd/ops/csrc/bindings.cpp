@@ -78,6 +78,7 @@ void launch_head_fwd(const void*, const void*, const float*, void*, float*,
 void launch_lsm_finalize(const void*, const float*, const float*, const long*,
                          const float*, float*, float*, int, long, int,
                          hipStream_t);
+void launch_lsm_partial(const void*, float*, float*, int, long, hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
                       float, float, float, float, float, hipStream_t);
 void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
@@ -305,6 +306,19 @@ void logsoftmax_nll_bwd(torch::Tensor logits, torch::Tensor label,
                      cur_stream());
 }
 
+// (max, sumexp) partials over 16384-col chunks -> pm/ps[GX][B]
+void lsm_partial(torch::Tensor logits, torch::Tensor pm, torch::Tensor ps) {
+  CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
+  CHK_DT(pm, torch::kFloat32); CHK_CONTIG(pm); CHK_CONTIG(ps);
+  const int B = logits.size(0);
+  const long L = logits.size(1);
+  const long GX = (L + 16383) / 16384;
+  TORCH_CHECK(pm.numel() == GX * B && ps.numel() == GX * B,
+              "lsm_partial partials shape");
+  launch_lsm_partial(logits.data_ptr(), pm.data_ptr<float>(),
+                     ps.data_ptr<float>(), B, L, cur_stream());
+}
+
 void dgrad(torch::Tensor dz, torch::Tensor w2, torch::Tensor dx) {
   CHK_CUDA(dz); CHK_CONTIG(dz); CHK_DT(dz, torch::kBFloat16);
   CHK_CONTIG(w2); CHK_DT(w2, torch::kBFloat16);
@@ -363,8 +377,10 @@ void logsoftmax_nll_finalize(torch::Tensor logits, torch::Tensor pm,
   CHK_DT(lse, torch::kFloat32);
   const int B = logits.size(0);
   const long L = logits.size(1);
-  const int GX = (int)((L + 255) / 256);
-  TORCH_CHECK(pm.numel() == (long)B * GX && ps.numel() == (long)B * GX,
+  // GX derived from the partials shape: 256-col blocks from the head_fwd
+  // epilogue, 4096-col blocks from lsm_partial — finalize is generic
+  const int GX = (int)(pm.numel() / B);
+  TORCH_CHECK((long)GX * B == pm.numel() && ps.numel() == pm.numel(),
               "partials");
   const float* wp = weight.defined() && weight.numel() > 0
                         ? weight.data_ptr<float>()
@@ -561,6 +577,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_bf16", &colsum_bf16);
   m.def("head_fwd", &head_fwd);
   m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
+  m.def("lsm_partial", &lsm_partial);
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
   m.def("head_bwd_prep", &head_bwd_prep);

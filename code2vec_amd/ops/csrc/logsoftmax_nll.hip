@@ -68,6 +68,69 @@ __global__ __launch_bounds__(256) void lsm_nll_fwd_kernel(
   }
 }
 
+// 2-D-grid online-softmax partials: pm/ps[gx][b] = (max, sumexp) of
+// logits[b, gx*CHUNK : (gx+1)*CHUNK).  The one-block-per-row kernel above
+// walks L serially per block (B blocks only) — at L = 261k that is
+// latency-bound at ~3.6 TB/s; this produces the same (pm, ps) layout the
+// head-forward stats epilogue emits, merged by lsm_finalize_kernel
+// (head_fwd.hip), and runs at the streaming floor.
+#define LSMP_CHUNK 16384  // 32 KB/block: enough in-flight loads to hide latency
+__global__ __launch_bounds__(256) void lsm_partial_kernel(
+    const bf16* __restrict__ logits, float* __restrict__ pm,
+    float* __restrict__ ps, int B, long L) {
+  const int gxn = (int)((L + LSMP_CHUNK - 1) / LSMP_CHUNK);
+  const int b = blockIdx.x / gxn;
+  const int gx = blockIdx.x % gxn;
+  const bf16* row = logits + (long)b * L + (long)gx * LSMP_CHUNK;
+  const long n = min((long)LSMP_CHUNK, L - (long)gx * LSMP_CHUNK);
+  float m = -3.0e38f, s = 0.f;
+  const long n8 = n & ~7L;
+  for (long j = (long)threadIdx.x * 8; j < n8; j += 256 * 8) {
+    bf16 v[8];
+    *(uint4*)v = *(const uint4*)(row + j);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const float x = bf2f(v[k]);
+      if (x > m) {
+        s *= __expf(m - x);
+        m = x;
+      }
+      s += __expf(x - m);
+    }
+  }
+  for (long j = n8 + threadIdx.x; j < n; j += 256) {
+    const float x = bf2f(row[j]);
+    if (x > m) {
+      s *= __expf(m - x);
+      m = x;
+    }
+    s += __expf(x - m);
+  }
+  __shared__ float red_m[4], red_s[4];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float om = __shfl_xor(m, off);
+    const float os = __shfl_xor(s, off);
+    const float nm = fmaxf(m, om);
+    s = s * __expf(m - nm) + os * __expf(om - nm);
+    m = nm;
+  }
+  if (lane == 0) { red_m[wave] = m; red_s[wave] = s; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float M0 = red_m[0], S0 = red_s[0];
+    for (int w = 1; w < 4; ++w) {
+      const float nm = fmaxf(M0, red_m[w]);
+      S0 = S0 * __expf(M0 - nm) + red_s[w] * __expf(red_m[w] - nm);
+      M0 = nm;
+    }
+    pm[(long)gx * B + b] = M0;
+    ps[(long)gx * B + b] = S0;
+  }
+}
+
 // dlogits[b,j] = gscale * (w_y/wsum) * (exp(x - lse_b) - [j == y])
 __global__ __launch_bounds__(256) void lsm_nll_bwd_kernel(
     const bf16* __restrict__ logits, const long* __restrict__ label,
@@ -108,6 +171,13 @@ void launch_lsm_nll_fwd(const void* logits, const long* label,
                         long L, hipStream_t stream) {
   lsm_nll_fwd_kernel<<<B, 256, 0, stream>>>((const bf16*)logits, label,
                                             weight, lse, acc, B, L);
+}
+
+void launch_lsm_partial(const void* logits, float* pm, float* ps, int B,
+                        long L, hipStream_t stream) {
+  const int gxn = (int)((L + LSMP_CHUNK - 1) / LSMP_CHUNK);
+  lsm_partial_kernel<<<(long)B * gxn, 256, 0, stream>>>(
+      (const bf16*)logits, pm, ps, B, L);
 }
 
 void launch_lsm_nll_bwd(const void* logits, const long* label,

@@ -139,6 +139,14 @@ def main():
             lambda: ext().logsoftmax_nll_finalize(logits, pm, ps, label,
                                                   weight, lse, acc),
             iters=10)
+        gx2 = (L + 16383) // 16384
+        pm2 = torch.empty(gx2, B, dtype=torch.float32, device=dev)
+        ps2 = torch.empty_like(pm2)
+        results["lsm_partial+final"] = timeit(
+            lambda: (ext().lsm_partial(logits, pm2, ps2),
+                     ext().logsoftmax_nll_finalize(logits, pm2, ps2, label,
+                                                   weight, lse, acc)),
+            iters=10)
         results["lsm_full_fwd"] = timeit(
             lambda: ext().logsoftmax_nll_fwd(logits, label, weight, lse,
                                              acc), iters=10)
