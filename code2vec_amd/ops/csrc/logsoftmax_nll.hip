@@ -83,20 +83,32 @@ __global__ __launch_bounds__(256) void lsm_partial_kernel(
   const int gx = blockIdx.x % gxn;
   const bf16* row = logits + (long)b * L + (long)gx * LSMP_CHUNK;
   const long n = min((long)LSMP_CHUNK, L - (long)gx * LSMP_CHUNK);
-  float m = -3.0e38f, s = 0.f;
+  // 8 independent (max, sum) accumulators, one per vector slot: the
+  // single-accumulator form chains every element through a dependent
+  // exp+compare (measured 4.9 TB/s); independent slots break the chain
   const long n8 = n & ~7L;
+  float m8[8], s8[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) { m8[k] = -3.0e38f; s8[k] = 0.f; }
   for (long j = (long)threadIdx.x * 8; j < n8; j += 256 * 8) {
     bf16 v[8];
     *(uint4*)v = *(const uint4*)(row + j);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
       const float x = bf2f(v[k]);
-      if (x > m) {
-        s *= __expf(m - x);
-        m = x;
+      if (x > m8[k]) {
+        s8[k] *= __expf(m8[k] - x);
+        m8[k] = x;
       }
-      s += __expf(x - m);
+      s8[k] += __expf(x - m8[k]);
     }
+  }
+  float m = m8[0], s = s8[0];
+#pragma unroll
+  for (int k = 1; k < 8; ++k) {
+    const float nm = fmaxf(m, m8[k]);
+    s = s * __expf(m - nm) + s8[k] * __expf(m8[k] - nm);
+    m = nm;
   }
   for (long j = n8 + threadIdx.x; j < n; j += 256) {
     const float x = bf2f(row[j]);

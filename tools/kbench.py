@@ -38,6 +38,7 @@ def main():
     B, C = 1024, 200
     T, P, L = 360632, 342846, 30000
     L = int(os.environ.get("C2V_KBENCH_L", L))  # e.g. 261000 (java-large)
+    T = int(os.environ.get("C2V_KBENCH_T", T))
     dt = dp = E = 100
     TS, PS, EP = round_up(dt), round_up(dp), round_up(E)
     KP = 2 * TS + PS
@@ -99,6 +100,15 @@ def main():
         logits = (cv.to(torch.bfloat16) @ wout.t()).contiguous()
         results["lsm_fwd"] = timeit(
             lambda: Fn.FusedLogSoftmaxNLL.apply(logits, label, weight))
+
+    if "membw" in ops:
+        # same-box streaming ceiling: D2D copy (read+write) of 2 GB
+        src = torch.empty(1 << 28, dtype=torch.float32, device=dev)
+        dst = torch.empty_like(src)
+        src.fill_(1.0)
+        t = timeit(lambda: dst.copy_(src), iters=10)
+        results["d2d_copy(2GB r+w)"] = t
+        print(f"# copy bandwidth: {2 * src.numel() * 4 / (t * 1e-6) / 1e12:.2f} TB/s")
 
     if "adam" in ops:
         p1 = term.clone().view(-1)
