@@ -96,7 +96,11 @@ def main() -> None:
         model = Code2VecTorch(opt, init_logical_params(opt, g))
         optim = torch.optim.Adam(model.parameters(), lr=0.01)
 
-    ddp = BucketedAllReduce(list(model.parameters()), world)
+    owned = (
+        [model.terminal_embedding, model.path_embedding] if on_gpu else None
+    )
+    ddp = BucketedAllReduce(list(model.parameters()), world,
+                            owned_params=owned)
     ddp.broadcast_parameters()
 
     B, C = cfg["batch"], cfg["contexts"]
@@ -122,8 +126,11 @@ def main() -> None:
         outputs, _, _ = model(s, p, e, y)
         loss = model.loss(outputs, y, class_weight)
         loss.backward()
-        ddp.finish()
-        optim.step()
+        if on_gpu:
+            ddp.finish_and_step(optim)
+        else:
+            ddp.finish()
+            optim.step()
 
     for i in range(args.warmup):
         step(i)

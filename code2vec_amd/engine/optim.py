@@ -46,10 +46,14 @@ class FusedAdam:
             self.state[p] = st
 
     @torch.no_grad()
-    def step(self) -> None:
+    def step(self, exclude_ids=None) -> None:
+        """One Adam step over params with grads.  ``exclude_ids`` (a set of
+        id(param)) skips params whose update will arrive via
+        ``step_rows`` in the same logical step (the DDP-overlapped table
+        path) — the step counter is bumped here, once."""
         self.step_count += 1
         for p in self.params:
-            if p.grad is None:
+            if p.grad is None or (exclude_ids and id(p) in exclude_ids):
                 continue
             st = self.state[p]
             Fn.adam_step(
@@ -57,6 +61,23 @@ class FusedAdam:
                 self.step_count, self.lr, self.beta1, self.beta2,
                 self.eps, self.weight_decay,
             )
+
+    @torch.no_grad()
+    def step_rows(self, p, grad2d, row_lo: int, row_hi: int) -> None:
+        """Adam on rows [row_lo, row_hi) of a 2-D param, using a caller-
+        owned grad tensor (p.grad may be None).  Uses the step count of
+        the enclosing ``step()`` call — elementwise Adam makes the row
+        partition exact."""
+        W = p.shape[1]
+        lo, hi = row_lo * W, row_hi * W
+        st = self.state[p]
+        Fn.adam_step(
+            p.data.view(-1)[lo:hi], grad2d.view(-1)[lo:hi],
+            None if st["master"] is None else st["master"][lo:hi],
+            st["m"][lo:hi], st["v"][lo:hi],
+            self.step_count, self.lr, self.beta1, self.beta2,
+            self.eps, self.weight_decay,
+        )
 
     def zero_grad(self, set_to_none: bool = False) -> None:
         # NOTE: when grads are BucketedAllReduce views, use ddp.zero_grad()

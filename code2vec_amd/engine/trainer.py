@@ -104,10 +104,17 @@ class Trainer:
                 weight_decay=config.weight_decay,
             )
 
+        # the embedding tables take the owner-buffer chunked-overlap path
+        # (their grads are produced last in backward; see parallel/ddp.py)
+        owned = (
+            [model.terminal_embedding, model.path_embedding]
+            if isinstance(model, Code2VecHIP) else None
+        )
         self.ddp = BucketedAllReduce(
-            list(model.parameters()), ctx.world_size
+            list(model.parameters()), ctx.world_size, owned_params=owned
         )
         self.ddp.broadcast_parameters()
+        self._overlapped_step = isinstance(self.optimizer, FusedAdam)
 
         self.roctx = _Roctx(self.device.type == "cuda")
         # epoch N+1's host-side rebuild runs here while N trains (the
@@ -277,10 +284,15 @@ class Trainer:
             self.roctx.pop()
             self.roctx.push("backward")
             loss.backward()
-            self.ddp.finish()
             self.roctx.pop()
             self.roctx.push("optimizer")
-            self.optimizer.step()
+            if self._overlapped_step:
+                # comm-overlapped: non-table Adam + chunkwise table Adam
+                # pipelined against the in-flight table all-reduces
+                self.ddp.finish_and_step(self.optimizer)
+            else:
+                self.ddp.finish()
+                self.optimizer.step()
             self.roctx.pop()
             self.roctx.pop()
 

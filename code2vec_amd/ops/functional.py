@@ -45,11 +45,16 @@ _DGRAD2 = os.environ.get("C2V_DGRAD2", "1") == "1"
 FUSED_HEAD = os.environ.get("C2V_FUSED_HEAD", "1") == "1"
 _NONE_T = torch.Tensor()  # "not provided" sentinel for optional kernel args
 
-# Optional early-gradient callbacks keyed by param.data_ptr(): the embedding
-# backward invokes them the moment a table's grad tensor is complete.
-# parallel/ddp.py no longer registers any (see the note in _on_grad_ready);
-# the hook point remains for schemes that own their buffers.
+# Owner-buffer gradient flow for the embedding tables (parallel/ddp.py):
+# when a param's data_ptr is in OWNED_GRAD_KEYS, the embedding backward
+# hands its finished grad tensor to EARLY_GRAD_CALLBACKS[key] the moment
+# its cast_clear completes and returns None to autograd.  Autograd then
+# never adopts (and possibly CLONES) the tensor — the round-1 hazard that
+# made in-place early all-reduce unsafe — so the DDP layer can launch
+# chunked async all-reduces on a buffer it owns while the REST of backward
+# (the other table's sort/scatter chain) still runs.
 EARLY_GRAD_CALLBACKS = {}
+OWNED_GRAD_KEYS = set()
 
 from . import ext, round_up
 
@@ -120,23 +125,30 @@ def _scatter_embedding_grads(starts, paths, ends, gout, term_shape,
     flags_p = _scratch_flags("path", path_shape[0], dev)
     dterm = torch.empty(term_shape, dtype=torch.bfloat16, device=dev)
     dpath = torch.empty(path_shape, dtype=torch.bfloat16, device=dev)
+    # term's FULL chain (sort+scatter+cast) runs before path's STARTS, so
+    # the early callback can put term's all-reduce on the wire while
+    # path's ~100 us of scatter kernels still execute (DP overlap)
     idx_se = torch.cat([starts.view(-1), ends.view(-1)])
     sorted_se, perm_se, counts_se = _group_by_index(idx_se, term_shape[0])
     ext().embed_scatter_sorted(sorted_se, perm_se, gout, dterm32, dterm,
                                flags_t, M, KP, 0, TS + PS, _SCATTER_R)
-    sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
-    ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
-                               flags_p, M, KP, TS, TS, _SCATTER_R)
     ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
     if term_key is not None:
         cb = EARLY_GRAD_CALLBACKS.get(term_key)
         if cb is not None:
             cb(dterm)
+    sorted_p, perm_p, counts_p = _group_by_index(paths.view(-1), path_shape[0])
+    ext().embed_scatter_sorted(sorted_p, perm_p, gout, dpath32, dpath,
+                               flags_p, M, KP, TS, TS, _SCATTER_R)
     ext().cast_clear_rows(dpath32, counts_p, flags_p, dpath)
     if path_key is not None:
         cb = EARLY_GRAD_CALLBACKS.get(path_key)
         if cb is not None:
             cb(dpath)
+    if term_key in OWNED_GRAD_KEYS:
+        dterm = None  # the callback owner keeps the buffer; autograd must
+    if path_key in OWNED_GRAD_KEYS:  # not adopt (or clone) it
+        dpath = None
     return dterm, dpath
 
 

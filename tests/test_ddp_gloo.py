@@ -252,3 +252,87 @@ def test_early_grad_callback_world2(tmp_path):
         pr.join(timeout=240)
         assert pr.exitcode == 0
     assert open(result_file).read() == "ok"
+
+
+def _worker_owned(rank, world, port, result_file):
+    """Owner-buffer chunked all-reduce + pipelined row-chunk step under
+    gloo: must equal the dense all-reduce + full step on the same grads."""
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    from code2vec_amd.ops import functional as Fn
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    class MockSGD:
+        """Implements the FusedAdam step/step_rows contract on CPU."""
+
+        def __init__(self, params, lr=0.1):
+            self.params = list(params)
+            self.lr = lr
+
+        def step(self, exclude_ids=None):
+            for p in self.params:
+                if p.grad is None or (exclude_ids and id(p) in exclude_ids):
+                    continue
+                p.data -= self.lr * p.grad
+
+        def step_rows(self, p, grad2d, lo, hi):
+            p.data[lo:hi] -= self.lr * grad2d[lo:hi].float()
+
+    torch.manual_seed(7)
+    N, W = 37, 8  # odd row count: uneven chunk tail
+    init = torch.randn(N, W)
+    p_owned = torch.nn.Parameter(init.clone())
+    p_plain = torch.nn.Parameter(init.clone())
+    grads = [torch.randn(N, W) * (rank + 1 + s) for s in range(2)]
+
+    ddp = BucketedAllReduce([p_owned], world, owned_params=[p_owned],
+                            owned_chunks=4)
+    optim = MockSGD([p_owned])
+    key = p_owned.data_ptr()
+    assert key in Fn.OWNED_GRAD_KEYS
+    for g in grads:
+        ddp.zero_grad()
+        Fn.EARLY_GRAD_CALLBACKS[key](g.clone())  # as the backward would
+        ddp.finish_and_step(optim)
+
+    # dense reference on the plain param
+    for g in grads:
+        gd = g.clone()
+        dist.all_reduce(gd)
+        p_plain.data -= 0.1 * (gd / world)
+
+    assert torch.allclose(p_owned.detach(), p_plain.detach(), atol=1e-6)
+
+    # the non-pipelined finish() publishes the reduced grad as p.grad
+    ddp.zero_grad()
+    Fn.EARLY_GRAD_CALLBACKS[key](grads[0].clone())
+    ddp.finish()
+    gd = grads[0].clone()
+    dist.all_reduce(gd)
+    assert torch.allclose(p_owned.grad, gd / world, atol=1e-6)
+    ddp.close()
+    assert key not in Fn.OWNED_GRAD_KEYS
+
+    if rank == 0:
+        with open(result_file, "w") as f:
+            f.write("ok")
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_owned_chunked_allreduce_world2(tmp_path):
+    ctx = mp.get_context("spawn")
+    result_file = str(tmp_path / "owned.txt")
+    procs = [ctx.Process(target=_worker_owned,
+                         args=(r, 2, 29551, result_file)) for r in range(2)]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    assert open(result_file).read() == "ok"

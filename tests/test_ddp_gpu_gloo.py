@@ -16,7 +16,7 @@ pytestmark = pytest.mark.gpu
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _worker(rank, world, port, result_file):
+def _worker(rank, world, port, result_file, mode="plain"):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -40,8 +40,11 @@ def _worker(rank, world, port, result_file):
                  batch_size=16, device=dev)
     g = torch.Generator().manual_seed(5)
     model = Code2VecHIP(opt, init_logical_params(opt, g), device=dev).train()
+    owned = ([model.terminal_embedding, model.path_embedding]
+             if mode == "owned" else None)
     ddp = BucketedAllReduce(list(model.parameters()), world,
-                            direct_threshold=64 * 1024)
+                            direct_threshold=64 * 1024, owned_params=owned,
+                            owned_chunks=3)
     ddp.broadcast_parameters()
     optim = FusedAdam(model.parameters(), lr=0.01)
     w = torch.ones(opt.label_count, device=dev)
@@ -57,8 +60,12 @@ def _worker(rank, world, port, result_file):
         out, _, _ = model(s, p, e, y)
         loss = model.loss(out, y, w)
         loss.backward()
-        ddp.finish()  # raises loudly if grad adoption failed
-        optim.step()
+        if mode == "owned":
+            # chunked owner-buffer all-reduce + pipelined per-chunk Adam
+            ddp.finish_and_step(optim)
+        else:
+            ddp.finish()  # raises loudly if grad adoption failed
+            optim.step()
     torch.cuda.synchronize()
 
     # replicas must stay bit-identical after synchronized updates
@@ -75,13 +82,24 @@ def _worker(rank, world, port, result_file):
 
 @pytest.mark.timeout(600)
 def test_hip_dp_world2_one_gpu(tmp_path):
-    ctx = mp.get_context("spawn")
-    result_file = str(tmp_path / "dp.txt")
-    procs = [ctx.Process(target=_worker, args=(r, 2, 29541, result_file))
-             for r in range(2)]
-    for pr in procs:
-        pr.start()
-    for pr in procs:
-        pr.join(timeout=500)
-        assert pr.exitcode == 0
-    assert open(result_file).read().startswith("ok")
+    """Both DP paths on the HIP model under gloo world=2; their results
+    must agree with each other (chunked owner-buffer all-reduce +
+    pipelined row-chunk Adam == plain reduce + full step)."""
+    digests = {}
+    for i, mode in enumerate(("plain", "owned")):
+        ctx = mp.get_context("spawn")
+        result_file = str(tmp_path / f"dp_{mode}.txt")
+        procs = [
+            ctx.Process(target=_worker,
+                        args=(r, 2, 29541 + i, result_file, mode))
+            for r in range(2)
+        ]
+        for pr in procs:
+            pr.start()
+        for pr in procs:
+            pr.join(timeout=500)
+            assert pr.exitcode == 0
+        txt = open(result_file).read()
+        assert txt.startswith("ok")
+        digests[mode] = txt.split()[1]
+    assert digests["plain"] == digests["owned"], digests

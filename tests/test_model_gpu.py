@@ -321,3 +321,50 @@ def test_torch_backend_on_gpu(dev):
     loss = model.loss(out, y, torch.ones(opt.label_count, device=dev))
     loss.backward()
     assert torch.isfinite(model.terminal_embedding.grad).all()
+
+
+def test_finish_and_step_matches_plain_step(dev):
+    """World=1: the overlapped finish_and_step (owner-buffer tables +
+    row-chunk Adam) must update parameters exactly like finish()+step()."""
+    import numpy as np
+
+    from code2vec_amd.data.synthetic import synthetic_batch
+    from code2vec_amd.engine.optim import FusedAdam
+    from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+    from code2vec_amd.utils.options import Option
+
+    opt = make_option(dropout_prob=0.0)
+    g = torch.Generator().manual_seed(21)
+    logical = init_logical_params(opt, g)
+    rng = np.random.default_rng(55)
+    s, p, e, y = synthetic_batch(rng, 16, opt.max_path_length,
+                                 opt.terminal_count, opt.path_count,
+                                 opt.label_count)
+    s = torch.from_numpy(s).to(dev); p = torch.from_numpy(p).to(dev)
+    e = torch.from_numpy(e).to(dev); y = torch.from_numpy(y).to(dev)
+    w = torch.ones(opt.label_count, device=dev)
+
+    finals = []
+    for mode in ("plain", "owned"):
+        m = Code2VecHIP(opt, logical, device=dev).train()
+        owned = ([m.terminal_embedding, m.path_embedding]
+                 if mode == "owned" else None)
+        ddp = BucketedAllReduce(list(m.parameters()), 1, owned_params=owned,
+                                owned_chunks=3)
+        optim = FusedAdam(m.parameters(), lr=0.01)
+        for _ in range(3):
+            ddp.zero_grad()
+            out, _, _ = m(s, p, e, y)
+            loss = m.loss(out, y, w)
+            loss.backward()
+            if mode == "owned":
+                ddp.finish_and_step(optim)
+            else:
+                ddp.finish()
+                optim.step()
+        finals.append({n: q.detach().float().cpu()
+                       for n, q in m.named_parameters()})
+        ddp.close()
+    for name in finals[0]:
+        assert torch.equal(finals[0][name], finals[1][name]), name
