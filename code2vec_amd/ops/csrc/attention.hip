@@ -255,23 +255,32 @@ __global__ __launch_bounds__(256) void attention_bwd_kernel(
       }
     }
   }
-  // combine da partials: each group's lane holds 8 cols (+128*i); groups
-  // 0..15 across 4 waves all accumulated different context spans -> LDS sum
+  // combine da partials DETERMINISTICALLY: the 4 lane-groups of each
+  // wave that share lane16 combine via a fixed shuffle tree, each wave
+  // writes its own partials row, and the final 4-way sum runs in fixed
+  // wave order.  (The previous LDS atomicAdd combine was ordering-
+  // nondeterministic at the fp32 ulp — identical runs diverged after a
+  // couple of bf16-rounded optimizer steps.)
   __syncthreads();  // ds/tile reads done; reuse partials region
-  for (int e = threadIdx.x; e < EP; e += blockDim.x) partials[e] = 0.f;
-  __syncthreads();
+  for (int e = threadIdx.x; e < 4 * EP; e += blockDim.x)
+    if ((e % EP) >= ts) partials[e] = 0.f;  // tail cols no lane writes
 #pragma unroll
   for (int i = 0; i < 3; ++i) {
     const int e_base = lane16 * 8 + 128 * i;
-    if (e_base < ts) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        atomicAdd(&partials[e_base + j], da_acc[i][j]);
+    for (int j = 0; j < 8; ++j) {
+      float v = da_acc[i][j];
+      v += __shfl_xor(v, 16);
+      v += __shfl_xor(v, 32);
+      if (lane < 16 && e_base < ts) partials[wave * EP + e_base + j] = v;
     }
   }
   __syncthreads();
-  for (int e = threadIdx.x; e < EP; e += blockDim.x)
-    da[(long)b * EP + e] = e < E ? partials[e] : 0.f;
+  for (int e = threadIdx.x; e < EP; e += blockDim.x) {
+    const float s4v = partials[e] + partials[EP + e] + partials[2 * EP + e] +
+                      partials[3 * EP + e];
+    da[(long)b * EP + e] = e < E ? s4v : 0.f;
+  }
   __syncthreads();  // smem reused by the next method
   }
 }

@@ -282,6 +282,7 @@ void logsoftmax_nll_fwd(torch::Tensor logits, torch::Tensor label,
   CHK_DT(label, torch::kInt64); CHK_DT(lse, torch::kFloat32);
   const int B = logits.size(0);
   const long L = logits.size(1);
+  TORCH_CHECK(acc.numel() == 2L * B, "acc must be [B, 2] partials");
   const float* wp = weight.defined() && weight.numel() > 0
                         ? weight.data_ptr<float>()
                         : nullptr;
@@ -382,6 +383,8 @@ void logsoftmax_nll_finalize(torch::Tensor logits, torch::Tensor pm,
   const int GX = (int)(pm.numel() / B);
   TORCH_CHECK((long)GX * B == pm.numel() && ps.numel() == pm.numel(),
               "partials");
+  TORCH_CHECK(acc.numel() == 2L * ((B + 15) / 16),
+              "acc must be [(B+15)/16, 2] partials");
   const float* wp = weight.defined() && weight.numel() > 0
                         ? weight.data_ptr<float>()
                         : nullptr;

@@ -349,34 +349,29 @@ __global__ __launch_bounds__(256) void combiner_bwd_kernel(
   }
   // block-combine the per-lane dgamma/dbeta partials in LDS, then one
   // partials-row store per block (summed host-side; avoids the
-  // all-waves-hit-128-addresses atomic serialization).
-  __shared__ float red_g[256 * 2 * NPAIR];
-  float* red_b = red_g + 128 * NPAIR;
+  // all-waves-hit-128-addresses atomic serialization).  Each column is
+  // owned by exactly ONE lane within a wave, so every wave writes its own
+  // row and the final 4-way sum runs in fixed order — the earlier LDS
+  // atomicAdd combine was ordering-nondeterministic at the fp32 ulp and
+  // made identical training runs diverge (see PERF.md determinism note).
+  __shared__ float red_g[4][128 * NPAIR];
+  __shared__ float red_b[4][128 * NPAIR];
 #pragma unroll
   for (int i = 0; i < NPAIR; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int c = cols[i] + j;
-      if (wave == 0 && c < 128 * NPAIR) {
-        red_g[c] = 0.f;
-        red_b[c] = 0.f;
-      }
-    }
-  __syncthreads();
-#pragma unroll
-  for (int i = 0; i < NPAIR; ++i)
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int c = cols[i] + j;
-      if (c < EP) {
-        atomicAdd(&red_g[c], acc_dg[i][j]);  // LDS atomics: 4 waves only
-        atomicAdd(&red_b[c], acc_db[i][j]);
+      if (c < 128 * NPAIR) {
+        red_g[wave][c] = (c < EP) ? acc_dg[i][j] : 0.f;
+        red_b[wave][c] = (c < EP) ? acc_db[i][j] : 0.f;
       }
     }
   __syncthreads();
   for (int c = threadIdx.x; c < EP; c += blockDim.x) {
-    dgamma_part[(long)blockIdx.x * EP + c] = red_g[c];
-    dbeta_part[(long)blockIdx.x * EP + c] = red_b[c];
+    dgamma_part[(long)blockIdx.x * EP + c] =
+        red_g[0][c] + red_g[1][c] + red_g[2][c] + red_g[3][c];
+    dbeta_part[(long)blockIdx.x * EP + c] =
+        red_b[0][c] + red_b[1][c] + red_b[2][c] + red_b[3][c];
   }
 }
 

@@ -291,11 +291,15 @@ __global__ __launch_bounds__(512) void head_fwd_kernel(
 // pathology #1).  Partials are laid out [lab_block][B] (coalesced
 // producer writes); they were just written, so the strided reads here
 // hit L2.
+// acc_part[block][2] per-block partials (summed by slab_sum_f32 in fixed
+// order): an atomic-add accumulation is ordering-nondeterministic at the
+// fp32 ulp level, which made otherwise-identical training runs diverge
+// after 2 steps of bf16 rounding.
 __global__ __launch_bounds__(1024) void lsm_finalize_kernel(
     const bf16* __restrict__ logits, const float* __restrict__ pm,
     const float* __restrict__ ps, const long* __restrict__ label,
     const float* __restrict__ weight, float* __restrict__ lse,
-    float* __restrict__ acc, int B, long L, int GXL) {
+    float* __restrict__ acc_part, int B, long L, int GXL) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int b = blockIdx.x * 16 + wave;
@@ -335,8 +339,8 @@ __global__ __launch_bounds__(1024) void lsm_finalize_kernel(
       a0 += red[w][0];
       a1 += red[w][1];
     }
-    atomic_add_f32(&acc[0], a0);
-    atomic_add_f32(&acc[1], a1);
+    acc_part[(long)blockIdx.x * 2] = a0;
+    acc_part[(long)blockIdx.x * 2 + 1] = a1;
   }
 }
 

@@ -8,10 +8,12 @@
 #include "common.h"
 
 // one block per row; online (max, sum) over the row's L columns
+// acc_part[b][2] per-row partials (summed deterministically on the host
+// side via slab_sum_f32 — see the ordering note in head_fwd.hip)
 __global__ __launch_bounds__(256) void lsm_nll_fwd_kernel(
     const bf16* __restrict__ logits, const long* __restrict__ label,
     const float* __restrict__ weight, float* __restrict__ lse,
-    float* __restrict__ acc, int B, long L) {
+    float* __restrict__ acc_part, int B, long L) {
   const int b = blockIdx.x;
   const bf16* row = logits + (long)b * L;
   float m = -3.0e38f, s = 0.f;
@@ -63,8 +65,8 @@ __global__ __launch_bounds__(256) void lsm_nll_fwd_kernel(
     lse[b] = l;
     const long y = label[b];
     const float wy = weight ? weight[y] : 1.0f;
-    atomic_add_f32(&acc[0], wy * (l - bf2f(row[y])));
-    atomic_add_f32(&acc[1], wy);
+    acc_part[(long)b * 2] = wy * (l - bf2f(row[y]));
+    acc_part[(long)b * 2 + 1] = wy;
   }
 }
 

@@ -466,20 +466,38 @@ class OutputHead(torch.autograd.Function):
         return dcv, dw, dbias
 
 
-def _lsm_stats(logits, label, weight, lse, acc):
-    """Fill (lse, acc) from a full pass over logits.  Large L takes the
-    2-D-grid partials kernel + finalize (the one-block-per-row walk is
-    latency-bound at ~3.6 TB/s at L = 261k: 150 -> ~75 us)."""
+def _acc_reduce(acc_part):
+    """Deterministic (fixed-order) reduction of per-block loss partials
+    [S, 2] -> acc[2] = (sum w_y*nll, sum w_y).  The previous atomic-add
+    accumulation was ordering-nondeterministic at the fp32 ulp level and
+    made identical runs diverge after two steps of bf16 rounding."""
+    acc = torch.empty(2, dtype=torch.float32, device=acc_part.device)
+    ext().slab_sum_f32(acc_part, acc)
+    return acc
+
+
+def _finalize_stats(logits, pm, ps, label, weight, lse):
+    B = logits.shape[0]
+    acc_p = torch.empty((B + 15) // 16, 2, dtype=torch.float32,
+                        device=logits.device)
+    ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc_p)
+    return _acc_reduce(acc_p)
+
+
+def _lsm_stats(logits, label, weight, lse):
+    """Fill lse and return acc from a full pass over logits.  Large L
+    takes the 2-D-grid partials kernel + finalize (the one-block-per-row
+    walk is latency-bound at ~3.6 TB/s at L = 261k: 150 -> ~75 us)."""
     B, L = logits.shape
     if L >= 32768:
         gx = (L + 16383) // 16384
         pm = torch.empty(gx, B, dtype=torch.float32, device=logits.device)
         ps = torch.empty_like(pm)
         ext().lsm_partial(logits, pm, ps)
-        ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse,
-                                      acc)
-    else:
-        ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc)
+        return _finalize_stats(logits, pm, ps, label, weight, lse)
+    acc_p = torch.empty(B, 2, dtype=torch.float32, device=logits.device)
+    ext().logsoftmax_nll_fwd(logits, label, weight, lse, acc_p)
+    return _acc_reduce(acc_p)
 
 
 class FusedLogSoftmaxNLL(torch.autograd.Function):
@@ -494,7 +512,6 @@ class FusedLogSoftmaxNLL(torch.autograd.Function):
         B, L = logits.shape
         lse = torch.empty(B, dtype=torch.float32, device=logits.device)
         # acc[0] = sum(w_y * nll), acc[1] = sum(w_y)
-        acc = torch.zeros(2, dtype=torch.float32, device=logits.device)
         partials = getattr(logits, "_c2v_lsm_partials", None)
         if partials is not None:
             # the head-forward epilogue already reduced per-row (max, sum)
@@ -502,10 +519,9 @@ class FusedLogSoftmaxNLL(torch.autograd.Function):
             # the whole [B, L] matrix
             pm, ps = partials
             del logits._c2v_lsm_partials
-            ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight,
-                                          lse, acc)
+            acc = _finalize_stats(logits, pm, ps, label, weight, lse)
         else:
-            _lsm_stats(logits, label, weight, lse, acc)
+            acc = _lsm_stats(logits, label, weight, lse)
         loss = acc[0] / acc[1]
         ctx.save_for_backward(logits, label, weight, lse, acc)
         return loss
@@ -563,15 +579,13 @@ class FusedHeadLoss(torch.autograd.Function):
     def forward(ctx, logits, cv_bf16, w, bias, label, weight):
         B, L = logits.shape
         lse = torch.empty(B, dtype=torch.float32, device=logits.device)
-        acc = torch.zeros(2, dtype=torch.float32, device=logits.device)
         partials = getattr(logits, "_c2v_lsm_partials", None)
         if partials is not None:
             pm, ps = partials
             del logits._c2v_lsm_partials
-            ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight,
-                                          lse, acc)
+            acc = _finalize_stats(logits, pm, ps, label, weight, lse)
         else:
-            _lsm_stats(logits, label, weight, lse, acc)
+            acc = _lsm_stats(logits, label, weight, lse)
         loss = acc[0] / acc[1]
         ctx.save_for_backward(logits, cv_bf16, w, label, weight, lse, acc)
         return loss

@@ -368,3 +368,45 @@ def test_finish_and_step_matches_plain_step(dev):
         ddp.close()
     for name in finals[0]:
         assert torch.equal(finals[0][name], finals[1][name]), name
+
+
+def test_training_bitwise_deterministic(dev):
+    """Two identical 3-step training runs produce bitwise-identical
+    parameters: every reduction in the framework is fixed-order (no
+    fp32 atomic accumulation ordering anywhere in the training path)."""
+    import numpy as np
+
+    from code2vec_amd.data.synthetic import synthetic_batch
+    from code2vec_amd.engine.optim import FusedAdam
+    from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+    from code2vec_amd.ops.functional import reseed_dropout_rng
+
+    opt = make_option(dropout_prob=0.25)  # dropout: counter RNG, reseeded
+    g = torch.Generator().manual_seed(31)
+    logical = init_logical_params(opt, g)
+    rng = np.random.default_rng(66)
+    s, p, e, y = synthetic_batch(rng, 16, opt.max_path_length,
+                                 opt.terminal_count, opt.path_count,
+                                 opt.label_count)
+    s = torch.from_numpy(s).to(dev); p = torch.from_numpy(p).to(dev)
+    e = torch.from_numpy(e).to(dev); y = torch.from_numpy(y).to(dev)
+    w = torch.ones(opt.label_count, device=dev)
+
+    def run():
+        reseed_dropout_rng(77)
+        m = Code2VecHIP(opt, logical, device=dev).train()
+        ddp = BucketedAllReduce(list(m.parameters()), 1)
+        optim = FusedAdam(m.parameters(), lr=0.01)
+        for _ in range(3):
+            ddp.zero_grad()
+            out, _, _ = m(s, p, e, y)
+            m.loss(out, y, w).backward()
+            ddp.finish()
+            optim.step()
+        return {n: q.detach().float().cpu().clone()
+                for n, q in m.named_parameters()}
+
+    a, b = run(), run()
+    for name in a:
+        assert torch.equal(a[name], b[name]), name

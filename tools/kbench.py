@@ -144,10 +144,10 @@ def main():
             lambda: ext().head_fwd(cvb, wout, bias, logits, pm, ps),
             iters=10)
         lse = torch.empty(B, dtype=torch.float32, device=dev)
-        acc = torch.zeros(2, dtype=torch.float32, device=dev)
+        acc_f = torch.empty((B + 15) // 16, 2, dtype=torch.float32, device=dev)
         results["lsm_finalize"] = timeit(
             lambda: ext().logsoftmax_nll_finalize(logits, pm, ps, label,
-                                                  weight, lse, acc),
+                                                  weight, lse, acc_f),
             iters=10)
         gx2 = (L + 16383) // 16384
         pm2 = torch.empty(gx2, B, dtype=torch.float32, device=dev)
@@ -155,11 +155,12 @@ def main():
         results["lsm_partial+final"] = timeit(
             lambda: (ext().lsm_partial(logits, pm2, ps2),
                      ext().logsoftmax_nll_finalize(logits, pm2, ps2, label,
-                                                   weight, lse, acc)),
+                                                   weight, lse, acc_f)),
             iters=10)
+        acc_b = torch.empty(B, 2, dtype=torch.float32, device=dev)
         results["lsm_full_fwd"] = timeit(
             lambda: ext().logsoftmax_nll_fwd(logits, label, weight, lse,
-                                             acc), iters=10)
+                                             acc_b), iters=10)
         bias_h = bias.to(torch.bfloat16)
         results["linear(hipblaslt)"] = timeit(
             lambda: torch.nn.functional.linear(cvb, wout, bias_h), iters=10)
@@ -175,8 +176,10 @@ def main():
         ps = torch.empty_like(pm)
         ext().head_fwd(cvb, wout, bias, logits, pm, ps)
         lse = torch.empty(B, dtype=torch.float32, device=dev)
-        acc = torch.zeros(2, dtype=torch.float32, device=dev)
-        ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc)
+        acc_f = torch.empty((B + 15) // 16, 2, dtype=torch.float32, device=dev)
+        ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc_f)
+        acc = torch.empty(2, dtype=torch.float32, device=dev)
+        ext().slab_sum_f32(acc_f, acc)
         g1 = torch.ones(1, device=dev)
         coef_lse = torch.zeros(B, 4, device=dev)
         ext().head_bwd_prep(label, weight, acc, g1, lse, coef_lse)
