@@ -44,8 +44,15 @@ from code2vec_amd.parallel.dist import init_distributed
 from code2vec_amd.utils.options import Option
 
 CONFIGS = {
-    # BASELINE.json config 2/3: top11-shaped (top11_dataset/params.txt:7-9)
-    "top11": dict(terminal_count=360632, path_count=342846, label_count=30000,
+    # BASELINE.json config 2/3: top11-shaped (top11_dataset/params.txt:7-9).
+    # label_count DERIVED from real reference data: Heaps' law
+    # V(n)=1.01*n^0.840 fit to the prefix-unique normalized-name curve of
+    # the reference's dataset/methods.txt (9,916 real methods, 2,280
+    # unique) gives V(605,945)=72,415; the synthetic generator's
+    # calibrated name process reproduces it (72,539 simulated) —
+    # tools/derive_label_vocab.py.  72,416 = fit rounded to the kernel
+    # 8-granule.
+    "top11": dict(terminal_count=360632, path_count=342846, label_count=72416,
                   embed=100, encode=100, batch=1024, contexts=200),
     # BASELINE.json config 4: java-large-scale synthetic
     "java-large": dict(terminal_count=1_100_000, path_count=1_300_000,
@@ -68,6 +75,14 @@ def main() -> None:
                     help="train: full step (the driver contract); infer: "
                          "hipGraph-captured batched code-vector export "
                          "(BASELINE config 5)")
+    ap.add_argument("--real-pipeline", action="store_true",
+                    help="route batches through the REAL input pipeline "
+                         "(SyntheticReader -> native epoch builder -> "
+                         "pinned-staged BatchIterator async H2D) instead "
+                         "of a pre-staged device pool")
+    ap.add_argument("--methods", type=int, default=605945,
+                    help="--real-pipeline corpus size (default: top11's "
+                         "method count)")
     args = ap.parse_args()
 
     cfg = CONFIGS[args.config]
@@ -106,12 +121,14 @@ def main() -> None:
     B, C = cfg["batch"], cfg["contexts"]
     rng = np.random.default_rng(1000 + ctx.rank)
     pool = []
-    for _ in range(args.pool):
-        s, p, e, y = synthetic_batch(rng, B, C, cfg["terminal_count"],
-                                     cfg["path_count"], cfg["label_count"])
-        pool.append(tuple(
-            torch.from_numpy(a).to(device) for a in (s, p, e, y)
-        ))
+    if not (args.real_pipeline and args.mode == "train"):
+        for _ in range(args.pool):
+            s, p, e, y = synthetic_batch(rng, B, C, cfg["terminal_count"],
+                                         cfg["path_count"],
+                                         cfg["label_count"])
+            pool.append(tuple(
+                torch.from_numpy(a).to(device) for a in (s, p, e, y)
+            ))
     class_weight = torch.ones(cfg["label_count"], device=device)
 
     if args.mode == "infer":
@@ -120,8 +137,35 @@ def main() -> None:
 
     model.train()
 
+    batch_iter = None
+    if args.real_pipeline:
+        from code2vec_amd.data.builder import DatasetBuilder
+        from code2vec_amd.data.synthetic import SyntheticReader
+        from code2vec_amd.engine.loader import BatchIterator
+
+        reader = SyntheticReader(
+            args.methods, cfg["terminal_count"], cfg["path_count"],
+            cfg["label_count"], max_contexts=2 * C, seed=7,
+        )
+        builder = DatasetBuilder(reader, opt, seed=123, rank=ctx.rank,
+                                 world_size=world)
+        loader = BatchIterator(builder.refresh_train_dataset(0), B,
+                               shuffle=True, seed=5, device=device)
+
+        def cycle():
+            while True:
+                for hb in loader:
+                    if hb["starts"].shape[0] == B:
+                        yield hb
+
+        batch_iter = cycle()
+
     def step(i: int) -> None:
-        s, p, e, y = pool[i % len(pool)]
+        if batch_iter is not None:
+            hb = next(batch_iter)
+            s, p, e, y = (hb["starts"], hb["paths"], hb["ends"], hb["label"])
+        else:
+            s, p, e, y = pool[i % len(pool)]
         ddp.zero_grad()
         outputs, _, _ = model(s, p, e, y)
         loss = model.loss(outputs, y, class_weight)
@@ -171,7 +215,12 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if on_gpu else "fp32",
-            "data": "synthetic (dense 200 contexts/method, random indices, random-init weights)",
+            "data": ("synthetic (random indices, random-init weights); "
+                     "REAL input pipeline: native epoch builder -> pinned "
+                     "staging -> async H2D per batch"
+                     if args.real_pipeline else
+                     "synthetic (dense 200 contexts/method, random indices, "
+                     "random-init weights)"),
             "config": {
                 "model": f"code2vec-{args.config}",
                 "global_batch": world * B,
@@ -183,7 +232,9 @@ def main() -> None:
                 "embed": cfg["embed"],
                 "encode": cfg["encode"],
                 "note": ("step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam"
-                        + ("; top11 label vocab size unpublished, 30k assumed" if args.config == "top11" else "")),
+                        + ("; top11 label vocab 72,416 derived via Heaps-law fit "
+                           "to the reference's real methods.txt name curve "
+                           "(tools/derive_label_vocab.py)" if args.config == "top11" else "")),
             },
         }))
 

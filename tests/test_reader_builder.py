@@ -392,3 +392,25 @@ def test_variable_task_equal_rows_across_ranks(tiny_corpus):
             b = DatasetBuilder(r, opt, seed=11, rank=rank, world_size=2)
             sizes.append(len(b.refresh_train_dataset(epoch=epoch)))
         assert sizes[0] == sizes[1], sizes
+
+
+def test_synthetic_reader_builder_integration():
+    """SyntheticReader (bench --real-pipeline) must drive DatasetBuilder
+    exactly like a file-backed CorpusReader."""
+    from code2vec_amd.data.synthetic import SyntheticReader
+
+    r = SyntheticReader(n_methods=100, terminal_count=50, path_count=40,
+                        label_count=20, max_contexts=30, seed=3)
+    opt = Option(terminal_count=50, path_count=40, label_count=20,
+                 max_path_length=16, terminal_embed_size=8,
+                 path_embed_size=8, encode_size=12)
+    b = DatasetBuilder(r, opt, seed=5)
+    d = b.refresh_train_dataset(epoch=0)
+    assert len(d) == len(b.train_items)
+    assert d.starts.shape == (len(d), 16)
+    assert (d.labels >= 0).all() and (d.labels < 20).all()
+    assert (d.starts < 50).all() and (d.paths < 40).all()
+    # resampling across epochs still applies
+    d0 = d.paths.copy()
+    d1 = b.refresh_train_dataset(epoch=1)
+    assert not np.array_equal(d0, d1.paths)
