@@ -61,6 +61,7 @@ class TrainerConfig:
         self.eval_method = args.eval_method
         self.random_seed = args.random_seed
         self.batch_size = args.batch_size
+        self.no_artifact_export = getattr(args, "no_artifact_export", False)
 
 
 class Trainer:
@@ -197,7 +198,8 @@ class Trainer:
                 if best_f1 is None or best_f1 < f1:
                     self._emit_metric("best_f1", f1, epoch)
                     best_f1 = f1
-                    if self.trial is None and self.ctx.is_rank0:
+                    if (self.trial is None and self.ctx.is_rank0
+                            and not cfg.no_artifact_export):
                         self._export_best(epoch)
 
                 # early stop: OR-logic reset (reference main.py:233-242)

@@ -75,6 +75,11 @@ def build_parser() -> argparse.ArgumentParser:
                    choices=["auto", "hip", "torch"])
     p.add_argument("--bucket_mb", type=int, default=32,
                    help="gradient all-reduce bucket size (MiB)")
+    p.add_argument("--no_artifact_export", action="store_true", default=False,
+                   help="MI355X extra: skip the best-F1 code.vec/checkpoint "
+                        "export (useful for pure throughput/F1 benchmark "
+                        "runs on top11-scale corpora where the text export "
+                        "dominates wall time)")
     p.add_argument("--init_model", type=str, default=None,
                    help="warm-start from a code2vec.model checkpoint "
                         "(reference state_dict format)")

@@ -28,10 +28,15 @@ def main():
     ap.add_argument("--max_contexts", type=int, default=200)
     ap.add_argument("--vars_per_method", type=int, default=2)
     ap.add_argument("--seed", type=int, default=1234)
+    ap.add_argument("--label_signal", type=float, default=0.0,
+                    help="fraction of each method's contexts drawn from its "
+                         "label's signature pool (makes names predictable "
+                         "from the context bag -> meaningful F1 curves)")
     a = ap.parse_args()
     spec = SyntheticSpec(n_methods=a.methods, n_terminals=a.terminals,
                          n_paths=a.paths, max_contexts=a.max_contexts,
-                         n_vars_per_method=a.vars_per_method, seed=a.seed)
+                         n_vars_per_method=a.vars_per_method, seed=a.seed,
+                         label_signal=a.label_signal)
     files = write_synthetic_corpus(a.out, spec)
     for k, v in files.items():
         print(f"{k}: {v}")
