@@ -210,7 +210,7 @@ class Code2VecHIP(nn.Module):
         ccv = y.view(B, C, self.EP)
         cv, attn = Fn.AttentionPool.apply(ccv, self.attention_a, starts, self.E)
         if opt.angular_margin_loss:
-            outputs = R.angular_margin_head(
+            outputs = Fn.angular_margin_head_hip(
                 cv.to(torch.bfloat16), self.output_weight, label,
                 self.cos_m, self.sin_m, opt.inverse_temp,
             )

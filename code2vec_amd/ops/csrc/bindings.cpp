@@ -69,6 +69,14 @@ void launch_head_bwd_prep(const long*, const float*, const float*,
                           const float*, const float*, float*, long,
                           hipStream_t);
 void launch_swizzle_cv(const void*, void*, long, hipStream_t);
+void launch_inv_rownorm(const void*, float*, long, hipStream_t);
+void launch_rowscale(const void*, const float*, void*, long, hipStream_t);
+void launch_angular_fwd(const void*, const void*, const long*, void*, void*,
+                        long, long, float, float, float, hipStream_t);
+void launch_angular_dcos(const void*, const void*, const long*, void*, long,
+                         long, float, float, float, hipStream_t);
+void launch_norm_project(const void*, const void*, const float*, void*,
+                         long, hipStream_t);
 void launch_head_bwd_dw(const void*, const void*, const float*, void*,
                         float*, long, long, hipStream_t);
 void launch_head_bwd_dcv(const void*, const void*, const float*, float*,
@@ -479,6 +487,68 @@ void head_bwd_dcv(torch::Tensor logits, torch::Tensor wt,
                       cur_stream());
 }
 
+// K11 angular-margin building blocks (all [N, 128] bf16 row spaces)
+void inv_rownorm(torch::Tensor x, torch::Tensor inv) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_DT(inv, torch::kFloat32);
+  TORCH_CHECK(x.size(1) == 128 && inv.numel() == x.size(0), "inv_rownorm");
+  launch_inv_rownorm(x.data_ptr(), inv.data_ptr<float>(), x.size(0),
+                     cur_stream());
+}
+
+void rowscale(torch::Tensor x, torch::Tensor inv, torch::Tensor out) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  TORCH_CHECK(x.size(1) == 128 && out.sizes() == x.sizes() &&
+              inv.numel() == x.size(0), "rowscale");
+  launch_rowscale(x.data_ptr(), inv.data_ptr<float>(), out.data_ptr(),
+                  x.size(0), cur_stream());
+}
+
+void angular_fwd(torch::Tensor ucv, torch::Tensor uw, torch::Tensor label,
+                 torch::Tensor out, torch::Tensor cos_out, double cos_m,
+                 double sin_m, double s) {
+  CHK_CUDA(ucv); CHK_CONTIG(ucv); CHK_DT(ucv, torch::kBFloat16);
+  CHK_CONTIG(uw); CHK_DT(uw, torch::kBFloat16);
+  CHK_DT(label, torch::kInt64);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  CHK_CONTIG(cos_out); CHK_DT(cos_out, torch::kBFloat16);
+  const long B = ucv.size(0), L = uw.size(0);
+  TORCH_CHECK(ucv.size(1) == 128 && uw.size(1) == 128 &&
+              out.size(0) == B && out.size(1) == L &&
+              cos_out.sizes() == out.sizes() && label.numel() == B,
+              "angular_fwd shapes");
+  launch_angular_fwd(ucv.data_ptr(), uw.data_ptr(), label.data_ptr<long>(),
+                     out.data_ptr(), cos_out.data_ptr(), B, L, (float)cos_m,
+                     (float)sin_m, (float)s, cur_stream());
+}
+
+void angular_dcos(torch::Tensor dout, torch::Tensor cosm,
+                  torch::Tensor label, torch::Tensor dcos, double cos_m,
+                  double sin_m, double s) {
+  CHK_CUDA(dout); CHK_CONTIG(dout); CHK_DT(dout, torch::kBFloat16);
+  CHK_CONTIG(cosm); CHK_DT(cosm, torch::kBFloat16);
+  CHK_CONTIG(dcos); CHK_DT(dcos, torch::kBFloat16);
+  const long B = dout.size(0), L = dout.size(1);
+  TORCH_CHECK(cosm.sizes() == dout.sizes() && dcos.sizes() == dout.sizes()
+              && label.numel() == B, "angular_dcos shapes");
+  launch_angular_dcos(dout.data_ptr(), cosm.data_ptr(),
+                      label.data_ptr<long>(), dcos.data_ptr(), B, L,
+                      (float)cos_m, (float)sin_m, (float)s, cur_stream());
+}
+
+void norm_project(torch::Tensor du, torch::Tensor u, torch::Tensor inv,
+                  torch::Tensor out) {
+  CHK_CUDA(du); CHK_CONTIG(du); CHK_DT(du, torch::kBFloat16);
+  CHK_CONTIG(u); CHK_DT(u, torch::kBFloat16);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  TORCH_CHECK(du.size(1) == 128 && u.sizes() == du.sizes() &&
+              out.sizes() == du.sizes() && inv.numel() == du.size(0),
+              "norm_project shapes");
+  launch_norm_project(du.data_ptr(), u.data_ptr(), inv.data_ptr<float>(),
+                      out.data_ptr(), du.size(0), cur_stream());
+}
+
 void transpose_w(torch::Tensor w, torch::Tensor wt) {
   CHK_CUDA(w); CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
   CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
@@ -585,6 +655,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_dgrad", &head_dgrad);
   m.def("head_bwd_prep", &head_bwd_prep);
   m.def("swizzle_cv", &swizzle_cv);
+  m.def("inv_rownorm", &inv_rownorm);
+  m.def("rowscale", &rowscale);
+  m.def("angular_fwd", &angular_fwd);
+  m.def("angular_dcos", &angular_dcos);
+  m.def("norm_project", &norm_project);
   m.def("head_bwd_dw", &head_bwd_dw);
   m.def("head_bwd_dcv", &head_bwd_dcv);
   m.def("transpose_w", &transpose_w);

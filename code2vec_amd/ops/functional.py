@@ -536,6 +536,81 @@ class FusedLogSoftmaxNLL(torch.autograd.Function):
         return dlogits, None, None
 
 
+class AngularMarginHead(torch.autograd.Function):
+    """K11: ArcFace-style head fully on HIP kernels (angular.hip).
+
+    Forward: per-row inverse norms -> unit matrices -> MFMA cosine GEMM
+    with the margin epilogue (emits scaled outputs AND the raw cosine for
+    backward).  Backward: dcos elementwise kernel; du = dcos @ U_w via the
+    split-K head_dgrad kernel; dv = dcos^T @ U_cv via head_wgrad; both
+    projected through the normalize backward (norm_project kernel).
+    Reference math: model/model.py:71-80 (th/mm computed-but-unused there).
+    EP = 128 only; callers gate and fall back to ops/reference.py.
+    """
+
+    @staticmethod
+    def forward(ctx, cv_bf16, w, label, cos_m, sin_m, s):
+        B = cv_bf16.shape[0]
+        L = w.shape[0]
+        dev = w.device
+        inv_c = torch.empty(B, dtype=torch.float32, device=dev)
+        inv_w = torch.empty(L, dtype=torch.float32, device=dev)
+        ext().inv_rownorm(cv_bf16, inv_c)
+        ext().inv_rownorm(w, inv_w)
+        ucv = torch.empty_like(cv_bf16)
+        uw = _scratch_bf16("ang_uw", (L, 128), dev)
+        ext().rowscale(cv_bf16, inv_c, ucv)
+        ext().rowscale(w, inv_w, uw)
+        out = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+        cos = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+        ext().angular_fwd(ucv, uw, label, out, cos, cos_m, sin_m, s)
+        ctx.save_for_backward(ucv, uw, inv_c, inv_w, cos, label)
+        ctx.consts = (cos_m, sin_m, s)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ucv, uw, inv_c, inv_w, cos, label = ctx.saved_tensors
+        cos_m, sin_m, s = ctx.consts
+        B, L = cos.shape
+        dev = cos.device
+        dcos = torch.empty_like(cos)
+        ext().angular_dcos(dout.contiguous().to(torch.bfloat16), cos, label,
+                           dcos, cos_m, sin_m, s)
+        # du = dcos @ U_w  (split-K head_dgrad; rocBLAS fallback off-shape)
+        if L % 8 == 0:
+            uwt = _scratch_bf16("ang_uwt", (128, L), dev)
+            ext().transpose_w(uw, uwt)
+            split = (L + 511) // 512
+            partials = _scratch_f32("ang_dcv", (split, B, 128), dev)
+            ext().head_dgrad(dcos, uwt, partials)
+            du = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
+            ext().slab_sum_bf16(partials, du)
+        else:
+            du = dcos @ uw
+        # dv = dcos^T @ U_cv  (head_wgrad; rocBLAS fallback off-shape)
+        dv = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
+        if B % 32 == 0 and L % 8 == 0:
+            ext().head_wgrad(dcos, ucv, dv)
+        else:
+            dv.copy_(dcos.t() @ ucv)
+        dcv = torch.empty_like(du)
+        dw = torch.empty_like(dv)
+        ext().norm_project(du, ucv, inv_c, dcv)
+        ext().norm_project(dv, uw, inv_w, dw)
+        return dcv, dw, None, None, None, None
+
+
+def angular_margin_head_hip(cv_bf16, w, label, cos_m, sin_m, s):
+    """Gated entry: HIP kernels when the shape fits, torch reference
+    otherwise (CPU, EP != 128)."""
+    from . import reference as R
+
+    if cv_bf16.is_cuda and cv_bf16.shape[1] == 128 and w.shape[1] == 128:
+        return AngularMarginHead.apply(cv_bf16, w, label, cos_m, sin_m, s)
+    return R.angular_margin_head(cv_bf16, w, label, cos_m, sin_m, s)
+
+
 def head_logits_with_stats(cv_bf16, w, bias):
     """Forward logits for the FUSED head+loss path (no autograd tracking —
     gradients flow through FusedHeadLoss instead).  Uses the custom MFMA

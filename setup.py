@@ -25,6 +25,7 @@ SRC = [
     "code2vec_amd/ops/csrc/head_fwd.hip",
     "code2vec_amd/ops/csrc/head_dgrad.hip",
     "code2vec_amd/ops/csrc/head_bwd.hip",
+    "code2vec_amd/ops/csrc/angular.hip",
     "code2vec_amd/ops/csrc/dgrad2.hip",
     "code2vec_amd/ops/csrc/dgrad.hip",
     "code2vec_amd/ops/csrc/head_wgrad.hip",

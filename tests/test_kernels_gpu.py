@@ -640,3 +640,57 @@ class TestFusedHeadLoss:
         with torch.no_grad():
             out_e, _, _ = m(s, p, e, y)
         assert not hasattr(out_e, "_c2v_fused_head")
+
+
+# ---------------------------------------------------------------------------
+class TestAngularMarginHead:
+    """K11 HIP kernels (angular.hip) vs the fp32 torch oracle."""
+
+    @pytest.mark.parametrize("B,L", [(64, 2048), (96, 1000), (33, 517)])
+    def test_forward_backward_vs_oracle(self, dev, B, L):
+        from code2vec_amd.ops.functional import AngularMarginHead
+
+        g = torch.Generator().manual_seed(29)
+        cv = (torch.randn(B, 128, generator=g) * 0.7).to(dev, torch.bfloat16)
+        w = (torch.randn(L, 128, generator=g) * 0.2).to(dev, torch.bfloat16)
+        label = torch.randint(0, L, (B,), generator=g).to(dev)
+        cos_m, sin_m, s = math.cos(0.5), math.sin(0.5), 30.0
+
+        cvh = cv.clone().requires_grad_(True)
+        wh = w.clone().requires_grad_(True)
+        out = AngularMarginHead.apply(cvh, wh, label, cos_m, sin_m, s)
+        dl = (torch.randn(B, L, generator=g) * 0.05).to(dev, torch.bfloat16)
+        out.backward(dl)
+
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        ref = R.angular_margin_head(cvr, wr, label, cos_m, sin_m, s)
+        ref.backward(dl.float())
+        assert relerr(out.float(), ref) < 3e-2
+        assert relerr(cvh.grad.float(), cvr.grad) < 4e-2
+        assert relerr(wh.grad.float(), wr.grad) < 4e-2
+
+    def test_model_path_uses_kernels(self, dev):
+        """Code2VecHIP with --angular_margin_loss goes through the HIP
+        AngularMarginHead (no torch head ops) and trains."""
+        from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+        from code2vec_amd.utils.options import Option
+        from code2vec_amd.ops import functional as Fn
+
+        opt = Option(terminal_count=400, path_count=300, label_count=512,
+                     max_path_length=12, terminal_embed_size=100,
+                     path_embed_size=100, encode_size=100, dropout_prob=0.0,
+                     angular_margin_loss=True)
+        g = torch.Generator().manual_seed(9)
+        m = Code2VecHIP(opt, init_logical_params(opt, g), device=dev).train()
+        s = torch.randint(1, 400, (16, 12), dtype=torch.int32, device=dev)
+        p = torch.randint(1, 300, (16, 12), dtype=torch.int32, device=dev)
+        e = torch.randint(1, 400, (16, 12), dtype=torch.int32, device=dev)
+        y = torch.randint(0, 512, (16,), device=dev)
+        out, _, _ = m(s, p, e, y)
+        assert out.grad_fn is not None
+        assert "AngularMarginHead" in type(out.grad_fn).__name__
+        loss = m.loss(out, y, torch.ones(512, device=dev))
+        loss.backward()
+        assert m.output_weight.grad is not None
+        assert m.terminal_embedding.grad is not None
