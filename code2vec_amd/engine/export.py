@@ -15,6 +15,8 @@ from typing import Optional
 
 import torch
 
+from ..ops import functional as Fn
+
 logger = logging.getLogger(__name__)
 
 
@@ -53,7 +55,7 @@ def write_code_vectors(
                     ids = batch["id"]
                     _, label, outputs, code_vector, _ = _forward(
                         model, batch, device, graphed)
-                    preds_prob, preds_label = torch.max(outputs.float(), dim=1)
+                    preds_prob, preds_label = Fn.row_max_argmax(outputs.detach())
                     cv = code_vector.float().cpu()
                     label_cpu = label.cpu()
                     preds_cpu = preds_label.cpu()
@@ -96,7 +98,7 @@ def print_sample(reader, model, loader, option, device) -> None:
     with torch.no_grad():
         for batch in loader:
             starts_d, label, outputs, _, attn = _forward(model, batch, device)
-            _, preds_label = torch.max(outputs.float(), dim=1)
+            _, preds_label = Fn.row_max_argmax(outputs.detach())
             starts = batch["starts"]
             paths = batch["paths"]
             ends = batch["ends"]

@@ -26,6 +26,7 @@ from . import metrics as M
 from .export import print_sample, write_code_vectors, write_vector_header
 from .loader import BatchIterator
 from .optim import FusedAdam
+from ..ops import functional as Fn
 
 logger = logging.getLogger(__name__)
 
@@ -324,7 +325,7 @@ class Trainer:
                 # deferred accumulation — no per-batch .item()/.cpu() sync
                 loss_acc += loss.detach().double()
                 expected.append(label)
-                actual.append(torch.argmax(outputs.float(), dim=1))
+                actual.append(Fn.row_max_argmax(outputs.detach())[1])
 
         if expected:
             expected = torch.cat(expected).cpu().tolist()

@@ -65,6 +65,8 @@ void launch_dgrad2(const void*, const void*, void*, long, long, hipStream_t);
 void launch_slab_sum_bf16(const float*, void*, int, long, hipStream_t);
 void launch_slab_sum_f32(const float*, float*, int, int, hipStream_t);
 void launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
+void launch_row_max_argmax(const void*, float*, long*, long, long,
+                           hipStream_t);
 void launch_head_bwd_prep(const long*, const float*, const float*,
                           const float*, const float*, float*, long,
                           hipStream_t);
@@ -549,6 +551,16 @@ void norm_project(torch::Tensor du, torch::Tensor u, torch::Tensor inv,
                       out.data_ptr(), du.size(0), cur_stream());
 }
 
+// K17: per-row (max value, argmax) over bf16 logits (torch.max ties)
+void row_max_argmax(torch::Tensor x, torch::Tensor vals, torch::Tensor idx) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
+  CHK_DT(vals, torch::kFloat32); CHK_DT(idx, torch::kInt64);
+  const long B = x.size(0), L = x.size(1);
+  TORCH_CHECK(vals.numel() == B && idx.numel() == B, "row_max_argmax");
+  launch_row_max_argmax(x.data_ptr(), vals.data_ptr<float>(),
+                        idx.data_ptr<long>(), B, L, cur_stream());
+}
+
 void transpose_w(torch::Tensor w, torch::Tensor wt) {
   CHK_CUDA(w); CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
   CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
@@ -655,6 +667,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_dgrad", &head_dgrad);
   m.def("head_bwd_prep", &head_bwd_prep);
   m.def("swizzle_cv", &swizzle_cv);
+  m.def("row_max_argmax", &row_max_argmax);
   m.def("inv_rownorm", &inv_rownorm);
   m.def("rowscale", &rowscale);
   m.def("angular_fwd", &angular_fwd);

@@ -36,7 +36,58 @@ __global__ __launch_bounds__(256) void slab_sum_f32_kernel(
   if (threadIdx.x == 0) out[col] = red[0] + red[1] + red[2] + red[3];
 }
 
+// K17: row max+argmax over bf16 logits (eval-path prediction).  One block
+// per row; ties resolve to the SMALLEST index (torch.max semantics).
+__global__ __launch_bounds__(256) void row_max_argmax_kernel(
+    const bf16* __restrict__ x, float* __restrict__ vals,
+    long* __restrict__ idx, long L) {
+  const bf16* row = x + (long)blockIdx.x * L;
+  float best = -3.4e38f;
+  long bi = 0;
+  for (long j = threadIdx.x; j < L; j += 256) {
+    const float v = bf2f(row[j]);
+    if (v > best || (v == best && j < bi)) {
+      best = v;
+      bi = j;
+    }
+  }
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off);
+    const long oi = (long)__shfl_xor((long long)bi, off);
+    if (ov > best || (ov == best && oi < bi)) {
+      best = ov;
+      bi = oi;
+    }
+  }
+  __shared__ float rv[4];
+  __shared__ long ri[4];
+  if (lane == 0) {
+    rv[wave] = best;
+    ri[wave] = bi;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < 4; ++w) {
+      if (rv[w] > best || (rv[w] == best && ri[w] < bi)) {
+        best = rv[w];
+        bi = ri[w];
+      }
+    }
+    vals[blockIdx.x] = best;
+    idx[blockIdx.x] = bi;
+  }
+}
+
 extern "C" {
+
+void launch_row_max_argmax(const void* x, float* vals, long* idx, long B,
+                           long L, hipStream_t stream) {
+  row_max_argmax_kernel<<<B, 256, 0, stream>>>((const bf16*)x, vals, idx,
+                                               L);
+}
 
 void launch_slab_sum_f32(const float* p, float* out, int S, int N,
                          hipStream_t stream) {

@@ -708,6 +708,20 @@ def fused_head_loss_supported(cv_bf16, w, training: bool) -> bool:
             and L < (1 << 24))  # y carried as f32 in the G recompute
 
 
+def row_max_argmax(logits: torch.Tensor):
+    """K17: per-row (max, argmax) for the eval/export prediction path —
+    the last eval op that went through a torch kernel.  Falls back to
+    torch.max off the HIP bf16 path."""
+    if logits.is_cuda and logits.dtype == torch.bfloat16 \
+            and logits.is_contiguous():
+        B = logits.shape[0]
+        vals = torch.empty(B, dtype=torch.float32, device=logits.device)
+        idx = torch.empty(B, dtype=torch.int64, device=logits.device)
+        ext().row_max_argmax(logits, vals, idx)
+        return vals, idx
+    return torch.max(logits.float(), dim=1)
+
+
 def adam_step(
     param: torch.Tensor,
     grad: torch.Tensor,

@@ -694,3 +694,26 @@ class TestAngularMarginHead:
         loss.backward()
         assert m.output_weight.grad is not None
         assert m.terminal_embedding.grad is not None
+
+
+class TestRowMaxArgmax:
+    def test_matches_torch_max(self, dev):
+        from code2vec_amd.ops.functional import row_max_argmax
+
+        g = torch.Generator().manual_seed(17)
+        for B, L in [(64, 30000), (33, 517), (8, 261000)]:
+            x = (torch.randn(B, L, generator=g)).to(dev, torch.bfloat16)
+            vals, idx = row_max_argmax(x)
+            rv, ri = torch.max(x.float(), dim=1)
+            assert torch.equal(idx, ri), (B, L)
+            assert torch.equal(vals, rv), (B, L)
+
+    def test_tie_break_first_index(self, dev):
+        from code2vec_amd.ops.functional import row_max_argmax
+
+        x = torch.zeros(4, 1000, dtype=torch.bfloat16, device=dev)
+        x[:, 100] = 2.0
+        x[:, 700] = 2.0  # tie: torch.max returns the FIRST index
+        _, idx = row_max_argmax(x)
+        assert torch.equal(idx, torch.full((4,), 100, dtype=torch.int64,
+                                           device=dev))
