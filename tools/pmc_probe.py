@@ -60,12 +60,49 @@ for _ in range(REPS):
 idx_se = torch.cat([starts.view(-1), ends.view(-1)])
 sorted_se, perm_se, counts_se = Fn._group_by_index(idx_se, T)
 dterm32 = torch.zeros(T, TS, dtype=torch.float32, device=dev)
-for _ in range(REPS):
-    ext().embed_scatter_sorted(sorted_se, perm_se, x, dterm32, M, KP, 0,
-                               TS + PS, 16)
 dterm = torch.empty(T, TS, dtype=torch.bfloat16, device=dev)
+flags_t = torch.zeros(T, dtype=torch.uint8, device=dev)
 for _ in range(REPS):
-    ext().cast_clear_rows(dterm32, counts_se, dterm)
+    ext().embed_scatter_sorted(sorted_se, perm_se, x, dterm32, dterm,
+                               flags_t, M, KP, 0, TS + PS, 16)
+for _ in range(REPS):
+    ext().cast_clear_rows(dterm32, counts_se, flags_t, dterm)
+
+# head forward + fused backward kernels at the derived top11 vocab
+L = 72416
+wout = (torch.randn(L, EP, generator=g, device=dev) * 0.05).to(torch.bfloat16)
+label = torch.randint(0, L, (B,), generator=g, device=dev)
+weight = torch.ones(L, device=dev)
+bias = torch.zeros(L, device=dev)
+logits = torch.empty(B, L, dtype=torch.bfloat16, device=dev)
+gx = (L + 255) // 256
+pm = torch.empty(gx, B, dtype=torch.float32, device=dev)
+ps = torch.empty_like(pm)
+for _ in range(REPS):
+    ext().head_fwd(cv.to(torch.bfloat16), wout, bias, logits, pm, ps)
+lse = torch.empty(B, dtype=torch.float32, device=dev)
+acc_f = torch.empty((B + 15) // 16, 2, dtype=torch.float32, device=dev)
+ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc_f)
+acc = torch.empty(2, dtype=torch.float32, device=dev)
+ext().slab_sum_f32(acc_f, acc)
+coef_lse = torch.zeros(B, 4, device=dev)
+g1s = torch.ones(1, device=dev)
+ext().head_bwd_prep(label, weight, acc, g1s, lse, coef_lse)
+cvb = cv.to(torch.bfloat16).contiguous()
+cvimg = torch.empty((B + 63) // 64 * 2, 8, 64, 8, dtype=torch.bfloat16,
+                    device=dev)
+ext().swizzle_cv(cvb, cvimg)
+dwh = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
+dbias = torch.empty(L, dtype=torch.float32, device=dev)
+for _ in range(REPS):
+    ext().head_bwd_dw(logits, cvimg, coef_lse, dwh, dbias)
+wt = torch.empty(128, L, dtype=torch.bfloat16, device=dev)
+ext().transpose_w(wout, wt)
+chunk = 1024
+dcv_p = torch.empty((L + chunk - 1) // chunk, B, 128, dtype=torch.float32,
+                    device=dev)
+for _ in range(REPS):
+    ext().head_bwd_dcv(logits, wt, coef_lse, dcv_p, chunk)
 
 p1 = term.view(-1)
 g1 = term.clone().view(-1)
