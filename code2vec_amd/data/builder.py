@@ -286,7 +286,9 @@ class DatasetBuilder:
         item_idx = _torch.from_numpy(np.ascontiguousarray(shard_idx, dtype=np.int64))
         starts, paths, ends = self._pinned_pool(pool_tag, N, C)
         seed = hash((self.seed, epoch, stream, self.rank)) & 0x7FFFFFFFFFFFFFFF
-        method_token_index = self.reader.terminal_vocab.stoi["@method_0"]
+        # -1 when the corpus has no @method_0 occurrences (matches nothing)
+        method_token_index = self.reader.terminal_vocab.stoi.get(
+            "@method_0", -1)
         _native.build_method_epoch(
             offsets_t, contexts_t, item_idx, starts, paths, ends,
             method_token_index, self.reader.QUESTION_TOKEN_INDEX, seed,
@@ -327,7 +329,8 @@ class DatasetBuilder:
         ends_rows: List[np.ndarray] = []
 
         if reader.infer_method:
-            method_token_index = reader.terminal_vocab.stoi["@method_0"]
+            method_token_index = reader.terminal_vocab.stoi.get(
+                "@method_0", -1)
             label_stoi = reader.label_vocab.stoi
             for item in items:
                 pcs = item.path_contexts

@@ -40,6 +40,11 @@ setup(
             extra_compile_args=["-O3", "-std=c++17", "-fopenmp"],
             extra_link_args=["-fopenmp"],
         ),
+        CppExtension(
+            name="code2vec_amd.data._c2v_extract",
+            sources=["code2vec_amd/data/csrc/extractor.cpp"],
+            extra_compile_args=["-O3", "-std=c++17"],
+        ),
         CUDAExtension(
             name="code2vec_amd.ops._c2v_hip",
             sources=SRC,

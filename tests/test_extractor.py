@@ -1,0 +1,143 @@
+"""Native Java -> path-context extractor (data/csrc/extractor.cpp) —
+capability parity with the reference's preprocessing notebook
+(create_path_contexts.ipynb cells 4-12)."""
+
+import os
+
+import pytest
+import torch  # noqa: F401  (loads libc10 for the extension)
+
+X = pytest.importorskip("code2vec_amd.data._c2v_extract")
+
+JAVA = """
+public class Calc {
+  private int count;
+  public int addTwice(int value, int bonus) {
+    int total = value + bonus;
+    for (int i = 0; i < 2; i++) {
+      total += compute(total);
+    }
+    if (total > 100) { return total - 1; }
+    while (total < 0) { total++; }
+    return addTwice(total, 0);
+  }
+  private int compute(int x) { return x * 2 + count; }
+  public int getCount() { return count; }
+  public void setCount(int c) { this.count = c; }
+  public abstract void pending(int a);
+  public String describe(String name) {
+    try {
+      return "x" + name.trim();
+    } catch (Exception e) {
+      throw new RuntimeException("bad", e);
+    }
+  }
+}
+"""
+
+
+def by_name(records):
+    return {r["name"]: r for r in records}
+
+
+def test_ignorable_rules():
+    r = by_name(X.extract_source(JAVA))
+    assert r["getCount"]["ignorable"]      # trivial getter
+    assert r["setCount"]["ignorable"]      # trivial setter
+    assert r["pending"]["ignorable"]       # abstract (no body)
+    assert not r["addTwice"]["ignorable"]
+    assert not r["describe"]["ignorable"]
+
+
+def test_anonymization_and_terminals():
+    r = by_name(X.extract_source(JAVA))["addTwice"]
+    al = dict(r["aliases"])
+    assert al == {"value": "@var_0", "bonus": "@var_1", "total": "@var_2",
+                  "i": "@var_3"}
+    terms = {t for c in r["contexts"] for t in (c[0], c[2])}
+    assert "@method_0" in terms            # own declaration + recursion
+    assert "@var_0" in terms and "@var_2" in terms
+    assert "@int_literal" in terms
+    assert "value" not in terms and "total" not in terms
+    # field names survive anonymization (lowercased at vocab time)
+    rc = by_name(X.extract_source(JAVA))["compute"]
+    cterms = {t for c in rc["contexts"] for t in (c[0], c[2])}
+    assert "count" in cterms
+
+
+def test_recursive_and_sibling_calls():
+    r = by_name(X.extract_source(JAVA))["addTwice"]
+    # calls to the method itself map to @method_0; calls to sibling class
+    # methods get @method_N aliases (MethodNameSpace, notebook cell 6)
+    call_terms = {t for c in r["contexts"] for t in (c[0], c[2])
+                  if t.startswith("@method")}
+    assert "@method_0" in call_terms
+    assert any(t.startswith("@method_") and t != "@method_0"
+               for t in call_terms)
+
+
+def test_path_length_and_width_limits():
+    short = by_name(X.extract_source(JAVA, 4, 3))["addTwice"]["contexts"]
+    default = by_name(X.extract_source(JAVA, 8, 3))["addTwice"]["contexts"]
+    wide = by_name(X.extract_source(JAVA, 8, 100))["addTwice"]["contexts"]
+    assert len(short) < len(default) <= len(wide)
+    for s, p, e in default:
+        n_nodes = p.count("↑") + p.count("↓") + 1
+        assert n_nodes <= 8, p
+
+
+def test_literals_and_strings():
+    r = by_name(X.extract_source(JAVA))["describe"]
+    terms = {t for c in r["contexts"] for t in (c[0], c[2])}
+    assert "@string_literal" in terms
+
+
+def test_dataset_roundtrip(tmp_path):
+    """extract_to_dataset files parse through CorpusReader and train."""
+    from code2vec_amd.data.builder import DatasetBuilder
+    from code2vec_amd.data.reader import CorpusReader
+    from code2vec_amd.models.code2vec import build_model
+    from code2vec_amd.utils.options import Option
+
+    src = tmp_path / "src"
+    src.mkdir()
+    for i in range(6):
+        (src / f"A{i}.java").write_text(JAVA.replace("Calc", f"Calc{i}"))
+    methods = tmp_path / "methods.txt"
+    with open(methods, "w") as f:
+        for i in range(6):
+            for m in ("addTwice", "compute", "describe", "getCount"):
+                f.write(f"A{i}.java\t{m}\n")
+    out = tmp_path / "ds"
+    out.mkdir()
+    stats = X.extract_to_dataset(str(methods), str(src), str(out), 8, 3)
+    assert stats["methods_written"] == 18  # getCount skipped as ignorable
+    assert stats["skipped_ignorable"] == 6
+    assert stats["skipped_unparseable"] == 0
+
+    reader = CorpusReader(str(out / "corpus.txt"),
+                          str(out / "path_idxs.txt"),
+                          str(out / "terminal_idxs.txt"))
+    assert len(reader.items) == 18
+    assert all(it.path_contexts.shape[0] > 0 for it in reader.items)
+    # aliases round-trip through the vars: section (reader keys by alias:
+    # aliases["@var_N"] = normalized original name)
+    assert any("@var_0" in it.aliases for it in reader.items)
+    assert any("value" in it.aliases.values() for it in reader.items)
+
+    opt = Option(terminal_count=len(reader.terminal_vocab),
+                 path_count=len(reader.path_vocab),
+                 label_count=len(reader.label_vocab),
+                 max_path_length=24, terminal_embed_size=12,
+                 path_embed_size=12, encode_size=16, dropout_prob=0.0)
+    b = DatasetBuilder(reader, opt, seed=3)
+    data = b.refresh_train_dataset(0)
+    model = build_model(opt, backend="torch")
+    s = torch.from_numpy(data.starts).long()
+    p = torch.from_numpy(data.paths).long()
+    e = torch.from_numpy(data.ends).long()
+    y = torch.from_numpy(data.labels)
+    outp, _, _ = model(s, p, e, y)
+    loss = model.loss(outp, y, torch.ones(opt.label_count))
+    loss.backward()
+    assert torch.isfinite(loss)
