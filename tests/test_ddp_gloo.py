@@ -245,7 +245,7 @@ def test_early_grad_callback_world2(tmp_path):
     ctx = mp.get_context("spawn")
     result_file = str(tmp_path / "cb.txt")
     procs = [ctx.Process(target=_worker_early_cb,
-                         args=(r, 2, 29533, result_file)) for r in range(2)]
+                         args=(r, 2, 29561, result_file)) for r in range(2)]
     for pr in procs:
         pr.start()
     for pr in procs:
