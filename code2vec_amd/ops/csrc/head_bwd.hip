@@ -261,6 +261,162 @@ __global__ __launch_bounds__(256) void swizzle_cv_kernel(
   }
 }
 
+// 128-label-tile variant of head_bwd_dw — MEASURED SLOWER and kept as an
+// A/B reference (C2V_HBDW128=<threshold> to enable): 125 vs 88 us at
+// L=72,416 and a wash (320 vs 326) at L=261k.  The hypothesis (halving
+// the block count amortizes per-block skeleton cost and the 256-B logits
+// row spans cut line amplification) loses to the doubled per-wave
+// B-fragment read + MFMA chain per stage between the same barriers.
+__global__ __launch_bounds__(512) void head_bwd_dw128_kernel(
+    const bf16* __restrict__ logits, const bf16* __restrict__ cvimg,
+    const float* __restrict__ coef_lse, bf16* __restrict__ dw,
+    float* __restrict__ dbias, long B, long L) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const long l0 = (long)blockIdx.x * 128;
+  const int krow = threadIdx.x >> 4;        // 0..31 (and +32: second chunk)
+  const int col8 = (threadIdx.x & 15) * 8;  // 0..120: label chunk start
+
+  __shared__ union {
+    bf16 img[2][2][8][HB_NSTR];  // [buf][ksub][label tile][image]
+    float red[32][129];
+  } sm;
+  __shared__ float red2[4][129];
+
+  const int n = col8 >> 4;
+  const int base_l = col8 & 15;
+  const int kgrp = (krow >> 3) & 3;  // same for krow and krow+32
+  const int jslot = krow & 7;
+  const long lc = l0 + col8;
+  const bool lok = lc < L;
+
+  f32x4 db0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 db1 = {0.f, 0.f, 0.f, 0.f};
+  const bf16x8 zero8 = {};
+
+  bf16x8 lv0, lv1;
+  f32x4 meta0, meta1;
+  bool rok0, rok1;
+
+#define DW1_LOAD(mb)                                                       \
+  do {                                                                     \
+    const long r0_ = (mb) + krow;                                          \
+    const long r1_ = r0_ + 32;                                             \
+    rok0 = r0_ < B && lok;                                                 \
+    rok1 = r1_ < B && lok;                                                 \
+    if (rok0) {                                                            \
+      meta0 = *(const f32x4*)(coef_lse + 4 * r0_);                         \
+      lv0 = *(const bf16x8*)(logits + r0_ * L + lc);                       \
+    }                                                                      \
+    if (rok1) {                                                            \
+      meta1 = *(const f32x4*)(coef_lse + 4 * r1_);                         \
+      lv1 = *(const bf16x8*)(logits + r1_ * L + lc);                       \
+    }                                                                      \
+  } while (0)
+
+#define DW1_G(gv, rok, meta, lv)                                           \
+  do {                                                                     \
+    gv = zero8;                                                            \
+    if (rok) {                                                             \
+      const float coef_ = meta[0];                                         \
+      const float lse_ = meta[1];                                          \
+      const long y_ = (long)meta[2];                                       \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
+        float g_ = coef_ * __expf(bf2f(lv[j]) - lse_);                     \
+        if (lc + j == y_) g_ -= coef_;                                     \
+        const bf16 gb_ = f2bf(g_);                                         \
+        gv[j] = gb_;                                                       \
+        if (j < 4) db0[j & 3] += bf2f(gb_);                                \
+        else db1[j & 3] += bf2f(gb_);                                      \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+#define DW1_WRITE(buf)                                                     \
+  do {                                                                     \
+    bf16x8 g0, g1;                                                         \
+    DW1_G(g0, rok0, meta0, lv0);                                           \
+    DW1_G(g1, rok1, meta1, lv1);                                           \
+    bf16* d0_ = sm.img[buf][0][n];                                         \
+    bf16* d1_ = sm.img[buf][1][n];                                         \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                        \
+      d0_[(base_l + j + kgrp * 16) * 8 + jslot] = g0[j];                   \
+      d1_[(base_l + j + kgrp * 16) * 8 + jslot] = g1[j];                   \
+    }                                                                      \
+  } while (0)
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = acc0, acc2 = acc0,
+        acc3 = acc0, acc4 = acc0, acc5 = acc0, acc6 = acc0, acc7 = acc0;
+
+  const int lt = wave;               // this wave's label tile (16 labels)
+  const bf16* cbase = cvimg + (long)lane * 8;
+
+  DW1_LOAD(0);
+  DW1_WRITE(0);
+  __syncthreads();
+  int buf = 0;
+  for (long mb = 0; mb < B; mb += HB_ROWS) {
+    const bool more = mb + HB_ROWS < B;
+    if (more) DW1_LOAD(mb + HB_ROWS);
+    const long kc0 = (mb >> 5) * 8 * 512;
+    const bf16x8 a0 = *(const bf16x8*)&sm.img[buf][0][lt][lane * 8];
+    const bf16x8 a1 = *(const bf16x8*)&sm.img[buf][1][lt][lane * 8];
+#define DW1_MM(ks, nt, accv)                                               \
+    {                                                                      \
+      const bf16x8 b_ =                                                    \
+          *(const bf16x8*)(cbase + kc0 + ((ks) * 8 + (nt)) * 512);         \
+      accv = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                      \
+          (ks) ? a1 : a0, b_, accv, 0, 0, 0);                              \
+    }
+    DW1_MM(0, 0, acc0) DW1_MM(0, 1, acc1) DW1_MM(0, 2, acc2)
+    DW1_MM(0, 3, acc3) DW1_MM(0, 4, acc4) DW1_MM(0, 5, acc5)
+    DW1_MM(0, 6, acc6) DW1_MM(0, 7, acc7)
+    DW1_MM(1, 0, acc0) DW1_MM(1, 1, acc1) DW1_MM(1, 2, acc2)
+    DW1_MM(1, 3, acc3) DW1_MM(1, 4, acc4) DW1_MM(1, 5, acc5)
+    DW1_MM(1, 6, acc6) DW1_MM(1, 7, acc7)
+#undef DW1_MM
+    if (more) DW1_WRITE(buf ^ 1);
+    buf ^= 1;
+    __syncthreads();
+  }
+#undef DW1_LOAD
+#undef DW1_G
+#undef DW1_WRITE
+
+  // dW: wave owns [16 labels x 128 EP]
+#define DW1_ST(nt, accv)                                                   \
+  _Pragma("unroll") for (int r = 0; r < 4; ++r) {                          \
+    const long lrow_ = l0 + lt * 16 + (lane >> 4) * 4 + r;                 \
+    if (lrow_ < L)                                                         \
+      dw[lrow_ * 128 + (nt) * 16 + (lane & 15)] = f2bf(accv[r]);           \
+  }
+  DW1_ST(0, acc0) DW1_ST(1, acc1) DW1_ST(2, acc2) DW1_ST(3, acc3)
+  DW1_ST(4, acc4) DW1_ST(5, acc5) DW1_ST(6, acc6) DW1_ST(7, acc7)
+#undef DW1_ST
+
+  // dbias: red[krow(32)][label(128)] then 4-way fixed-order combine
+#pragma unroll
+  for (int j = 0; j < 4; ++j) sm.red[krow][col8 + j] = db0[j];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) sm.red[krow][col8 + 4 + j] = db1[j];
+  __syncthreads();
+  {
+    const int lbl = threadIdx.x & 127;
+    const int kg0 = (threadIdx.x >> 7) * 8;
+    float p = 0.f;
+#pragma unroll
+    for (int kg = 0; kg < 8; ++kg) p += sm.red[kg0 + kg][lbl];
+    red2[threadIdx.x >> 7][lbl] = p;
+  }
+  __syncthreads();
+  if (threadIdx.x < 128 && l0 + threadIdx.x < L) {
+    float s = 0.f;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) s += red2[q][threadIdx.x];
+    dbias[l0 + threadIdx.x] = s;
+  }
+}
+
 // dcv split-K partials: head_dgrad.hip's kernel with A = G recomputed.
 __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
     const bf16* __restrict__ logits, const bf16* __restrict__ wt,
@@ -365,6 +521,18 @@ void launch_swizzle_cv(const void* cv, void* cvimg, long B,
 void launch_head_bwd_dw(const void* logits, const void* cvimg,
                         const float* coef_lse, void* dw, float* dbias,
                         long B, long L, hipStream_t stream) {
+  // 64-label tiles by default; the 128-tile variant measured slower at
+  // both top11 and java-large shapes (see the kernel comment) and is
+  // reachable for experiments via C2V_HBDW128=<L threshold>
+  const char* te = getenv("C2V_HBDW128");
+  const long thresh = te ? atol(te) : (1L << 60);
+  if (L >= thresh) {
+    const int grid = (int)((L + 127) / 128);
+    head_bwd_dw128_kernel<<<grid, 512, 0, stream>>>(
+        (const bf16*)logits, (const bf16*)cvimg, coef_lse, (bf16*)dw, dbias,
+        B, L);
+    return;
+  }
   const int grid = (int)((L + HB_LB - 1) / HB_LB);
   const char* ve = getenv("C2V_HBDW_VARIANT");
   const int variant = ve ? atoi(ve) : 0;
