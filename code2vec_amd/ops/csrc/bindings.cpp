@@ -70,7 +70,7 @@ void launch_row_max_argmax(const void*, float*, long*, long, long,
 void launch_head_bwd_prep(const long*, const float*, const float*,
                           const float*, const float*, float*, long,
                           hipStream_t);
-void launch_swizzle_cv(const void*, void*, long, hipStream_t);
+void launch_swizzle_cv(const void*, void*, long, long, hipStream_t);
 void launch_inv_rownorm(const void*, float*, long, hipStream_t);
 void launch_rowscale(const void*, const float*, void*, long, hipStream_t);
 void launch_angular_fwd(const void*, const void*, const long*, void*, void*,
@@ -435,14 +435,18 @@ void head_bwd_prep(torch::Tensor label, torch::Tensor weight,
                        coef_lse.data_ptr<float>(), B, cur_stream());
 }
 
-// cv [B, 128] -> B-fragment image [ceil(B/64)*2, 8, 64, 8].
+// [N, 128] -> B-fragment image [nchunk, 8, 64, 8]; nchunk (from the
+// output shape) must cover N at the consumer's stage granularity.
 void swizzle_cv(torch::Tensor cv, torch::Tensor cvimg) {
   CHK_CUDA(cv); CHK_CONTIG(cv); CHK_DT(cv, torch::kBFloat16);
   CHK_CONTIG(cvimg); CHK_DT(cvimg, torch::kBFloat16);
   const long B = cv.size(0);
-  TORCH_CHECK(cv.size(1) == 128, "cv must be [B, 128]");
-  TORCH_CHECK(cvimg.numel() == (B + 63) / 64 * 2 * 4096, "cvimg shape");
-  launch_swizzle_cv(cv.data_ptr(), cvimg.data_ptr(), B, cur_stream());
+  TORCH_CHECK(cv.size(1) == 128, "operand must be [N, 128]");
+  const long nchunk = cvimg.numel() / 4096;
+  TORCH_CHECK(nchunk * 4096 == cvimg.numel() && nchunk >= (B + 31) / 32,
+              "image shape must be [nchunk >= ceil(N/32), 8, 64, 8]");
+  launch_swizzle_cv(cv.data_ptr(), cvimg.data_ptr(), B, nchunk,
+                    cur_stream());
 }
 
 // Fused head+loss backward kernel 1: dw[L, 128] bf16 + dbias[L] f32 from
@@ -468,22 +472,24 @@ void head_bwd_dw(torch::Tensor logits, torch::Tensor cvimg,
 }
 
 // Fused head+loss backward kernel 2: dcv split-K partials
-// [ceil(L/chunk), B, 128] f32 with G recomputed; wt = W^T [128, L].
-void head_bwd_dcv(torch::Tensor logits, torch::Tensor wt,
+// [ceil(L/chunk), B, 128] f32 with G recomputed; wimg = swizzle_cv(W)
+// at 128-label granularity ([ceil(L/128)*4, 8, 64, 8]).
+void head_bwd_dcv(torch::Tensor logits, torch::Tensor wimg,
                   torch::Tensor coef_lse, torch::Tensor partials,
                   long chunk) {
   CHK_CUDA(logits); CHK_CONTIG(logits); CHK_DT(logits, torch::kBFloat16);
-  CHK_CONTIG(wt); CHK_DT(wt, torch::kBFloat16);
+  CHK_CONTIG(wimg); CHK_DT(wimg, torch::kBFloat16);
   CHK_CONTIG(coef_lse); CHK_DT(coef_lse, torch::kFloat32);
   CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
   const long B = logits.size(0), L = logits.size(1);
-  TORCH_CHECK(wt.size(0) == 128 && wt.size(1) == L, "wt must be [128, L]");
+  TORCH_CHECK(wimg.numel() == (L + 127) / 128 * 4 * 4096,
+              "wimg must be [ceil(L/128)*4, 8, 64, 8]");
   TORCH_CHECK(L % 8 == 0 && chunk % 128 == 0, "head_bwd_dcv shape gates");
   TORCH_CHECK(L < (1L << 24), "label index carried as f32 needs L < 2^24");
   TORCH_CHECK(coef_lse.numel() == 4 * B, "coef_lse must be [B, 4]");
   TORCH_CHECK(partials.numel() == (L + chunk - 1) / chunk * B * 128,
               "head_bwd_dcv partials shape");
-  launch_head_bwd_dcv(logits.data_ptr(), wt.data_ptr(),
+  launch_head_bwd_dcv(logits.data_ptr(), wimg.data_ptr(),
                       coef_lse.data_ptr<float>(),
                       partials.data_ptr<float>(), B, L, (int)chunk,
                       cur_stream());

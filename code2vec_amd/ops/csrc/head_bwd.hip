@@ -231,11 +231,12 @@ __global__ __launch_bounds__(512) void head_bwd_dw_kernel(
   }
 }
 
-// cv [B, 128] row-major -> MFMA B-fragment image
-// [ceil(B/32)][8 nt][64 lanes][8]: entry [kc][nt][l][j] =
-// cv[kc*32 + (l>>4)*8 + j][nt*16 + (l&15)] (zero-padded past B).  One
-// 32-row tile per block staged through LDS; output writes are contiguous
-// 16 B per thread.
+// [N, 128] row-major -> MFMA B-fragment image [nchunk][8 nt][64 lanes][8]:
+// entry [kc][nt][l][j] = x[kc*32 + (l>>4)*8 + j][nt*16 + (l&15)]
+// (zero-padded past N).  One 32-row tile per block staged through LDS;
+// output writes are contiguous 16 B per thread.  Used for BOTH operands
+// that appear as MFMA B fragments with a row index as k: cv [B, 128] in
+// head_bwd_dw and W [L, 128] in head_bwd_dcv.
 __global__ __launch_bounds__(256) void swizzle_cv_kernel(
     const bf16* __restrict__ cv, bf16* __restrict__ cvimg, long B) {
   const long kc = blockIdx.x;
@@ -418,8 +419,13 @@ __global__ __launch_bounds__(512) void head_bwd_dw128_kernel(
 }
 
 // dcv split-K partials: head_dgrad.hip's kernel with A = G recomputed.
+// The W operand arrives as a pre-swizzled B-fragment image (swizzle_cv
+// layout, 128-label-granular): per 128-label sub-stage the LDS staging is
+// ONE contiguous 32-KB copy (replacing a transposed-W staging whose
+// 256-B row reads at stride 2L were line-amplified, plus the per-step
+// transpose_w kernel entirely).
 __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
-    const bf16* __restrict__ logits, const bf16* __restrict__ wt,
+    const bf16* __restrict__ logits, const bf16* __restrict__ wimg,
     const float* __restrict__ coef_lse, float* __restrict__ partials, long B,
     long L, int chunk, int GYB) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -445,7 +451,8 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
   }
   const bf16* ap = logits + row * L + kj;
 
-  __shared__ bf16 wst[128][132];  // padded: conflict-free b128 reads
+  // one 128-label sub-stage of the W image: [4 kc][8 nt][64 lanes][8]
+  __shared__ bf16 wst[4 * 8 * 512];
 
   f32x4 acc[8];
 #pragma unroll
@@ -456,13 +463,10 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
   for (int s = 0; s < nsub; ++s) {
     const long l0 = (long)sc * chunk + (long)s * 128;
     if (l0 >= L) break;
-    for (int t = threadIdx.x; t < 128 * 16; t += 512) {
-      const int e = t >> 4;
-      const int c = t & 15;
-      const bf16x8 v = (l0 + c * 8 < L)
-          ? *(const bf16x8*)(wt + (long)e * L + l0 + c * 8) : zero8;
-      *(bf16x8*)&wst[e][c * 8] = v;
-    }
+    // contiguous 32-KB image copy (image is zero-padded past L)
+    const bf16* src = wimg + (l0 >> 5) * 4096;
+    for (int t = threadIdx.x; t < 2048; t += 512)
+      *(bf16x8*)&wst[t * 8] = *(const bf16x8*)(src + t * 8);
     __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
@@ -480,7 +484,7 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         const bf16x8 b =
-            *(const bf16x8*)&wst[nt * 16 + (lane & 15)][kk * 32 + kj];
+            *(const bf16x8*)&wst[(kk * 8 + nt) * 512 + lane * 8];
         acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt],
                                                           0, 0, 0);
       }
@@ -510,11 +514,12 @@ void launch_head_bwd_prep(const long* label, const float* weight,
       label, weight, acc_ws, gscale, lse, coef_lse, B);
 }
 
-void launch_swizzle_cv(const void* cv, void* cvimg, long B,
+void launch_swizzle_cv(const void* cv, void* cvimg, long B, long nchunk,
                        hipStream_t stream) {
-  // chunk count rounded to the dW kernel's 64-row stage granularity so
-  // tail-stage fragment reads always hit written (zero-padded) chunks
-  swizzle_cv_kernel<<<((B + 63) / 64) * 2, 256, 0, stream>>>(
+  // nchunk is caller-chosen (rounded to the consumer's stage granularity:
+  // 64 rows for head_bwd_dw's cv, 128 for head_bwd_dcv's W) so tail-stage
+  // fragment reads always hit written (zero-padded) chunks
+  swizzle_cv_kernel<<<(int)nchunk, 256, 0, stream>>>(
       (const bf16*)cv, (bf16*)cvimg, B);
 }
 
@@ -541,14 +546,14 @@ void launch_head_bwd_dw(const void* logits, const void* cvimg,
       L, variant);
 }
 
-void launch_head_bwd_dcv(const void* logits, const void* wt,
+void launch_head_bwd_dcv(const void* logits, const void* wimg,
                          const float* coef_lse, float* partials, long B,
                          long L, int chunk, hipStream_t stream) {
   const int GYB = (int)((B + 127) / 128);
   const int split = (int)((L + chunk - 1) / chunk);
   head_bwd_dcv_kernel<<<GYB * split, 512, 0, stream>>>(
-      (const bf16*)logits, (const bf16*)wt, coef_lse, partials, B, L, chunk,
-      GYB);
+      (const bf16*)logits, (const bf16*)wimg, coef_lse, partials, B, L,
+      chunk, GYB);
 }
 
 }  // extern "C"

@@ -686,14 +686,18 @@ class FusedHeadLoss(torch.autograd.Function):
         ext().head_bwd_dw(logits, cvimg, coef_lse, dw, dbias)
         # dcv (batch-major split-K): label chunk balances split-K slab
         # traffic against grid width (A/B-swept: 1024 beats 512 by ~30
-        # us/step at top11; 4096 at java-large keeps slabs ~33 MB)
+        # us/step at top11; 4096 at java-large keeps slabs ~33 MB).  W is
+        # pre-swizzled into the B-fragment image the kernel stages from
+        # (one contiguous 32-KB LDS copy per sub-stage; replaces the
+        # transpose_w kernel + line-amplified transposed-W staging).
         chunk = int(os.environ.get("C2V_HB_DCV_CHUNK", "0")) or (
-            1024 if L <= 65536 else 4096)
+            1024 if L <= 131072 else 4096)
         split = (L + chunk - 1) // chunk
-        wt = _scratch_bf16("head_wt", (128, L), dev)
-        ext().transpose_w(w, wt)
+        wimg = _scratch_bf16("head_wimg", ((L + 127) // 128 * 4, 8, 64, 8),
+                             dev)
+        ext().swizzle_cv(w, wimg)
         partials = _scratch_f32("head_fused_dcv", (split, B, 128), dev)
-        ext().head_bwd_dcv(logits, wt, coef_lse, partials, chunk)
+        ext().head_bwd_dcv(logits, wimg, coef_lse, partials, chunk)
         dcv = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
         ext().slab_sum_bf16(partials, dcv)
         return None, dcv, dw, dbias, None, None

@@ -191,13 +191,17 @@ def main():
         results["swizzle_cv"] = timeit(lambda: ext().swizzle_cv(cvb, cvimg))
         results["head_bwd_dw"] = timeit(
             lambda: ext().head_bwd_dw(logits, cvimg, coef_lse, dw, dbias))
-        chunk = 512 if L <= 65536 else 4096
+        chunk = int(os.environ.get("C2V_HB_DCV_CHUNK", "0")) or (
+            1024 if L <= 131072 else 4096)
         split = (L + chunk - 1) // chunk
         partials = torch.empty(split, B, 128, dtype=torch.float32, device=dev)
-        wt = torch.empty(128, L, dtype=torch.bfloat16, device=dev)
-        ext().transpose_w(wout, wt)
+        wimg = torch.empty((L + 127) // 128 * 4, 8, 64, 8,
+                           dtype=torch.bfloat16, device=dev)
+        ext().swizzle_cv(wout, wimg)
+        results["swizzle_w"] = timeit(lambda: ext().swizzle_cv(wout, wimg))
         results["head_bwd_dcv"] = timeit(
-            lambda: ext().head_bwd_dcv(logits, wt, coef_lse, partials, chunk))
+            lambda: ext().head_bwd_dcv(logits, wimg, coef_lse, partials,
+                                       chunk))
 
     for k, vv in results.items():
         print(f"{k:24s} {vv:10.1f} us")

@@ -96,13 +96,14 @@ dwh = torch.empty(L, 128, dtype=torch.bfloat16, device=dev)
 dbias = torch.empty(L, dtype=torch.float32, device=dev)
 for _ in range(REPS):
     ext().head_bwd_dw(logits, cvimg, coef_lse, dwh, dbias)
-wt = torch.empty(128, L, dtype=torch.bfloat16, device=dev)
-ext().transpose_w(wout, wt)
+wimg = torch.empty((L + 127) // 128 * 4, 8, 64, 8, dtype=torch.bfloat16,
+                   device=dev)
+ext().swizzle_cv(wout, wimg)
 chunk = 1024
 dcv_p = torch.empty((L + chunk - 1) // chunk, B, 128, dtype=torch.float32,
                     device=dev)
 for _ in range(REPS):
-    ext().head_bwd_dcv(logits, wt, coef_lse, dcv_p, chunk)
+    ext().head_bwd_dcv(logits, wimg, coef_lse, dcv_p, chunk)
 
 p1 = term.view(-1)
 g1 = term.clone().view(-1)
