@@ -26,6 +26,7 @@ void launch_count_indices(const int*, int*, long, hipStream_t);
 void launch_cast_clear_rows(float*, const int*, unsigned char*, void*, long,
                             int, hipStream_t);
 void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
+void launch_exclusive_scan(const int*, int*, int*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
                          float, unsigned long long, unsigned long long, int,
@@ -420,6 +421,20 @@ void head_dgrad(torch::Tensor dlogits, torch::Tensor wt,
 }
 
 // Fused head+loss backward prep: coef_lse[B, 4] = (coef_b, lse_b, y_b, 0).
+// exclusive prefix scan x[0..n-1] -> out[0..n-1] (int32); partial is a
+// scratch of >= ceil(n/1024) ints
+void exclusive_scan(torch::Tensor x, torch::Tensor partial,
+                    torch::Tensor out) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kInt32);
+  CHK_CONTIG(partial); CHK_DT(partial, torch::kInt32);
+  CHK_CONTIG(out); CHK_DT(out, torch::kInt32);
+  const long n = x.numel();
+  TORCH_CHECK(out.numel() >= n, "scan out too small");
+  TORCH_CHECK(partial.numel() >= (n + 1023) / 1024, "scan partial too small");
+  launch_exclusive_scan(x.data_ptr<int>(), partial.data_ptr<int>(),
+                        out.data_ptr<int>(), n, cur_stream());
+}
+
 void head_bwd_prep(torch::Tensor label, torch::Tensor weight,
                    torch::Tensor acc, torch::Tensor gscale, torch::Tensor lse,
                    torch::Tensor coef_lse) {
@@ -671,6 +686,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lsm_partial", &lsm_partial);
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
+  m.def("exclusive_scan", &exclusive_scan);
   m.def("head_bwd_prep", &head_bwd_prep);
   m.def("swizzle_cv", &swizzle_cv);
   m.def("row_max_argmax", &row_max_argmax);

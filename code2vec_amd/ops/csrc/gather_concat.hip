@@ -276,7 +276,7 @@ __global__ void scatter_group_kernel(const int* __restrict__ idx,
 // cache-line touch — 360k of them — and ran latency-bound at 37 us vs
 // ~8 us of actual byte traffic).
 __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
-                                       const int* __restrict__ counts,
+                                       int* __restrict__ counts,
                                        unsigned char* __restrict__ flags,
                                        bf16* __restrict__ out, long T, int S) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -297,6 +297,9 @@ __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
         valid && !boundary && rl != 0 && counts[rl] > 0;
     const bool zero_row = valid && !boundary && !interior;
     if (boundary) flags[rl] = 0;
+    // consume the histogram: the counts buffer is persistent and must be
+    // all-zero before the next step's count_indices pass
+    if (valid && counts[rl] != 0) counts[rl] = 0;
     const unsigned long long mb = __ballot(boundary);
     unsigned long long m = mb | __ballot(zero_row);
     while (m) {
@@ -321,7 +324,107 @@ __global__ void cast_clear_rows_kernel(float* __restrict__ dtable,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Exclusive prefix scan of counts[0..n-1] -> cursor[0..n-1] in three tiny
+// kernels (1024 elements per block; the middle kernel scans the <=1600
+// block sums serially in one block).  rocprim's lookback scan plus its
+// init kernel plus the torch.zeros fills cost ~21 us per table per step
+// at 360k-1.3M rows; this chain is ~6 us and needs no zeroed inputs.
+#define SCAN_B 1024
+
+__global__ __launch_bounds__(256) void scan_partials_kernel(
+    const int* __restrict__ x, int* __restrict__ partial, long n) {
+  const long b0 = (long)blockIdx.x * SCAN_B;
+  int v = 0;
+  for (int t = threadIdx.x; t < SCAN_B; t += 256) {
+    const long i = b0 + t;
+    v += i < n ? x[i] : 0;
+  }
+  // block reduce
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off);
+  __shared__ int red[4];
+  if (lane == 0) red[wave] = v;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    partial[blockIdx.x] = red[0] + red[1] + red[2] + red[3];
+}
+
+__global__ __launch_bounds__(1024) void scan_spine_kernel(
+    int* __restrict__ partial, int nb) {
+  // single block: exclusive scan of the block sums
+  __shared__ int carry;
+  if (threadIdx.x == 0) carry = 0;
+  __syncthreads();
+  // process in tiles of 1024 with a Hillis-Steele scan in LDS
+  __shared__ int tile[1024];
+  for (int base = 0; base < nb; base += 1024) {
+    const int i = base + threadIdx.x;
+    int v = i < nb ? partial[i] : 0;
+    tile[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+      int add = threadIdx.x >= off ? tile[threadIdx.x - off] : 0;
+      __syncthreads();
+      tile[threadIdx.x] += add;
+      __syncthreads();
+    }
+    const int incl = tile[threadIdx.x];
+    if (i < nb) partial[i] = carry + incl - v;  // exclusive
+    __syncthreads();
+    if (threadIdx.x == 1023) carry += tile[1023];
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(256) void scan_apply_kernel(
+    const int* __restrict__ x, const int* __restrict__ partial,
+    int* __restrict__ out, long n) {
+  const long b0 = (long)blockIdx.x * SCAN_B;
+  __shared__ int tile[SCAN_B];
+  for (int t = threadIdx.x; t < SCAN_B; t += 256) {
+    const long i = b0 + t;
+    tile[t] = i < n ? x[i] : 0;
+  }
+  __syncthreads();
+  // serial-ish per-thread scan: each thread owns 4 consecutive elems
+  // (SCAN_B/256); do a two-level exclusive scan in LDS
+  __shared__ int tsum[256];
+  {
+    const int t0 = threadIdx.x * 4;
+    int s0 = tile[t0] + tile[t0 + 1] + tile[t0 + 2] + tile[t0 + 3];
+    tsum[threadIdx.x] = s0;
+  }
+  __syncthreads();
+  // Hillis-Steele over the 256 thread sums
+  for (int off = 1; off < 256; off <<= 1) {
+    int add = threadIdx.x >= off ? tsum[threadIdx.x - off] : 0;
+    __syncthreads();
+    tsum[threadIdx.x] += add;
+    __syncthreads();
+  }
+  {
+    const int t0 = threadIdx.x * 4;
+    int run = partial[blockIdx.x] +
+              (threadIdx.x > 0 ? tsum[threadIdx.x - 1] : 0);
+    for (int k = 0; k < 4; ++k) {
+      const long i = b0 + t0 + k;
+      if (i < n) out[i] = run;
+      run += tile[t0 + k];
+    }
+  }
+}
+
 extern "C" {
+
+void launch_exclusive_scan(const int* x, int* partial, int* out, long n,
+                           hipStream_t stream) {
+  const int nb = (int)((n + SCAN_B - 1) / SCAN_B);
+  scan_partials_kernel<<<nb, 256, 0, stream>>>(x, partial, n);
+  scan_spine_kernel<<<1, 1024, 0, stream>>>(partial, nb);
+  scan_apply_kernel<<<nb, 256, 0, stream>>>(x, partial, out, n);
+}
 
 void launch_cast_clear_rows(float* dtable, const int* counts,
                             unsigned char* flags, void* out, long T, int S,
@@ -330,8 +433,8 @@ void launch_cast_clear_rows(float* dtable, const int* counts,
   const int wpb = block / WAVE;
   const long chunks = (T + WAVE - 1) / WAVE;
   const int grid = (int)min((chunks + wpb - 1) / wpb, (long)8192);
-  cast_clear_rows_kernel<<<grid, block, 0, stream>>>(dtable, counts, flags,
-                                                     (bf16*)out, T, S);
+  cast_clear_rows_kernel<<<grid, block, 0, stream>>>(
+      dtable, (int*)counts, flags, (bf16*)out, T, S);
 }
 
 void launch_count_indices(const int* idx, int* counts, long N,

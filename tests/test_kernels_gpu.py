@@ -717,3 +717,40 @@ class TestRowMaxArgmax:
         _, idx = row_max_argmax(x)
         assert torch.equal(idx, torch.full((4,), 100, dtype=torch.int64,
                                            device=dev))
+
+
+class TestExclusiveScan:
+    @pytest.mark.parametrize("n", [7, 1024, 4096, 360633, 1300001])
+    def test_matches_cumsum(self, dev, n):
+        from code2vec_amd.ops import ext
+
+        g = torch.Generator().manual_seed(n)
+        x = torch.randint(0, 9, (n,), generator=g, dtype=torch.int32).to(dev)
+        partial = torch.empty((n + 1023) // 1024, dtype=torch.int32,
+                              device=dev)
+        out = torch.empty(n, dtype=torch.int32, device=dev)
+        ext().exclusive_scan(x, partial, out)
+        ref = torch.zeros(n, dtype=torch.int64, device=dev)
+        ref[1:] = torch.cumsum(x.long()[:-1], 0)
+        assert torch.equal(out.long(), ref)
+
+    def test_pooled_counts_invariant_across_steps(self, dev):
+        """The pooled counting sort must give identical results on
+        repeated use: cast_clear_rows restores the all-zero histogram."""
+        from code2vec_amd.ops import functional as Fn
+        from code2vec_amd.ops import ext
+
+        T, S, M = 5000, 104, 4096
+        g = torch.Generator().manual_seed(3)
+        for step in range(3):
+            idx = torch.randint(0, T, (2 * M,), generator=g,
+                                dtype=torch.int32).to(dev)
+            a = Fn._group_by_index(idx, T, pool_tag="term")
+            b = Fn._group_by_index(idx, T)  # fresh buffers
+            assert torch.equal(a[0], b[0])
+            assert torch.equal(a[2], b[2])
+            # consume the pooled counts the way the scatter path does
+            dt32 = Fn._scratch_f32("term", (T, S), dev)
+            flags = Fn._scratch_flags("term", T, dev)
+            dbf = torch.empty(T, S, dtype=torch.bfloat16, device=dev)
+            ext().cast_clear_rows(dt32, a[2], flags, dbf)
