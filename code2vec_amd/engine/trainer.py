@@ -226,6 +226,10 @@ class Trainer:
         finally:
             if self.summary_writer is not None:
                 self.summary_writer.close()
+            # deregister the owned-grad callbacks: a later model in the
+            # same process could reuse this model's table addresses and
+            # silently hand its grads to a stale callback otherwise
+            self.ddp.close()
 
         return 1.0 - f1
 

@@ -12,6 +12,18 @@ import torch.multiprocessing as mp
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port():
+    """OS-assigned free TCP port (avoids TIME_WAIT collisions when the
+    suite runs twice in quick succession)."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def _worker(rank, world, port, result_file):
     import torch.distributed as dist
 
@@ -67,7 +79,7 @@ def test_ddp2_matches_single_process_large_batch(tmp_path, world):
     """DP=2/4 with batch shards + grad averaging == single process on
     the full batch (same init, fp32 => near-bitwise)."""
     ctx = mp.get_context("spawn")
-    port = 29531 + world
+    port = _free_port()
     result_file = str(tmp_path / "params.pt")
     procs = [ctx.Process(target=_worker, args=(r, world, port, result_file))
              for r in range(world)]
@@ -163,9 +175,10 @@ def _worker_trainer(rank, world, port, tmpdir, result_file, n_methods=40):
 def test_trainer_world2_runs(tmp_path):
     ctx = mp.get_context("spawn")
     result_file = str(tmp_path / "obj.txt")
+    port = _free_port()
     procs = [
         ctx.Process(target=_worker_trainer,
-                    args=(r, 2, 29532, str(tmp_path), result_file))
+                    args=(r, 2, port, str(tmp_path), result_file))
         for r in range(2)
     ]
     for pr in procs:
@@ -184,9 +197,10 @@ def test_trainer_world2_uneven_dataset(tmp_path):
     deadlock (the ADVICE.md stride-sharding finding).  Must complete."""
     ctx = mp.get_context("spawn")
     result_file = str(tmp_path / "obj.txt")
+    port = _free_port()
     procs = [
         ctx.Process(target=_worker_trainer,
-                    args=(r, 2, 29537, str(tmp_path), result_file, 41))
+                    args=(r, 2, port, str(tmp_path), result_file, 41))
         for r in range(2)
     ]
     for pr in procs:
@@ -244,8 +258,9 @@ def _worker_early_cb(rank, world, port, result_file):
 def test_early_grad_callback_world2(tmp_path):
     ctx = mp.get_context("spawn")
     result_file = str(tmp_path / "cb.txt")
+    port = _free_port()
     procs = [ctx.Process(target=_worker_early_cb,
-                         args=(r, 2, 29561, result_file)) for r in range(2)]
+                         args=(r, 2, port, result_file)) for r in range(2)]
     for pr in procs:
         pr.start()
     for pr in procs:
@@ -328,8 +343,9 @@ def _worker_owned(rank, world, port, result_file):
 def test_owned_chunked_allreduce_world2(tmp_path):
     ctx = mp.get_context("spawn")
     result_file = str(tmp_path / "owned.txt")
+    port = _free_port()
     procs = [ctx.Process(target=_worker_owned,
-                         args=(r, 2, 29551, result_file)) for r in range(2)]
+                         args=(r, 2, port, result_file)) for r in range(2)]
     for pr in procs:
         pr.start()
     for pr in procs:

@@ -16,6 +16,18 @@ pytestmark = pytest.mark.gpu
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port():
+    """OS-assigned free TCP port (avoids TIME_WAIT collisions when the
+    suite runs twice in quick succession)."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def _worker(rank, world, port, result_file, mode="plain"):
     import torch.distributed as dist
 
@@ -89,9 +101,10 @@ def test_hip_dp_world2_one_gpu(tmp_path):
     for i, mode in enumerate(("plain", "owned")):
         ctx = mp.get_context("spawn")
         result_file = str(tmp_path / f"dp_{mode}.txt")
+        port = _free_port()
         procs = [
             ctx.Process(target=_worker,
-                        args=(r, 2, 29541 + i, result_file, mode))
+                        args=(r, 2, port, result_file, mode))
             for r in range(2)
         ]
         for pr in procs:

@@ -410,3 +410,97 @@ def test_training_bitwise_deterministic(dev):
     a, b = run(), run()
     for name in a:
         assert torch.equal(a[name], b[name]), name
+
+
+def test_trainer_e2e_variable_task_gpu(dev, tmp_path):
+    """Full Trainer run on the HIP backend with the VARIABLE-name task
+    (infer_variable + shuffle_variable_indexes) — the numpy builder path
+    feeding real GPU training."""
+    from code2vec_amd.data.builder import DatasetBuilder
+    from code2vec_amd.data.reader import CorpusReader
+    from code2vec_amd.data.synthetic import SyntheticSpec, write_synthetic_corpus
+    from code2vec_amd.engine.trainer import Trainer, TrainerConfig
+    from code2vec_amd.models.code2vec import build_model, init_logical_params
+    from code2vec_amd.parallel.dist import DistContext
+    from code2vec_amd.utils.options import Option
+
+    files = write_synthetic_corpus(
+        str(tmp_path / "data"),
+        SyntheticSpec(n_methods=60, n_terminals=300, n_paths=200,
+                      max_contexts=20, n_vars_per_method=3, seed=11),
+    )
+    reader = CorpusReader(files["corpus_path"], files["path_idx_path"],
+                          files["terminal_idx_path"], infer_method=True,
+                          infer_variable=True, shuffle_variable_indexes=True)
+    opt = Option(terminal_count=len(reader.terminal_vocab),
+                 path_count=len(reader.path_vocab),
+                 label_count=len(reader.label_vocab),
+                 max_path_length=16, terminal_embed_size=100,
+                 path_embed_size=100, encode_size=100, dropout_prob=0.25,
+                 batch_size=16, device=dev)
+    builder = DatasetBuilder(reader, opt, seed=3)
+    model = build_model(opt, backend="hip",
+                        logical=init_logical_params(
+                            opt, torch.Generator().manual_seed(2)),
+                        device=dev)
+    ctx = DistContext(0, 1, 0, dev)
+
+    class A:
+        max_epoch = 2; lr = 0.01; beta_min = 0.9; beta_max = 0.999
+        weight_decay = 0.0; model_path = str(tmp_path / "out")
+        vectors_path = str(tmp_path / "out" / "code.vec")
+        test_result_path = str(tmp_path / "out" / "results.tsv")
+        env = None; print_sample_cycle = 0
+        eval_method = "subtoken"; random_seed = 3; batch_size = 16
+
+    trainer = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
+    obj = trainer.train()
+    assert 0.0 <= obj <= 1.0
+    # best-F1 artifacts written in the reference formats
+    assert (tmp_path / "out" / "code.vec").exists()
+    assert (tmp_path / "out" / "code2vec.model").exists()
+    assert (tmp_path / "out" / "results.tsv").exists()
+
+
+@pytest.mark.parametrize("method", ["exact", "ave_subtoken"])
+def test_trainer_eval_methods_gpu(dev, tmp_path, method):
+    """The two non-default eval methods through the HIP Trainer."""
+    from code2vec_amd.data.builder import DatasetBuilder
+    from code2vec_amd.data.reader import CorpusReader
+    from code2vec_amd.data.synthetic import SyntheticSpec, write_synthetic_corpus
+    from code2vec_amd.engine.trainer import Trainer, TrainerConfig
+    from code2vec_amd.models.code2vec import build_model, init_logical_params
+    from code2vec_amd.parallel.dist import DistContext
+    from code2vec_amd.utils.options import Option
+
+    files = write_synthetic_corpus(
+        str(tmp_path / "d"),
+        SyntheticSpec(n_methods=50, n_terminals=200, n_paths=150,
+                      max_contexts=12, seed=13),
+    )
+    reader = CorpusReader(files["corpus_path"], files["path_idx_path"],
+                          files["terminal_idx_path"])
+    opt = Option(terminal_count=len(reader.terminal_vocab),
+                 path_count=len(reader.path_vocab),
+                 label_count=len(reader.label_vocab),
+                 max_path_length=12, terminal_embed_size=100,
+                 path_embed_size=100, encode_size=100, dropout_prob=0.0,
+                 batch_size=16, eval_method=method, device=dev)
+    builder = DatasetBuilder(reader, opt, seed=3)
+    model = build_model(opt, backend="hip",
+                        logical=init_logical_params(
+                            opt, torch.Generator().manual_seed(4)),
+                        device=dev)
+    ctx = DistContext(0, 1, 0, dev)
+
+    class A:
+        max_epoch = 1; lr = 0.01; beta_min = 0.9; beta_max = 0.999
+        weight_decay = 0.0; model_path = str(tmp_path / "o")
+        vectors_path = str(tmp_path / "o" / "code.vec")
+        test_result_path = None; env = None; print_sample_cycle = 0
+        eval_method = method; random_seed = 3; batch_size = 16
+        no_artifact_export = True
+
+    trainer = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
+    obj = trainer.train()
+    assert 0.0 <= obj <= 1.0
