@@ -237,3 +237,67 @@ def test_seed_determinism_end_to_end(tiny_corpus):
         return loss
 
     assert one_loss() == one_loss()
+
+
+def test_find_hyperparams_with_stub_optuna(tmp_path, tiny_corpus, monkeypatch):
+    """HPO wiring (reference main.py:429-488) exercised end to end with a
+    minimal optuna stand-in (optuna itself is not installed here): the
+    objective must mutate Option in place, train, report per-epoch values
+    and honor pruning."""
+    import sys
+    import types
+
+    import main as main_mod
+
+    calls = {"suggest": [], "reports": 0, "pruned": 0}
+
+    class Trial:
+        def __init__(self, n):
+            self.n = n
+
+        def suggest_float(self, name, lo, hi, log=False):
+            calls["suggest"].append(name)
+            return float(lo)
+
+        def report(self, value, step):
+            calls["reports"] += 1
+
+        def should_prune(self):
+            # prune the second trial after its first epoch
+            return self.n == 1 and calls["reports"] > 0
+
+    class TrialPruned(Exception):
+        pass
+
+    class Study:
+        best_params = {"adam_lr": 1e-5}
+        best_value = 0.5
+
+        def optimize(self, objective, n_trials):
+            for n in range(n_trials):
+                try:
+                    objective(Trial(n))
+                except TrialPruned:
+                    calls["pruned"] += 1
+
+    stub = types.ModuleType("optuna")
+    stub.TrialPruned = TrialPruned
+    stub.create_study = lambda pruner=None: Study()
+    stub.pruners = types.SimpleNamespace(MedianPruner=lambda: None)
+    monkeypatch.setitem(sys.modules, "optuna", stub)
+
+    main_mod.main([
+        "--corpus_path", tiny_corpus["corpus_path"],
+        "--path_idx_path", tiny_corpus["path_idx_path"],
+        "--terminal_idx_path", tiny_corpus["terminal_idx_path"],
+        "--find_hyperparams", "--num_trials", "2", "--max_epoch", "2",
+        "--no_cuda", "--encode_size", "16", "--terminal_embed_size", "8",
+        "--path_embed_size", "8", "--batch_size", "8",
+        "--model_path", str(tmp_path / "out"),
+        "--vectors_path", str(tmp_path / "out" / "code.vec"),
+    ])
+    # same search space as the reference (main.py:447-456)
+    assert set(calls["suggest"]) == {"encode_size", "dropout_prob",
+                                     "batch_size", "adam_lr", "weight_decay"}
+    assert calls["reports"] >= 2   # per-epoch trial.report
+    assert calls["pruned"] == 1    # TrialPruned raised and handled
