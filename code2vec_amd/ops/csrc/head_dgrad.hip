@@ -121,17 +121,24 @@ __global__ __launch_bounds__(256) void transpose_w_kernel(
 }
 
 // partials[S][N] f32 -> out[N] bf16 (slab reduce + cast in one pass).
-// Scalar per thread: N is small (B*EP ~ 131k), so a vectorized variant
-// leaves half the CUs idle — more threads beat wider loads here.
+// Two independent accumulator chains per thread (i and i+stride): the
+// serial per-thread load chain over S slabs was latency-bound at ~2 TB/s
+// with 70+ slabs; doubling outstanding loads recovers most of it while
+// keeping the full thread count (a wide-vector variant idles CUs).
 __global__ __launch_bounds__(256) void slab_sum_bf16_kernel(
     const float* __restrict__ partials, bf16* __restrict__ out, int S,
     long N) {
+  const long half = (N + 1) / 2;
   const long i = (long)blockIdx.x * 256 + threadIdx.x;
-  if (i >= N) return;
-  float acc = 0.f;
-  for (int s = 0; s < S; ++s)
+  if (i >= half) return;
+  const long i2 = i + half;
+  float acc = 0.f, acc2 = 0.f;
+  for (int s = 0; s < S; ++s) {
     acc += __builtin_nontemporal_load(partials + s * N + i);
+    if (i2 < N) acc2 += __builtin_nontemporal_load(partials + s * N + i2);
+  }
   out[i] = f2bf(acc);
+  if (i2 < N) out[i2] = f2bf(acc2);
 }
 
 extern "C" {
@@ -144,7 +151,8 @@ void launch_transpose_w(const void* w, void* wt, long L,
 
 void launch_slab_sum_bf16(const float* partials, void* out, int S, long N,
                           hipStream_t stream) {
-  slab_sum_bf16_kernel<<<(N + 255) / 256, 256, 0, stream>>>(
+  const long half = (N + 1) / 2;
+  slab_sum_bf16_kernel<<<(half + 255) / 256, 256, 0, stream>>>(
       partials, (bf16*)out, S, N);
 }
 

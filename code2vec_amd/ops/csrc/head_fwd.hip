@@ -245,8 +245,10 @@ __global__ __launch_bounds__(512) void head_fwd_kernel(
   __syncthreads();
 
   // store: 8 batch rows per wave, 8 lanes x 16 B = one full 128-B line
-  // per row per pass; nontemporal — logits are 61 MB (far beyond L2) and
-  // not re-read until the loss backward
+  // per row per pass.  Nontemporal by default (logits far exceed L2);
+  // C2V_HF_NT=0 uses normal stores so the tail of the matrix stays in L2
+  // for the fused backward that reads logits right after (A/B knob —
+  // the NT choice predates the recompute-G backward).
   const long cend = (L - (long)bx * HF_LABS < HF_LABS)
                         ? L - (long)bx * HF_LABS : HF_LABS;
   const int bcr = wave * 8 + (lane >> 3);
@@ -257,8 +259,11 @@ __global__ __launch_bounds__(512) void head_fwd_kernel(
     if (grow < B && lo < cend && !(variant & 1)) {
       bf16* dst = out + grow * L + (long)bx * HF_LABS + lo;
       if (lo + 8 <= cend) {
-        __builtin_nontemporal_store(*(const bf16x8*)&patch[bcr][lo],
-                                    (bf16x8*)dst);
+        if (variant & 32)
+          *(bf16x8*)dst = *(const bf16x8*)&patch[bcr][lo];
+        else
+          __builtin_nontemporal_store(*(const bf16x8*)&patch[bcr][lo],
+                                      (bf16x8*)dst);
       } else {
         for (int j = 0; j < cend - lo; ++j) dst[j] = patch[bcr][lo + j];
       }
@@ -353,8 +358,12 @@ void launch_head_fwd(const void* cv, const void* w, const float* bias,
   const int GYB = (int)((B + HF_BATCH - 1) / HF_BATCH);
   const dim3 grid((long)GXL * GYB);
   const int stats = pm != nullptr;
+  // normal stores by default since the fused recompute-G backward reads
+  // logits right after (A/B: 1.403 vs 1.407 ms/step at L=72k; the NT
+  // choice predates that backward).  C2V_HF_NT=1 restores nontemporal.
   static const int variant =
-      getenv("C2V_HF_VARIANT") ? atoi(getenv("C2V_HF_VARIANT")) : 0;
+      (getenv("C2V_HF_VARIANT") ? atoi(getenv("C2V_HF_VARIANT")) : 0) |
+      ((getenv("C2V_HF_NT") && getenv("C2V_HF_NT")[0] == '1') ? 0 : 32);
 #define HFCASE(nkt, st)                                                      \
   head_fwd_kernel<nkt, st><<<grid, 512, 0, stream>>>(                        \
       (const bf16*)cv, (const bf16*)w, bias, (bf16*)out, pm, ps, B, L, EP,   \

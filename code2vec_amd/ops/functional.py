@@ -30,11 +30,12 @@ _SCATTER_R = int(os.environ.get("C2V_SCATTER_R", "16"))
 FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
 # custom output-head forward (+ fused loss statistics); C2V_HEAD_FWD=0
 # falls back to hipBLASLt linear + full-pass loss forward.  Measured
-# faster up to mid-size label vocabs (top11 L=30k: 69 -> 45 us for the
-# head+loss pair) but ~100 us/step slower at java-large's L=261k, so
+# faster through the derived top11 vocab (with normal stores + the fused
+# backward: 1.403 vs 1.412 ms/step at L=72,416) but still slower at
+# java-large's L=261k (strided W fragment reads stream from HBM), so
 # vocabs past C2V_HEAD_FWD_MAXL take the library path.
 _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
-_HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "65536"))
+_HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "98304"))
 # custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
 # combiner dgrad through dgrad2.hip (C2V_DGRAD2=0 -> rocBLAS)
