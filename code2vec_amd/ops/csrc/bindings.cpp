@@ -85,7 +85,8 @@ void launch_head_bwd_dw(const void*, const void*, const float*, void*,
 void launch_head_bwd_dcv(const void*, const void*, const float*, float*,
                          long, long, int, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
-                     float*, long, long, int, hipStream_t);
+                     float*, long, long, int, int, hipStream_t);
+void launch_swizzle_a(const void*, void*, long, long, hipStream_t);
 void launch_lsm_finalize(const void*, const float*, const float*, const long*,
                          const float*, float*, float*, int, long, int,
                          hipStream_t);
@@ -377,7 +378,49 @@ void head_fwd(torch::Tensor cv, torch::Tensor w, torch::Tensor bias,
     psp = ps.data_ptr<float>();
   }
   launch_head_fwd(cv.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
-                  out.data_ptr(), pmp, psp, B, L, EP, cur_stream());
+                  out.data_ptr(), pmp, psp, B, L, EP, /*aimg=*/0,
+                  cur_stream());
+}
+
+// head_fwd consuming the swizzle_a W image ([ceil(L/256)*16, 4, 64, 8]);
+// out/pm/ps contracts as head_fwd
+void head_fwd_img(torch::Tensor cv, torch::Tensor wimg, torch::Tensor bias,
+                  torch::Tensor out, torch::Tensor pm, torch::Tensor ps,
+                  int64_t L) {
+  CHK_CUDA(cv); CHK_CONTIG(cv); CHK_DT(cv, torch::kBFloat16);
+  CHK_CONTIG(wimg); CHK_DT(wimg, torch::kBFloat16);
+  CHK_DT(bias, torch::kFloat32); CHK_CONTIG(bias);
+  CHK_CONTIG(out); CHK_DT(out, torch::kBFloat16);
+  const long B = cv.size(0);
+  const int EP = cv.size(1);
+  TORCH_CHECK(EP == 128, "head_fwd_img is EP=128 only");
+  TORCH_CHECK(wimg.numel() == (L + 255) / 256 * 16 * 2048,
+              "wimg must be [ceil(L/256)*16, 4, 64, 8]");
+  TORCH_CHECK(out.size(0) == B && out.size(1) == L && bias.numel() == L,
+              "head_fwd_img shapes");
+  float* pmp = nullptr;
+  float* psp = nullptr;
+  if (pm.defined() && pm.numel() > 0) {
+    const long GXL = (L + 255) / 256;
+    CHK_DT(pm, torch::kFloat32); CHK_CONTIG(pm);
+    TORCH_CHECK(pm.numel() == B * GXL && ps.numel() == B * GXL, "partials");
+    pmp = pm.data_ptr<float>();
+    psp = ps.data_ptr<float>();
+  }
+  launch_head_fwd(cv.data_ptr(), wimg.data_ptr(), bias.data_ptr<float>(),
+                  out.data_ptr(), pmp, psp, B, L, EP, /*aimg=*/1,
+                  cur_stream());
+}
+
+void swizzle_a(torch::Tensor w, torch::Tensor wimg) {
+  CHK_CUDA(w); CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
+  CHK_CONTIG(wimg); CHK_DT(wimg, torch::kBFloat16);
+  const long L = w.size(0);
+  TORCH_CHECK(w.size(1) == 128, "w must be [L, 128]");
+  const long nrt = wimg.numel() / 2048;
+  TORCH_CHECK(nrt * 2048 == wimg.numel() && nrt >= (L + 15) / 16,
+              "wimg must be [nrt >= ceil(L/16), 4, 64, 8]");
+  launch_swizzle_a(w.data_ptr(), wimg.data_ptr(), L, nrt, cur_stream());
 }
 
 void logsoftmax_nll_finalize(torch::Tensor logits, torch::Tensor pm,
@@ -682,6 +725,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad", &wgrad);
   m.def("colsum_bf16", &colsum_bf16);
   m.def("head_fwd", &head_fwd);
+  m.def("head_fwd_img", &head_fwd_img);
+  m.def("swizzle_a", &swizzle_a);
   m.def("logsoftmax_nll_finalize", &logsoftmax_nll_finalize);
   m.def("lsm_partial", &lsm_partial);
   m.def("head_wgrad", &head_wgrad);

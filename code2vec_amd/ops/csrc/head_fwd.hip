@@ -43,7 +43,10 @@
 // variant: phase-isolation bitmask for perf experiments (C2V_HF_VARIANT):
 //   1 = skip global stores, 4 = skip B (cv) loads, 8 = skip A (w) loads,
 //   16 = skip the LDS transpose writes
-template <int NKT, int STATS>  // NKT = EP/32 when compile-time (0 = runtime)
+// AIMG=1: the A operand arrives as the swizzle_a fragment image (wimg
+// aliasing the w pointer) — contiguous 1-KB wave reads instead of
+// 256-B-strided per-lane row loads (the large-L fix; EP=128 only).
+template <int NKT, int STATS, int AIMG = 0>
 __global__ __launch_bounds__(512) void head_fwd_kernel(
     const bf16* __restrict__ cv, const bf16* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ out,
@@ -69,13 +72,14 @@ __global__ __launch_bounds__(512) void head_fwd_kernel(
   const int NK = NKT ? NKT : EP / 32;
 
   // A rows = labels (W rows), B cols = batch rows (cv rows); both k-runs
-  // contiguous
+  // contiguous.  AIMG: per-(row-tile, k-chunk) image reads instead.
   const long arow0 = lab0 + (lane & 15);
   const long arow1 = arow0 + 16;
-  const bf16* ap0 = w + arow0 * EP + kj;
-  const bf16* ap1 = w + arow1 * EP + kj;
-  const bool a0ok = arow0 < L && !(variant & 8);
-  const bool a1ok = arow1 < L && !(variant & 8);
+  const bf16* ap0 = AIMG ? w + ((lab0 >> 4) * 4) * 512 + lane * 8
+                         : w + arow0 * EP + kj;
+  const bf16* ap1 = AIMG ? ap0 + 4 * 512 : w + arow1 * EP + kj;
+  const bool a0ok = (AIMG || arow0 < L) && !(variant & 8);
+  const bool a1ok = (AIMG || arow1 < L) && !(variant & 8);
   const long bcol = b0 + (lane & 15);  // + nt*16 per col-tile
 
   f32x4 acc[2][4];
@@ -93,8 +97,10 @@ __global__ __launch_bounds__(512) void head_fwd_kernel(
   __shared__ bf16 smem2d[HF_BATCH][HF_LABS + 8];
   __shared__ float wstat[8][HF_BATCH][2];
 
-#define HF_LD(v, ok, p, kk) \
-  const bf16x8 v = (ok) ? *(const bf16x8*)((p) + (kk)*32) : zero8;
+#define HF_LD(v, ok, p, kk)                                          \
+  const bf16x8 v =                                                   \
+      (ok) ? *(const bf16x8*)((p) + (AIMG ? (kk) * 512 : (kk) * 32)) \
+           : zero8;
 #define HF_LB(v, nt, kk) \
   const bf16x8 v = \
       *(const bf16x8*)&smem2d[nt * 16 + (lane & 15)][(kk)*32 + kj];
@@ -349,29 +355,73 @@ __global__ __launch_bounds__(1024) void lsm_finalize_kernel(
   }
 }
 
+// W [L, 128] row-major -> MFMA A-fragment image [nrt][4 kc][64 lanes][8]:
+// entry [rt][kc][l][j] = w[rt*16 + (l&15)][kc*32 + (l>>4)*8 + j]
+// (zero-padded past L).  Makes head_fwd's A-operand reads contiguous
+// 1-KB wave reads at large L, where the per-lane 256-B-strided row loads
+// stream W from HBM inefficiently (variant-isolated ~80 us at L=261k).
+__global__ __launch_bounds__(256) void swizzle_a_kernel(
+    const bf16* __restrict__ w, bf16* __restrict__ wimg, long L) {
+  const long rt = blockIdx.x;
+  __shared__ bf16 tile[16][132];
+  const bf16x8 zero8 = {};
+  for (int t = threadIdx.x; t < 16 * 16; t += 256) {
+    const int r = t >> 4;
+    const int c = t & 15;
+    const long row = rt * 16 + r;
+    const bf16x8 v =
+        (row < L) ? *(const bf16x8*)(w + row * 128 + c * 8) : zero8;
+    *(bf16x8*)&tile[r][c * 8] = v;
+  }
+  __syncthreads();
+  // 256 outputs of 16 B: (kc, lane) pairs, one per thread
+  {
+    const int kc = threadIdx.x >> 6;
+    const int l = threadIdx.x & 63;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = tile[l & 15][kc * 32 + (l >> 4) * 8 + j];
+    *(bf16x8*)(wimg + ((rt * 4 + kc) * 64 + (long)l) * 8) = v;
+  }
+}
+
 extern "C" {
+
+void launch_swizzle_a(const void* w, void* wimg, long L, long nrt,
+                      hipStream_t stream) {
+  swizzle_a_kernel<<<(int)nrt, 256, 0, stream>>>((const bf16*)w,
+                                                 (bf16*)wimg, L);
+}
 
 void launch_head_fwd(const void* cv, const void* w, const float* bias,
                      void* out, float* pm, float* ps, long B, long L, int EP,
-                     hipStream_t stream) {
+                     int aimg, hipStream_t stream) {
   const int GXL = (int)((L + HF_LABS - 1) / HF_LABS);
   const int GYB = (int)((B + HF_BATCH - 1) / HF_BATCH);
   const dim3 grid((long)GXL * GYB);
   const int stats = pm != nullptr;
-  // normal stores by default since the fused recompute-G backward reads
-  // logits right after (A/B: 1.403 vs 1.407 ms/step at L=72k; the NT
-  // choice predates that backward).  C2V_HF_NT=1 restores nontemporal.
-  static const int variant =
-      (getenv("C2V_HF_VARIANT") ? atoi(getenv("C2V_HF_VARIANT")) : 0) |
-      ((getenv("C2V_HF_NT") && getenv("C2V_HF_NT")[0] == '1') ? 0 : 32);
-#define HFCASE(nkt, st)                                                      \
-  head_fwd_kernel<nkt, st><<<grid, 512, 0, stream>>>(                        \
+  // store mode by size: NORMAL stores up to ~the L2 scale (the fused
+  // recompute-G backward reads logits right after; A/B 1.403 vs 1.407
+  // ms/step at L=72k), NONTEMPORAL above it (write-allocate doubles the
+  // 534-MB logits write traffic at L=261k).  C2V_HF_NT=0/1 overrides.
+  static const int base_variant =
+      getenv("C2V_HF_VARIANT") ? atoi(getenv("C2V_HF_VARIANT")) : 0;
+  static const char* nt_env = getenv("C2V_HF_NT");
+  const bool nt = nt_env ? (nt_env[0] == '1') : (L > 98304);
+  const int variant = base_variant | (nt ? 0 : 32);
+#define HFCASE(nkt, st, ai)                                                  \
+  head_fwd_kernel<nkt, st, ai><<<grid, 512, 0, stream>>>(                    \
       (const bf16*)cv, (const bf16*)w, bias, (bf16*)out, pm, ps, B, L, EP,   \
       GYB, variant)
   if (EP == 128) {
-    if (stats) HFCASE(4, 1); else HFCASE(4, 0);
+    if (aimg) {
+      if (stats) HFCASE(4, 1, 1); else HFCASE(4, 0, 1);
+    } else {
+      if (stats) HFCASE(4, 1, 0); else HFCASE(4, 0, 0);
+    }
   } else {
-    if (stats) HFCASE(0, 1); else HFCASE(0, 0);
+    if (stats) HFCASE(0, 1, 0); else HFCASE(0, 0, 0);
   }
 #undef HFCASE
 }

@@ -36,6 +36,10 @@ FUSE_GATHER_COMBINER = os.environ.get("C2V_FUSE") == "1"
 # vocabs past C2V_HEAD_FWD_MAXL take the library path.
 _HEAD_FWD = os.environ.get("C2V_HEAD_FWD", "1") == "1"
 _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "98304"))
+# above MAXL: custom head forward consuming a swizzle_a W fragment image
+# (contiguous A reads; re-enables the fused stats epilogue at java-large
+# scale).  C2V_HF_AIMG=0 falls back to hipBLASLt + lsm_partial there.
+_HF_AIMG = os.environ.get("C2V_HF_AIMG", "1") == "1"
 # custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
 # combiner dgrad through dgrad2.hip (C2V_DGRAD2=0 -> rocBLAS)
@@ -666,6 +670,22 @@ def head_logits_with_stats(cv_bf16, w, bias):
             pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
             ps = torch.empty_like(pm)
             ext().head_fwd(cv_bf16, w, bias.float(), logits, pm, ps)
+            logits._c2v_lsm_partials = (pm, ps)
+        elif _HEAD_FWD and _HF_AIMG and EP == 128 and L % 8 == 0:
+            # large L: W pre-swizzled into the A-fragment image so the
+            # custom kernel's W reads are contiguous (strided per-lane
+            # row loads streamed W from HBM inefficiently at L=261k)
+            wimg = _scratch_bf16("hf_aimg",
+                                 ((L + 255) // 256 * 16, 4, 64, 8),
+                                 w.device)
+            ext().swizzle_a(w, wimg)
+            logits = torch.empty(B, L, dtype=torch.bfloat16,
+                                 device=cv_bf16.device)
+            gx = (L + 255) // 256
+            pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
+            ps = torch.empty_like(pm)
+            ext().head_fwd_img(cv_bf16, wimg, bias.float(), logits, pm, ps,
+                               L)
             logits._c2v_lsm_partials = (pm, ps)
         else:
             logits = torch.nn.functional.linear(

@@ -164,6 +164,14 @@ def main():
         bias_h = bias.to(torch.bfloat16)
         results["linear(hipblaslt)"] = timeit(
             lambda: torch.nn.functional.linear(cvb, wout, bias_h), iters=10)
+        wimg_a = torch.empty((L + 255) // 256 * 16, 4, 64, 8,
+                             dtype=torch.bfloat16, device=dev)
+        ext().swizzle_a(wout, wimg_a)
+        results["swizzle_a"] = timeit(lambda: ext().swizzle_a(wout, wimg_a),
+                                      iters=10)
+        results["head_fwd_img(stats)"] = timeit(
+            lambda: ext().head_fwd_img(cvb, wimg_a, bias, logits, pm, ps,
+                                       L), iters=10)
 
     if "head_bwd" in ops:
         from code2vec_amd.ops import ext
