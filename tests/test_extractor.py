@@ -141,3 +141,89 @@ def test_dataset_roundtrip(tmp_path):
     loss = model.loss(outp, y, torch.ones(opt.label_count))
     loss.backward()
     assert torch.isfinite(loss)
+
+
+TORTURE = """
+package com.example;
+
+import java.util.*;
+import java.util.function.*;
+
+@SuppressWarnings("unchecked")
+public class Torture<T extends Comparable<T>> {
+  private final Map<String, List<T>> cache = new HashMap<>();
+  static int[] GRID = new int[]{1, 2, 3};
+
+  public synchronized <R> List<R> transform(List<T> items,
+                                            Function<T, R> fn) throws Exception {
+    List<R> out = new ArrayList<>(items.size());
+    for (T item : items) {
+      out.add(fn.apply(item));
+    }
+    items.sort((a, b) -> a.compareTo(b));
+    out.removeIf(x -> x == null);
+    return out;
+  }
+
+  public int classify(int code, String label) {
+    switch (code) {
+      case 1:
+      case 2:
+        return code * 2;
+      case 3: {
+        int acc = 0;
+        do { acc += code--; } while (code > 0);
+        return acc;
+      }
+      default:
+        break;
+    }
+    outer:
+    for (int i = 0; i < GRID.length; i++) {
+      for (int j = i; j < 10; j += 2) {
+        if (GRID[i] > j) continue outer;
+        if (label != null && label.length() > (j & 3)) break outer;
+      }
+    }
+    try (AutoCloseable res = null) {
+      return label instanceof String s ? s.length() : -1;
+    } catch (Exception | Error e) {
+      throw new RuntimeException("fail: " + e, e);
+    } finally {
+      assert code >= 0 : "neg";
+    }
+  }
+
+  public double math(double x) {
+    long bits = 0x7fffL << 2;
+    float f = 1.5e-3f;
+    char c = '\\n';
+    String tb = "multi";
+    return (x + f) * (bits >>> 1) + (c == '\\n' ? 1.0 : 2.0) + tb.length()
+        + Math.sqrt((double) (int) x);
+  }
+}
+"""
+
+
+def test_extractor_torture_java():
+    """Generics, lambdas, method refs, labeled loops, switch fallthrough,
+    try-with-resources, multi-catch, instanceof patterns, text-ish
+    literals: all three methods must extract (not error out)."""
+    recs = by_name(X.extract_source(TORTURE))
+    for name in ("transform", "classify", "math"):
+        assert name in recs, sorted(recs)
+        r = recs[name]
+        assert "error" not in r, (name, r.get("error"))
+        assert len(r["contexts"]) > 0, name
+    t = recs["transform"]
+    al = dict(t["aliases"])
+    assert "items" in al and "fn" in al and "item" in al
+    # lambda params are vars too
+    assert "a" in al and "b" in al and "x" in al
+    c = recs["classify"]
+    cterms = {x for ctx in c["contexts"] for x in (ctx[0], ctx[2])}
+    assert "@int_literal" in cterms
+    m = recs["math"]
+    mterms = {x for ctx in m["contexts"] for x in (ctx[0], ctx[2])}
+    assert "@double_literal" in mterms and "@char_literal" in mterms

@@ -301,3 +301,25 @@ def test_find_hyperparams_with_stub_optuna(tmp_path, tiny_corpus, monkeypatch):
                                      "batch_size", "adam_lr", "weight_decay"}
     assert calls["reports"] >= 2   # per-epoch trial.report
     assert calls["pruned"] == 1    # TrialPruned raised and handled
+
+
+def test_floyd_metric_line_format(tmp_path, tiny_corpus, capsys):
+    """--env floyd prints the machine-readable metric lines via print()
+    (reference main.py:183-190 contract)."""
+    import main as main_mod
+
+    main_mod.main([
+        "--corpus_path", tiny_corpus["corpus_path"],
+        "--path_idx_path", tiny_corpus["path_idx_path"],
+        "--terminal_idx_path", tiny_corpus["terminal_idx_path"],
+        "--max_epoch", "1", "--no_cuda", "--env", "floyd",
+        "--encode_size", "16", "--terminal_embed_size", "8",
+        "--path_embed_size", "8", "--batch_size", "8",
+        "--model_path", str(tmp_path / "out"),
+        "--vectors_path", str(tmp_path / "out" / "code.vec"),
+    ])
+    out = capsys.readouterr().out
+    for name in ("train_loss", "test_loss", "accuracy", "precision",
+                 "recall", "f1", "best_f1"):
+        assert f'{{"metric": "{name}", "value": ' in out, name
+    assert "epoch 0" in out
