@@ -352,3 +352,69 @@ def test_owned_chunked_allreduce_world2(tmp_path):
         pr.join(timeout=240)
         assert pr.exitcode == 0
     assert open(result_file).read() == "ok"
+
+
+def _worker_tp_head(rank, world, port, result_file):
+    """Vocab-sharded head+loss (parallel/tp_head.py) == dense reference:
+    loss, dcv, and the reassembled dW/dbias shards must match."""
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    from code2vec_amd.ops import reference as R
+    from code2vec_amd.parallel.tp_head import (shard_rows,
+                                               vocab_parallel_head_loss)
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    torch.manual_seed(11)
+    B, E, L = 16, 32, 51  # odd L: uneven last shard
+    cv = torch.randn(B, E)
+    w = torch.randn(L, E) * 0.2
+    b = torch.randn(L) * 0.1
+    y = torch.randint(0, L, (B,))
+    wt = torch.rand(L) + 0.5
+
+    per = (L + world - 1) // world
+    cvh = cv.clone().requires_grad_(True)
+    wsh = shard_rows(w, rank, world).clone().requires_grad_(True)
+    bsh = shard_rows(b, rank, world).clone().requires_grad_(True)
+    loss = vocab_parallel_head_loss(cvh, wsh, bsh, y,
+                                    shard_rows(wt, rank, world),
+                                    rank * per, L)
+    loss.backward()
+
+    cvr = cv.clone().requires_grad_(True)
+    wr = w.clone().requires_grad_(True)
+    br = b.clone().requires_grad_(True)
+    ref = R.logsoftmax_nll(cvr @ wr.t() + br, y, wt)
+    ref.backward()
+
+    assert torch.allclose(loss, ref, atol=1e-5), (float(loss), float(ref))
+    assert torch.allclose(cvh.grad, cvr.grad, atol=1e-5)
+    assert torch.allclose(wsh.grad, shard_rows(wr.grad, rank, world),
+                          atol=1e-5)
+    assert torch.allclose(bsh.grad, shard_rows(br.grad, rank, world),
+                          atol=1e-5)
+    if rank == 0:
+        with open(result_file, "w") as f:
+            f.write("ok")
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("world", [2, 3])
+def test_vocab_parallel_head_matches_dense(tmp_path, world):
+    ctx = mp.get_context("spawn")
+    port = _free_port()
+    result_file = str(tmp_path / "tp.txt")
+    procs = [ctx.Process(target=_worker_tp_head,
+                         args=(r, world, port, result_file))
+             for r in range(world)]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=240)
+        assert pr.exitcode == 0
+    assert open(result_file).read() == "ok"
