@@ -176,6 +176,58 @@ def main() -> None:
             ddp.finish()
             optim.step()
 
+    # hipGraph-captured whole training step (N=1 pool mode) — opt-in
+    # EXPERIMENT (C2V_GRAPH_STEP=1): capture works and replays bitwise-
+    # correctly (the dropout RNG offset and Adam beta^t live in device
+    # memory precisely so replays advance them), but MEASURED SLIGHTLY
+    # SLOWER than eager (1.441 vs 1.429 ms/step at top11): the eager
+    # loop already pipelines enqueue across steps, so graph launch saves
+    # nothing here and the per-batch copy into static buffers adds a
+    # little.  Kept as validated infrastructure (test
+    # test_graph_captured_step_matches_eager).
+    graph_note = ""
+    if (on_gpu and world == 1 and args.mode == "train"
+            and not args.real_pipeline
+            and os.environ.get("C2V_GRAPH_STEP", "0") == "1"):
+        try:
+            static = tuple(t.clone() for t in pool[0])
+
+            def graph_body():
+                s, p, e, y = static
+                ddp.zero_grad()
+                outputs, _, _ = model(s, p, e, y)
+                loss = model.loss(outputs, y, class_weight)
+                loss.backward()
+                ddp.finish_and_step(optim)
+
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):  # allocator/TunableOp warmup
+                    graph_body()
+            torch.cuda.current_stream().wait_stream(side)
+            optim.bc_pow.fill_(1.0)  # capture records, does not execute:
+            Fn = None                # continue beta^t from the real count
+            from code2vec_amd.ops import functional as Fn  # noqa: F811
+            for _ in range(optim.step_count):
+                optim.bc_pow[0] *= optim.beta1
+                optim.bc_pow[1] *= optim.beta2
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                graph_body()
+
+            def graphed_step(i: int) -> None:
+                for dst, src in zip(static, pool[i % len(pool)]):
+                    dst.copy_(src, non_blocking=True)
+                graph.replay()
+
+            step = graphed_step
+            graph_note = "; step captured in a hipGraph (replayed per batch)"
+        except Exception as exc:  # noqa: BLE001 - eager fallback
+            if ctx.is_rank0:
+                print(f"# hipGraph capture unavailable, eager step: {exc}",
+                      file=sys.stderr)
+
     for i in range(args.warmup):
         step(i)
 
@@ -234,7 +286,8 @@ def main() -> None:
                 "note": ("step = fwd + fused log-softmax/NLL loss + bwd + bucketed RCCL all-reduce + fused Adam"
                         + ("; top11 label vocab 72,416 derived via Heaps-law fit "
                            "to the reference's real methods.txt name curve "
-                           "(tools/derive_label_vocab.py)" if args.config == "top11" else "")),
+                           "(tools/derive_label_vocab.py)" if args.config == "top11" else "")
+                        + graph_note),
             },
         }))
 

@@ -16,6 +16,7 @@ from typing import Iterable, Tuple
 import torch
 
 from ..ops import functional as Fn
+from ..ops import ext as _ext  # noqa: F401  (adam_tick access via Fn.ext)
 
 
 class FusedAdam:
@@ -33,6 +34,10 @@ class FusedAdam:
         self.eps = eps
         self.weight_decay = weight_decay
         self.step_count = 0
+        # device-resident (beta1^t, beta2^t): hipGraph-capturable bias
+        # correction (advanced by one adam_tick kernel per step)
+        dev = self.params[0].device if self.params else torch.device("cpu")
+        self.bc_pow = torch.ones(2, dtype=torch.float64, device=dev)
         self.state = {}
         for p in self.params:
             st = {
@@ -52,13 +57,14 @@ class FusedAdam:
         ``step_rows`` in the same logical step (the DDP-overlapped table
         path) — the step counter is bumped here, once."""
         self.step_count += 1
+        Fn.ext().adam_tick(self.bc_pow, self.beta1, self.beta2)
         for p in self.params:
             if p.grad is None or (exclude_ids and id(p) in exclude_ids):
                 continue
             st = self.state[p]
             Fn.adam_step(
                 p.data, p.grad, st["master"], st["m"], st["v"],
-                self.step_count, self.lr, self.beta1, self.beta2,
+                self.bc_pow, self.lr, self.beta1, self.beta2,
                 self.eps, self.weight_decay,
             )
 
@@ -75,7 +81,7 @@ class FusedAdam:
             p.data.view(-1)[lo:hi], grad2d.view(-1)[lo:hi],
             None if st["master"] is None else st["master"][lo:hi],
             st["m"][lo:hi], st["v"][lo:hi],
-            self.step_count, self.lr, self.beta1, self.beta2,
+            self.bc_pow, self.lr, self.beta1, self.beta2,
             self.eps, self.weight_decay,
         )
 

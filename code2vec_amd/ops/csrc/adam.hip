@@ -8,6 +8,18 @@
 
 #include "common.h"
 
+// bc_pow: DEVICE double[2] = (beta1^t, beta2^t), advanced once per
+// optimizer step by adam_tick_kernel — device-resident so a hipGraph-
+// captured training step keeps the bias correction advancing per replay
+// (a host-computed inv_bc argument would freeze t at capture time).
+__global__ void adam_tick_kernel(double* __restrict__ bc_pow, float b1,
+                                 float b2) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    bc_pow[0] *= (double)b1;
+    bc_pow[1] *= (double)b2;
+  }
+}
+
 template <int NT, int U>  // NT: nontemporal; U: independent 4-elem chunks
                           // per loop iteration (ILP / outstanding loads)
 __global__ void adam_bf16_kernel(bf16* __restrict__ p,
@@ -15,8 +27,10 @@ __global__ void adam_bf16_kernel(bf16* __restrict__ p,
                                  float* __restrict__ master,
                                  float* __restrict__ m, float* __restrict__ v,
                                  long n, float lr, float b1, float b2,
-                                 float eps, float wd, float inv_bc1,
-                                 float inv_bc2) {
+                                 float eps, float wd,
+                                 const double* __restrict__ bc_pow) {
+  const float inv_bc1 = (float)(1.0 / (1.0 - bc_pow[0]));
+  const float inv_bc2 = (float)(1.0 / (1.0 - bc_pow[1]));
   const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   const long cstride = (long)gridDim.x * blockDim.x * 4;  // chunk stride
   const long stride = cstride * U;
@@ -87,8 +101,10 @@ __global__ void adam_f32_kernel(float* __restrict__ p,
                                 const float* __restrict__ g,
                                 float* __restrict__ m, float* __restrict__ v,
                                 long n, float lr, float b1, float b2,
-                                float eps, float wd, float inv_bc1,
-                                float inv_bc2) {
+                                float eps, float wd,
+                                const double* __restrict__ bc_pow) {
+  const float inv_bc1 = (float)(1.0 / (1.0 - bc_pow[0]));
+  const float inv_bc2 = (float)(1.0 / (1.0 - bc_pow[1]));
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long j = i; j < n; j += stride) {
@@ -101,11 +117,15 @@ __global__ void adam_f32_kernel(float* __restrict__ p,
 
 extern "C" {
 
+void launch_adam_tick(double* bc_pow, float b1, float b2,
+                      hipStream_t stream) {
+  adam_tick_kernel<<<1, 1, 0, stream>>>(bc_pow, b1, b2);
+}
+
 void launch_adam_bf16(void* p, const void* g, float* master, float* m,
-                      float* v, long n, int step, float lr, float b1,
-                      float b2, float eps, float wd, hipStream_t stream) {
-  const float inv_bc1 = 1.0f / (1.0f - powf(b1, (float)step));
-  const float inv_bc2 = 1.0f / (1.0f - powf(b2, (float)step));
+                      float* v, long n, const double* bc_pow, float lr,
+                      float b1, float b2, float eps, float wd,
+                      hipStream_t stream) {
   const int block = 256;
   // defaults swept on hardware against the same-box D2D copy ceiling
   // (5.5 TB/s): ILP 2 / grid 64k runs the 3.2-GB table update at 5.2 TB/s
@@ -122,26 +142,24 @@ void launch_adam_bf16(void* p, const void* g, float* master, float* m,
   if (nt == NTV && U == UV) {                                               \
     adam_bf16_kernel<NTV, UV><<<grid, block, 0, stream>>>(                  \
         (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,     \
-        inv_bc1, inv_bc2);                                                  \
+        bc_pow);                                                            \
     return;                                                                 \
   }
   ADAM_CASE(1, 1) ADAM_CASE(1, 2) ADAM_CASE(1, 4)
   ADAM_CASE(0, 1) ADAM_CASE(0, 2) ADAM_CASE(0, 4)
   adam_bf16_kernel<1, 2><<<grid, block, 0, stream>>>(
       (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,
-      inv_bc1, inv_bc2);
+      bc_pow);
 }
 
 void launch_adam_f32(float* p, const float* g, float* m, float* v, long n,
-                     int step, float lr, float b1, float b2, float eps,
-                     float wd, hipStream_t stream) {
-  const float inv_bc1 = 1.0f / (1.0f - powf(b1, (float)step));
-  const float inv_bc2 = 1.0f / (1.0f - powf(b2, (float)step));
+                     const double* bc_pow, float lr, float b1, float b2,
+                     float eps, float wd, hipStream_t stream) {
   const int block = 256;
   const long want = (n + block - 1) / block;
   const int grid = (int)min(want > 0 ? want : 1, (long)4096);
   adam_f32_kernel<<<grid, block, 0, stream>>>(p, g, m, v, n, lr, b1, b2, eps,
-                                              wd, inv_bc1, inv_bc2);
+                                              wd, bc_pow);
 }
 
 }  // extern "C"

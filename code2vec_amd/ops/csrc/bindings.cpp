@@ -29,14 +29,15 @@ void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_exclusive_scan(const int*, int*, int*, long, hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
-                         float, unsigned long long, unsigned long long, int,
-                         hipStream_t);
+                         float, unsigned long long,
+                         const unsigned long long*, int, hipStream_t);
+void launch_bump_u64(void*, unsigned long long, hipStream_t);
 void launch_gather_combiner_fwd(const int*, const int*, const int*,
                                 const void*, const void*, int, int,
                                 const void*, const float*, const float*,
                                 void*, void*, float*, float*, long, int, int,
                                 int, float, unsigned long long,
-                                unsigned long long, hipStream_t);
+                                const unsigned long long*, hipStream_t);
 void launch_wgrad_gather(const int*, const int*, const int*, const void*,
                          const void*, int, int, const void*, float*, long,
                          int, int, int, hipStream_t);
@@ -91,10 +92,13 @@ void launch_lsm_finalize(const void*, const float*, const float*, const long*,
                          const float*, float*, float*, int, long, int,
                          hipStream_t);
 void launch_lsm_partial(const void*, float*, float*, int, long, hipStream_t);
-void launch_adam_bf16(void*, const void*, float*, float*, float*, long, int,
-                      float, float, float, float, float, hipStream_t);
-void launch_adam_f32(float*, const float*, float*, float*, long, int, float,
-                     float, float, float, float, hipStream_t);
+void launch_adam_tick(double*, float, float, hipStream_t);
+void launch_adam_bf16(void*, const void*, float*, float*, float*, long,
+                      const double*, float, float, float, float, float,
+                      hipStream_t);
+void launch_adam_f32(float*, const float*, float*, float*, long,
+                     const double*, float, float, float, float, float,
+                     hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -167,10 +171,13 @@ void cast_clear_rows(torch::Tensor dtable, torch::Tensor counts,
                          S, cur_stream());
 }
 
+// rng_off: int64 [1] DEVICE scalar holding the dropout counter offset;
+// consumed by the kernel and advanced by M*EP on the same stream (device-
+// side state so hipGraph replays keep drawing fresh masks).
 void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
                   torch::Tensor beta, torch::Tensor out, torch::Tensor z,
                   torch::Tensor mean, torch::Tensor rstd, int64_t E,
-                  double p, int64_t seed, int64_t offset,
+                  double p, int64_t seed, torch::Tensor rng_off,
                   int64_t epilogue_mode) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
   CHK_CONTIG(w); CHK_DT(w, torch::kBFloat16);
@@ -180,12 +187,16 @@ void combiner_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor gamma,
   const long M = x.size(0);
   const int KP = x.size(1), EP = w.size(0);  // w is TRANSPOSED: [EP, KP]
   TORCH_CHECK(w.size(1) == KP && KP % 32 == 0 && EP % 32 == 0, "w shape");
+  CHK_DT(rng_off, torch::kInt64);
   launch_combiner_fwd(x.data_ptr(), w.data_ptr(), gamma.data_ptr<float>(),
                       beta.data_ptr<float>(), out.data_ptr(), z.data_ptr(),
                       mean.data_ptr<float>(), rstd.data_ptr<float>(), M, KP,
                       EP, (int)E, (float)p, (unsigned long long)seed,
-                      (unsigned long long)offset, (int)epilogue_mode,
-                      cur_stream());
+                      (const unsigned long long*)rng_off.data_ptr(),
+                      (int)epilogue_mode, cur_stream());
+  if (p > 0.0)
+    launch_bump_u64(rng_off.data_ptr(), (unsigned long long)(M * EP),
+                    cur_stream());
 }
 
 void gather_combiner_fwd(torch::Tensor starts, torch::Tensor paths,
@@ -194,7 +205,8 @@ void gather_combiner_fwd(torch::Tensor starts, torch::Tensor paths,
                          torch::Tensor gamma, torch::Tensor beta,
                          torch::Tensor out, torch::Tensor z,
                          torch::Tensor mean, torch::Tensor rstd, int64_t KP,
-                         int64_t E, double p, int64_t seed, int64_t offset) {
+                         int64_t E, double p, int64_t seed,
+                         torch::Tensor rng_off) {
   CHK_CUDA(starts); CHK_CONTIG(starts); CHK_DT(starts, torch::kInt32);
   CHK_DT(term, torch::kBFloat16); CHK_CONTIG(term);
   CHK_DT(path, torch::kBFloat16); CHK_CONTIG(path);
@@ -204,13 +216,18 @@ void gather_combiner_fwd(torch::Tensor starts, torch::Tensor paths,
   const int EP = w.size(0);
   const int TS = term.size(1), PS = path.size(1);
   TORCH_CHECK(w.size(1) == KP && KP >= 2 * TS + PS, "shapes");
+  CHK_DT(rng_off, torch::kInt64);
   launch_gather_combiner_fwd(
       starts.data_ptr<int>(), paths.data_ptr<int>(), ends.data_ptr<int>(),
       term.data_ptr(), path.data_ptr(), TS, PS, w.data_ptr(),
       gamma.data_ptr<float>(), beta.data_ptr<float>(), out.data_ptr(),
       z.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), M,
       (int)KP, EP, (int)E, (float)p, (unsigned long long)seed,
-      (unsigned long long)offset, cur_stream());
+      (const unsigned long long*)rng_off.data_ptr(), cur_stream());
+  if (p > 0.0)
+    launch_bump_u64(rng_off.data_ptr(),
+                    (unsigned long long)(M * (long)w.size(0)),
+                    cur_stream());
 }
 
 void wgrad_gather(torch::Tensor starts, torch::Tensor paths,
@@ -686,26 +703,37 @@ void wgrad(torch::Tensor x, torch::Tensor dz, torch::Tensor partials) {
                KP, EP, partials.size(0), cur_stream());
 }
 
+// bc_pow: float64 [2] DEVICE tensor = (beta1^t, beta2^t); advance it with
+// adam_tick once per optimizer step (device-resident so hipGraph replays
+// keep the bias correction advancing).
+void adam_tick(torch::Tensor bc_pow, double b1, double b2) {
+  CHK_CUDA(bc_pow); CHK_DT(bc_pow, torch::kFloat64);
+  TORCH_CHECK(bc_pow.numel() == 2, "bc_pow must be [2]");
+  launch_adam_tick((double*)bc_pow.data_ptr(), (float)b1, (float)b2,
+                   cur_stream());
+}
+
 void adam_step_bf16(torch::Tensor p, torch::Tensor g, torch::Tensor master,
-                    torch::Tensor m, torch::Tensor v, int64_t step, double lr,
-                    double b1, double b2, double eps, double wd) {
+                    torch::Tensor m, torch::Tensor v, torch::Tensor bc_pow,
+                    double lr, double b1, double b2, double eps, double wd) {
   CHK_CUDA(p); CHK_DT(p, torch::kBFloat16); CHK_DT(g, torch::kBFloat16);
-  CHK_DT(master, torch::kFloat32);
+  CHK_DT(master, torch::kFloat32); CHK_DT(bc_pow, torch::kFloat64);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == master.numel(), "sizes");
   launch_adam_bf16(p.data_ptr(), g.data_ptr(), master.data_ptr<float>(),
                    m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
-                   (int)step, (float)lr, (float)b1, (float)b2, (float)eps,
-                   (float)wd, cur_stream());
+                   (const double*)bc_pow.data_ptr(), (float)lr, (float)b1,
+                   (float)b2, (float)eps, (float)wd, cur_stream());
 }
 
 void adam_step_f32(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                   torch::Tensor v, int64_t step, double lr, double b1,
-                   double b2, double eps, double wd) {
+                   torch::Tensor v, torch::Tensor bc_pow, double lr,
+                   double b1, double b2, double eps, double wd) {
   CHK_CUDA(p); CHK_DT(p, torch::kFloat32); CHK_DT(g, torch::kFloat32);
+  CHK_DT(bc_pow, torch::kFloat64);
   launch_adam_f32(p.data_ptr<float>(), g.data_ptr<float>(),
                   m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
-                  (int)step, (float)lr, (float)b1, (float)b2, (float)eps,
-                  (float)wd, cur_stream());
+                  (const double*)bc_pow.data_ptr(), (float)lr, (float)b1,
+                  (float)b2, (float)eps, (float)wd, cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -747,6 +775,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("slab_sum_bf16", &slab_sum_bf16);
   m.def("slab_sum_f32", &slab_sum_f32);
   m.def("dgrad", &dgrad);
+  m.def("adam_tick", &adam_tick);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);
 }

@@ -46,7 +46,7 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
     bf16* __restrict__ out, bf16* __restrict__ z_save,
     float* __restrict__ mean_save, float* __restrict__ rstd_save,
     long M, int KP, int E, float p, float inv1mp,
-    unsigned long long seed, unsigned long long offset,
+    unsigned long long seed, const unsigned long long* __restrict__ rng_off,
     const int* __restrict__ g_starts, const int* __restrict__ g_paths,
     const int* __restrict__ g_ends, const bf16* __restrict__ g_term,
     const bf16* __restrict__ g_path, int TS, int PS) {
@@ -54,6 +54,10 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const long row0 = (long)blockIdx.x * (MT * 64);
+  // RNG offset lives in DEVICE memory so hipGraph replays of a captured
+  // training step advance the dropout stream (a by-value argument would
+  // freeze the mask at capture time); bumped by bump_u64_kernel per call
+  const unsigned long long offset = p > 0.0f ? rng_off[0] : 0ull;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* lds_gamma = (float*)smem;
@@ -375,13 +379,24 @@ __global__ __launch_bounds__(256) void combiner_bwd_kernel(
   }
 }
 
+__global__ void bump_u64_kernel(unsigned long long* p,
+                                unsigned long long delta) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) p[0] += delta;
+}
+
 extern "C" {
+
+void launch_bump_u64(void* p, unsigned long long delta,
+                     hipStream_t stream) {
+  bump_u64_kernel<<<1, 1, 0, stream>>>((unsigned long long*)p, delta);
+}
 
 void launch_combiner_fwd_impl(const void* X, const void* W,
                          const float* gamma,
                          const float* beta, void* out, void* z, float* mean,
                          float* rstd, long M, int KP, int EP, int E, float p,
-                         unsigned long long seed, unsigned long long offset,
+                         unsigned long long seed,
+                         const unsigned long long* rng_off,
                          int epilogue_mode, int gather, const int* starts,
                          const int* paths, const int* ends, const void* term,
                          const void* path, int TS, int PS,
@@ -404,22 +419,22 @@ void launch_combiner_fwd_impl(const void* X, const void* W,
     if (gather && mt == 1)                                                    \
       combiner_fwd_kernel<nt, 1, 1, 1><<<grid, 256, smem, stream>>>(          \
           nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset, starts, paths,       \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, starts, paths,      \
           ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
     else if (gather)                                                          \
       combiner_fwd_kernel<nt, 1, 1, 2><<<grid, 256, smem, stream>>>(          \
           nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset, starts, paths,       \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, starts, paths,      \
           ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
     else if (epi == 1)                                                        \
       combiner_fwd_kernel<nt, 1, 0><<<grid, 256, smem, stream>>>(             \
           (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset, nullptr, nullptr,    \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, nullptr, nullptr,   \
           nullptr, nullptr, nullptr, 0, 0);                                   \
     else                                                                      \
       combiner_fwd_kernel<nt, 0, 0><<<grid, 256, smem, stream>>>(             \
           (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
-          mean, rstd, M, KP, E, p, inv1mp, seed, offset, nullptr, nullptr,    \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, nullptr, nullptr,   \
           nullptr, nullptr, nullptr, 0, 0);                                   \
     break;
   switch (NT) {
@@ -434,10 +449,11 @@ void launch_combiner_fwd_impl(const void* X, const void* W,
 void launch_combiner_fwd(const void* X, const void* W, const float* gamma,
                          const float* beta, void* out, void* z, float* mean,
                          float* rstd, long M, int KP, int EP, int E, float p,
-                         unsigned long long seed, unsigned long long offset,
+                         unsigned long long seed,
+                         const unsigned long long* rng_off,
                          int epilogue_mode, hipStream_t stream) {
   launch_combiner_fwd_impl(X, W, gamma, beta, out, z, mean, rstd, M, KP, EP,
-                           E, p, seed, offset, epilogue_mode, 0, nullptr,
+                           E, p, seed, rng_off, epilogue_mode, 0, nullptr,
                            nullptr, nullptr, nullptr, nullptr, 0, 0, stream);
 }
 
@@ -449,10 +465,10 @@ void launch_gather_combiner_fwd(const int* starts, const int* paths,
                                 float* mean, float* rstd, long M, int KP,
                                 int EP, int E, float p,
                                 unsigned long long seed,
-                                unsigned long long offset,
+                                const unsigned long long* rng_off,
                                 hipStream_t stream) {
   launch_combiner_fwd_impl(nullptr, W, gamma, beta, out, z, mean, rstd, M,
-                           KP, EP, E, p, seed, offset, 1, 1, starts, paths,
+                           KP, EP, E, p, seed, rng_off, 1, 1, starts, paths,
                            ends, term, path, TS, PS, stream);
 }
 

@@ -63,22 +63,28 @@ OWNED_GRAD_KEYS = set()
 
 from . import ext, round_up
 
-_rng_state = {"seed": None, "offset": 0}
+_rng_state = {"seed": None, "dev_off": {}}
 
 
-def _next_philox(n: int):
-    """Per-call (seed, offset) for the counter-based dropout RNG."""
+def _philox_state(device):
+    """(seed, device offset scalar) for the counter-based dropout RNG.
+    The offset lives in DEVICE memory and is advanced by the consuming
+    kernel launch chain, so a hipGraph-captured training step keeps
+    drawing fresh masks on every replay."""
     if _rng_state["seed"] is None:
         _rng_state["seed"] = torch.initial_seed() & 0x7FFFFFFFFFFFFFFF
-    seed = _rng_state["seed"]
-    offset = _rng_state["offset"]
-    _rng_state["offset"] += n
-    return seed, offset
+    key = str(device)
+    off = _rng_state["dev_off"].get(key)
+    if off is None:
+        off = torch.zeros(1, dtype=torch.int64, device=device)
+        _rng_state["dev_off"][key] = off
+    return _rng_state["seed"], off
 
 
 def reseed_dropout_rng(seed: int) -> None:
     _rng_state["seed"] = seed & 0x7FFFFFFFFFFFFFFF
-    _rng_state["offset"] = 0
+    for off in _rng_state["dev_off"].values():
+        off.zero_()
 
 
 class GatherConcat(torch.autograd.Function):
@@ -290,8 +296,10 @@ class CombinerLNTanh(torch.autograd.Function):
         mean = torch.empty(M, dtype=torch.float32, device=x.device)
         rstd = torch.empty(M, dtype=torch.float32, device=x.device)
         p_eff = float(p) if training else 0.0
-        seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
-        ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff, seed, offset, _FWD_EPI)
+        seed, off_t = (_philox_state(x.device) if p_eff > 0.0
+                       else (0, _philox_state(x.device)[1]))
+        ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, p_eff,
+                           seed, off_t, _FWD_EPI)
         ctx.save_for_backward(x, w, gamma, beta, z, mean, rstd, out)
         ctx.meta = (E, p_eff)
         return out
@@ -358,10 +366,11 @@ class FusedGatherCombiner(torch.autograd.Function):
         mean = torch.empty(M, dtype=torch.float32, device=dev)
         rstd = torch.empty(M, dtype=torch.float32, device=dev)
         p_eff = float(p) if training else 0.0
-        seed, offset = _next_philox(M * EP) if p_eff > 0.0 else (0, 0)
+        seed, off_t = (_philox_state(dev) if p_eff > 0.0
+                       else (0, _philox_state(dev)[1]))
         ext().gather_combiner_fwd(starts, paths, ends, term_w, path_w, w,
                                   gamma, beta, out, z, mean, rstd, KP, E,
-                                  p_eff, seed, offset)
+                                  p_eff, seed, off_t)
         ctx.save_for_backward(starts, paths, ends, term_w, path_w, w, gamma,
                               beta, z, mean, rstd, out)
         ctx.meta = (E, p_eff)
@@ -791,7 +800,7 @@ def adam_step(
     master: Optional[torch.Tensor],
     m: torch.Tensor,
     v: torch.Tensor,
-    step: int,
+    bc_pow: torch.Tensor,
     lr: float,
     beta1: float,
     beta2: float,
@@ -799,15 +808,17 @@ def adam_step(
     weight_decay: float,
 ) -> None:
     """K16: fused Adam.  bf16 params carry an f32 master (updated in f32,
-    rounded to bf16 param); f32 params update in place (master is None)."""
+    rounded to bf16 param); f32 params update in place (master is None).
+    ``bc_pow`` is the optimizer's device-resident (beta1^t, beta2^t)
+    float64 pair (advanced by adam_tick once per step)."""
     if param.dtype == torch.bfloat16:
         assert master is not None
         ext().adam_step_bf16(
             param.view(-1), grad.view(-1), master, m, v,
-            step, lr, beta1, beta2, eps, weight_decay,
+            bc_pow, lr, beta1, beta2, eps, weight_decay,
         )
     else:
         ext().adam_step_f32(
             param.view(-1), grad.view(-1).float(), m, v,
-            step, lr, beta1, beta2, eps, weight_decay,
+            bc_pow, lr, beta1, beta2, eps, weight_decay,
         )

@@ -504,3 +504,91 @@ def test_trainer_eval_methods_gpu(dev, tmp_path, method):
     trainer = Trainer(TrainerConfig(A), opt, reader, builder, model, ctx)
     obj = trainer.train()
     assert 0.0 <= obj <= 1.0
+
+
+def test_graph_captured_step_matches_eager(dev):
+    """A hipGraph-captured training step must produce bitwise-identical
+    parameters to the eager step over several replays (device-resident
+    dropout offset advances per replay)."""
+    import numpy as np
+
+    from code2vec_amd.data.synthetic import synthetic_batch
+    from code2vec_amd.engine.optim import FusedAdam
+    from code2vec_amd.models.code2vec import Code2VecHIP, init_logical_params
+    from code2vec_amd.ops.functional import reseed_dropout_rng
+    from code2vec_amd.parallel.ddp import BucketedAllReduce
+    from code2vec_amd.utils.options import Option
+
+    opt = Option(terminal_count=600, path_count=500, label_count=1024,
+                 max_path_length=16, terminal_embed_size=100,
+                 path_embed_size=100, encode_size=100, dropout_prob=0.25)
+    g = torch.Generator().manual_seed(41)
+    logical = init_logical_params(opt, g)
+    rng = np.random.default_rng(77)
+    batches = []
+    for _ in range(3):
+        s, p, e, y = synthetic_batch(rng, 16, opt.max_path_length,
+                                     opt.terminal_count, opt.path_count,
+                                     opt.label_count)
+        batches.append(tuple(torch.from_numpy(a).to(dev)
+                             for a in (s, p, e, y)))
+    w = torch.ones(opt.label_count, device=dev)
+
+    def run(graphed):
+        reseed_dropout_rng(123)
+        m = Code2VecHIP(opt, logical, device=dev).train()
+        owned = [m.terminal_embedding, m.path_embedding]
+        ddp = BucketedAllReduce(list(m.parameters()), 1, owned_params=owned)
+        optim = FusedAdam(m.parameters(), lr=0.01)
+
+        static = tuple(t.clone() for t in batches[0])
+
+        def body():
+            s, p, e, y = static
+            ddp.zero_grad()
+            out, _, _ = m(s, p, e, y)
+            m.loss(out, y, w).backward()
+            ddp.finish_and_step(optim)
+
+        if graphed:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    body()
+            torch.cuda.current_stream().wait_stream(side)
+            # fresh model state for the measured run (warmup mutated it)
+            reseed_dropout_rng(123)
+            m2 = Code2VecHIP(opt, logical, device=dev).train()
+            with torch.no_grad():
+                for a, b in zip(m.parameters(), m2.parameters()):
+                    a.copy_(b)
+            for p_, st in optim.state.items():
+                st["m"].zero_()
+                st["v"].zero_()
+                if st["master"] is not None:
+                    st["master"].copy_(p_.detach().float().view(-1))
+            optim.step_count = 0
+            optim.bc_pow.fill_(1.0)  # warmup advanced the device beta^t
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                body()
+            for b_ in batches:
+                for dst, src in zip(static, b_):
+                    dst.copy_(src, non_blocking=True)
+                graph.replay()
+        else:
+            for b_ in batches:
+                for dst, src in zip(static, b_):
+                    dst.copy_(src, non_blocking=True)
+                body()
+        torch.cuda.synchronize()
+        out = {n: q.detach().float().cpu().clone()
+               for n, q in m.named_parameters()}
+        ddp.close()
+        return out
+
+    eager = run(False)
+    graphed = run(True)
+    for name in eager:
+        assert torch.equal(eager[name], graphed[name]), name

@@ -117,8 +117,9 @@ def main():
         m = torch.zeros_like(master)
         v = torch.zeros_like(master)
         from code2vec_amd.ops import ext
+        bc = torch.full((2,), 0.5, dtype=torch.float64, device=dev)
         results["adam(term)"] = timeit(
-            lambda: ext().adam_step_bf16(p1, g1, master, m, v, 5, 0.01,
+            lambda: ext().adam_step_bf16(p1, g1, master, m, v, bc, 0.01,
                                          0.9, 0.999, 1e-8, 0.0))
 
     if "wgrad" in ops:

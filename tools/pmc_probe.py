@@ -49,7 +49,8 @@ partials = torch.empty(256, KP, EP, dtype=torch.float32, device=dev)
 for _ in range(REPS):
     ext().gather_concat_fwd(starts, pth, ends, term, path, x)
 for _ in range(REPS):
-    ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, 0.0, 0, 0, 1)
+    ext().combiner_fwd(x, w, gamma, beta, out, z, mean, rstd, E, 0.0, 0,
+                       torch.zeros(1, dtype=torch.int64, device=dev), 1)
 ccv = out.view(B, C, EP)
 for _ in range(REPS):
     ext().attention_fwd(ccv, a, starts, cv, attn, E)
@@ -111,7 +112,10 @@ master = p1.float()
 m_ = torch.zeros_like(master)
 v_ = torch.zeros_like(master)
 for _ in range(REPS):
-    ext().adam_step_bf16(p1, g1, master, m_, v_, 5, 0.01, 0.9, 0.999, 1e-8, 0.0)
+    ext().adam_step_bf16(p1, g1, master, m_, v_,
+                         torch.full((2,), 0.5, dtype=torch.float64,
+                                    device=dev), 0.01, 0.9, 0.999, 1e-8,
+                         0.0)
 
 torch.cuda.synchronize()
 print("pmc probe done")
