@@ -39,8 +39,12 @@
 // tables directly (segment-resolved), never materializing the concat
 // tensor.  Segment boundaries are 8-element multiples, so a 16-B fragment
 // piece never straddles one.
-template <int NT, int EPI, int GATHER, int MT = 2>  // MT: 16-row M tiles per wave
-__global__ __launch_bounds__(256) void combiner_fwd_kernel(
+// MT: 16-row M tiles per wave.  MINW: min waves/SIMD the register allocator
+// must honor (occupancy experiment C2V_COMBINER_OCC — the default NT=8/MT=2
+// build lands at 230 VGPR+AGPR = 2 waves/SIMD; forcing 3 trades ILP/spill
+// for latency hiding on the TA-issue-bound A-fragment loads).
+template <int NT, int EPI, int GATHER, int MT = 2, int MINW = 1>
+__global__ __launch_bounds__(256, MINW) void combiner_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ Wt,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     bf16* __restrict__ out, bf16* __restrict__ z_save,
@@ -61,7 +65,7 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* lds_gamma = (float*)smem;
-  float* lds_beta = lds_gamma + EP;
+  float* lds_beta = lds_gamma + EP;  // gamma/beta live for the whole kernel
   for (int c = threadIdx.x; c < EP; c += blockDim.x) {
     lds_gamma[c] = gamma[c];
     lds_beta[c] = beta[c];
@@ -165,9 +169,13 @@ __global__ __launch_bounds__(256) void combiner_fwd_kernel(
   // as contiguous 16-B chunks — the wave's 32 output rows are contiguous
   // in memory, so the flush is one coalesced 8-KB span per tensor.
   const float invE = 1.0f / (float)E;
+  // The bounce tile ALIASES the B staging buffer: the K-loop's final
+  // __syncthreads() guarantees every wave is done reading lds_b, so the
+  // epilogue reuses its space — block LDS drops from 56.3 KB to 35.8 KB
+  // (EP=128) and LDS stops capping occupancy at 2 blocks/CU.
   bf16* lds_t = nullptr;
   if (EPI == 1) {
-    lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) + 2 * (size_t)EP * 40 +
+    lds_t = (bf16*)(smem + 2 * EP * sizeof(float)) +
             (size_t)wave * (MT * 16) * (EP + 8);
   }
   float mean_r[MT][4], rstd_r[MT][4];
@@ -404,14 +412,26 @@ void launch_combiner_fwd_impl(const void* X, const void* W,
   const int NT = EP / 16;
   const char* mt_env = getenv("C2V_COMBINER_MT");
   const int mt = (mt_env && mt_env[0] == '1') ? 1 : 2;
+  // Occupancy-forced register allocation (C2V_COMBINER_OCC=0/3/4): the
+  // unconstrained build burns 230 VGPR+AGPR -> 2 waves/SIMD; with the
+  // epilogue bounce aliased over the dead B buffer, LDS fits 4 blocks/CU
+  // and __launch_bounds__(256,4) repacks to 128 VGPR (6 spills) ->
+  // measured 116.7 / 97.1 / 92.3 us (occ 2/3/4) at top11 — default 4.
+  // (guarded to NT<=10 in the dispatch: larger encode widths spill >1 KB
+  // scratch/lane under the cap and must keep the unconstrained build)
+  const char* occ_env = getenv("C2V_COMBINER_OCC");
+  const int occ = occ_env ? atoi(occ_env) : 4;
   const long grid = (M + mt * 64 - 1) / (mt * 64);
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
   int epi = epilogue_mode;
   (void)gather;
-  int smem = 2 * EP * sizeof(float) + 2 * EP * 40 * (int)sizeof(bf16);
+  const int bbuf = 2 * EP * 40 * (int)sizeof(bf16);
+  int smem = 2 * EP * sizeof(float) + bbuf;
   if (epi == 1) {
+    // bounce aliases the dead B buffer; allocate the max of the two
     const int bounce = 4 * mt * 16 * (EP + 8) * (int)sizeof(bf16);
-    if (smem + bounce <= 160 * 1024 - 2048) smem += bounce;
+    const int need = 2 * EP * (int)sizeof(float) + (bounce > bbuf ? bounce : bbuf);
+    if (need <= 160 * 1024 - 2048) smem = need;
     else epi = 0;
   }
 #define CASE(nt)                                                              \
@@ -426,6 +446,16 @@ void launch_combiner_fwd_impl(const void* X, const void* W,
           nullptr, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,         \
           mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, starts, paths,      \
           ends, (const bf16*)term, (const bf16*)path, TS, PS);                \
+    else if (epi == 1 && occ == 4 && nt <= 10)                                \
+      combiner_fwd_kernel<nt, 1, 0, 2, 4><<<grid, 256, smem, stream>>>(       \
+          (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, nullptr, nullptr,   \
+          nullptr, nullptr, nullptr, 0, 0);                                   \
+    else if (epi == 1 && occ == 3 && nt <= 10)                                \
+      combiner_fwd_kernel<nt, 1, 0, 2, 3><<<grid, 256, smem, stream>>>(       \
+          (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \
+          mean, rstd, M, KP, E, p, inv1mp, seed, rng_off, nullptr, nullptr,   \
+          nullptr, nullptr, nullptr, 0, 0);                                   \
     else if (epi == 1)                                                        \
       combiner_fwd_kernel<nt, 1, 0><<<grid, 256, smem, stream>>>(             \
           (const bf16*)X, (const bf16*)W, gamma, beta, (bf16*)out, (bf16*)z,  \

@@ -17,8 +17,8 @@
 
 // GATHER=1: the X operand is re-gathered from the embedding tables during
 // staging (the fused forward never materializes the concat tensor).
-template <int NT, int IT, int GATHER>
-__global__ __launch_bounds__(512) void wgrad_kernel(
+template <int NT, int IT, int GATHER, int MINW = 1>
+__global__ __launch_bounds__(512, MINW) void wgrad_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ dZ,
     float* __restrict__ partials, long M, int KP, int EP, long rows_per_block,
     const int* __restrict__ g_starts, const int* __restrict__ g_paths,
@@ -182,6 +182,10 @@ void launch_wgrad_impl(const void* X, const void* dZ, float* partials,
   const int smem = 2 * ((KP / 16) + (EP / 16)) * (64 * 8 + 8) * 2;
   const int NT = EP / 16;
   const int IT = (KP / 16 + 7) / 8;
+  // C2V_WGRAD_OCC=3/4 forces a waves/SIMD register cap (occupancy A/B;
+  // 512-thread blocks make 4 the only level that adds a resident block)
+  const char* occ_env = getenv("C2V_WGRAD_OCC");
+  const int occ = occ_env ? atoi(occ_env) : 0;
   // supported shapes: KP/16 divisible into 8 waves; dispatch common cases
 #define WCASE(nt, it)                                                        \
   if (NT == nt && IT == it) {                                                \
@@ -190,6 +194,16 @@ void launch_wgrad_impl(const void* X, const void* dZ, float* partials,
           nullptr, (const bf16*)dZ, partials, M, KP, EP, rows_per_block,     \
           starts, paths, ends, (const bf16*)term, (const bf16*)path, TS,     \
           PS);                                                               \
+    else if (occ == 4)                                                       \
+      wgrad_kernel<nt, it, 0, 4><<<nblocks, 512, smem, stream>>>(            \
+          (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,              \
+          rows_per_block, nullptr, nullptr, nullptr, nullptr, nullptr, 0,    \
+          0);                                                                \
+    else if (occ == 3)                                                       \
+      wgrad_kernel<nt, it, 0, 3><<<nblocks, 512, smem, stream>>>(            \
+          (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,              \
+          rows_per_block, nullptr, nullptr, nullptr, nullptr, nullptr, 0,    \
+          0);                                                                \
     else                                                                     \
       wgrad_kernel<nt, it, 0><<<nblocks, 512, smem, stream>>>(               \
           (const bf16*)X, (const bf16*)dZ, partials, M, KP, EP,              \
