@@ -71,7 +71,10 @@ __global__ __launch_bounds__(256) void head_bwd_prep_kernel(
 // 69 us); (b) staging cv rows through LDS fragment images per stage —
 // 16 extra ds_write_b16 per thread per stage on the CU-shared LDS pipe
 // (~33 us of skeleton).
-__global__ __launch_bounds__(512) void head_bwd_dw_kernel(
+// MINW: forced min waves/SIMD (C2V_HBDW_OCC env A/B; 512-thread blocks
+// make 6 the next level above the default 4 — costs 136 B/lane scratch)
+template <int MINW = 1>
+__global__ __launch_bounds__(512, MINW) void head_bwd_dw_kernel(
     const bf16* __restrict__ logits, const bf16* __restrict__ cvimg,
     const float* __restrict__ coef_lse, bf16* __restrict__ dw,
     float* __restrict__ dbias, long B, long L, int variant) {
@@ -541,9 +544,15 @@ void launch_head_bwd_dw(const void* logits, const void* cvimg,
   const int grid = (int)((L + HB_LB - 1) / HB_LB);
   const char* ve = getenv("C2V_HBDW_VARIANT");
   const int variant = ve ? atoi(ve) : 0;
-  head_bwd_dw_kernel<<<grid, 512, 0, stream>>>(
-      (const bf16*)logits, (const bf16*)cvimg, coef_lse, (bf16*)dw, dbias, B,
-      L, variant);
+  const char* occ_env = getenv("C2V_HBDW_OCC");
+  if (occ_env && occ_env[0] == '6')
+    head_bwd_dw_kernel<6><<<grid, 512, 0, stream>>>(
+        (const bf16*)logits, (const bf16*)cvimg, coef_lse, (bf16*)dw, dbias,
+        B, L, variant);
+  else
+    head_bwd_dw_kernel<<<grid, 512, 0, stream>>>(
+        (const bf16*)logits, (const bf16*)cvimg, coef_lse, (bf16*)dw, dbias,
+        B, L, variant);
 }
 
 void launch_head_bwd_dcv(const void* logits, const void* wimg,
