@@ -46,8 +46,11 @@
 // AIMG=1: the A operand arrives as the swizzle_a fragment image (wimg
 // aliasing the w pointer) — contiguous 1-KB wave reads instead of
 // 256-B-strided per-lane row loads (the large-L fix; EP=128 only).
-template <int NKT, int STATS, int AIMG = 0>
-__global__ __launch_bounds__(512) void head_fwd_kernel(
+// MINW=6: the unconstrained build allocates 92 VGPR (2 resident blocks =
+// 4 waves/SIMD); a forced 6-wave cap repacks to 74 VGPR with ZERO spill,
+// and LDS (37.9 KB) still fits the 3rd block.  C2V_HF_OCC=1 reverts.
+template <int NKT, int STATS, int AIMG = 0, int MINW = 6>
+__global__ __launch_bounds__(512, MINW) void head_fwd_kernel(
     const bf16* __restrict__ cv, const bf16* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ out,
     float* __restrict__ pm, float* __restrict__ ps, long B, long L, int EP,
@@ -410,8 +413,16 @@ void launch_head_fwd(const void* cv, const void* w, const float* bias,
   static const char* nt_env = getenv("C2V_HF_NT");
   const bool nt = nt_env ? (nt_env[0] == '1') : (L > 98304);
   const int variant = base_variant | (nt ? 0 : 32);
+  // C2V_HF_OCC=1 reverts to the unconstrained register allocation
+  static const char* occ_env = getenv("C2V_HF_OCC");
+  const bool hf_occ1 = occ_env && occ_env[0] == '1';
 #define HFCASE(nkt, st, ai)                                                  \
-  head_fwd_kernel<nkt, st, ai><<<grid, 512, 0, stream>>>(                    \
+  if (hf_occ1)                                                               \
+    head_fwd_kernel<nkt, st, ai, 1><<<grid, 512, 0, stream>>>(                \
+        (const bf16*)cv, (const bf16*)w, bias, (bf16*)out, pm, ps, B, L,     \
+        EP, GYB, variant);                                                    \
+  else                                                                        \
+    head_fwd_kernel<nkt, st, ai><<<grid, 512, 0, stream>>>(                   \
       (const bf16*)cv, (const bf16*)w, bias, (bf16*)out, pm, ps, B, L, EP,   \
       GYB, variant)
   if (EP == 128) {
