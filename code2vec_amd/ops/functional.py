@@ -672,18 +672,14 @@ def head_logits_with_stats(cv_bf16, w, bias):
     B, EP = cv_bf16.shape
     L = w.shape[0]
     with torch.no_grad():
-        if _HEAD_FWD and EP % 32 == 0 and L <= _HEAD_FWD_MAXL:
-            logits = torch.empty(B, L, dtype=torch.bfloat16,
-                                 device=cv_bf16.device)
-            gx = (L + 255) // 256
-            pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
-            ps = torch.empty_like(pm)
-            ext().head_fwd(cv_bf16, w, bias.float(), logits, pm, ps)
-            logits._c2v_lsm_partials = (pm, ps)
-        elif _HEAD_FWD and _HF_AIMG and EP == 128 and L % 8 == 0:
-            # large L: W pre-swizzled into the A-fragment image so the
-            # custom kernel's W reads are contiguous (strided per-lane
-            # row loads streamed W from HBM inefficiently at L=261k)
+        if _HEAD_FWD and _HF_AIMG and EP == 128 and L % 8 == 0:
+            # W pre-swizzled into the A-fragment image so the kernel's W
+            # reads are contiguous 1-KB wave reads.  Originally the
+            # large-L fix (strided per-lane row loads streamed W from
+            # HBM inefficiently at L=261k), but measured faster than the
+            # direct-load kernel at top11 scale too (63.7 vs 79.4 us at
+            # L=72,416; step 1.337 vs 1.350 ms) — the swizzle cost
+            # (~7 us, L2-resident) is well under the fragment-read win.
             wimg = _scratch_bf16("hf_aimg",
                                  ((L + 255) // 256 * 16, 4, 64, 8),
                                  w.device)
@@ -695,6 +691,16 @@ def head_logits_with_stats(cv_bf16, w, bias):
             ps = torch.empty_like(pm)
             ext().head_fwd_img(cv_bf16, wimg, bias.float(), logits, pm, ps,
                                L)
+            logits._c2v_lsm_partials = (pm, ps)
+        elif _HEAD_FWD and EP % 32 == 0 and L <= _HEAD_FWD_MAXL:
+            # direct-load kernel: the EP != 128 path (C2V_HF_AIMG=0 also
+            # lands here below C2V_HEAD_FWD_MAXL)
+            logits = torch.empty(B, L, dtype=torch.bfloat16,
+                                 device=cv_bf16.device)
+            gx = (L + 255) // 256
+            pm = torch.empty(gx, B, dtype=torch.float32, device=w.device)
+            ps = torch.empty_like(pm)
+            ext().head_fwd(cv_bf16, w, bias.float(), logits, pm, ps)
             logits._c2v_lsm_partials = (pm, ps)
         else:
             logits = torch.nn.functional.linear(
