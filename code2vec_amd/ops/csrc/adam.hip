@@ -20,9 +20,10 @@ __global__ void adam_tick_kernel(double* __restrict__ bc_pow, float b1,
   }
 }
 
-template <int NT, int U>  // NT: nontemporal; U: independent 4-elem chunks
-                          // per loop iteration (ILP / outstanding loads)
-__global__ void adam_bf16_kernel(bf16* __restrict__ p,
+template <int NT, int U, int MINW = 1>  // NT: nontemporal; U: independent
+                          // 4-elem chunks per loop iteration; MINW: forced
+                          // waves/SIMD (C2V_ADAM_OCC occupancy A/B)
+__global__ __launch_bounds__(256, MINW) void adam_bf16_kernel(bf16* __restrict__ p,
                                  const bf16* __restrict__ g,
                                  float* __restrict__ master,
                                  float* __restrict__ m, float* __restrict__ v,
@@ -138,11 +139,19 @@ void launch_adam_bf16(void* p, const void* g, float* master, float* m,
   const int grid = (int)min(want > 0 ? want : 1, cap);
   const char* nt_env = getenv("C2V_ADAM_NT");
   const int nt = (nt_env == nullptr || nt_env[0] != '0') ? 1 : 0;
+  // C2V_ADAM_OCC=5 forces a 5-waves/SIMD register cap (occupancy A/B)
+  static const char* occ_env = getenv("C2V_ADAM_OCC");
+  const bool occ5 = occ_env && occ_env[0] == '5';
 #define ADAM_CASE(NTV, UV)                                                  \
   if (nt == NTV && U == UV) {                                               \
-    adam_bf16_kernel<NTV, UV><<<grid, block, 0, stream>>>(                  \
-        (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,     \
-        bc_pow);                                                            \
+    if (occ5)                                                               \
+      adam_bf16_kernel<NTV, UV, 5><<<grid, block, 0, stream>>>(             \
+          (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,   \
+          bc_pow);                                                          \
+    else                                                                    \
+      adam_bf16_kernel<NTV, UV><<<grid, block, 0, stream>>>(                \
+          (bf16*)p, (const bf16*)g, master, m, v, n, lr, b1, b2, eps, wd,   \
+          bc_pow);                                                          \
     return;                                                                 \
   }
   ADAM_CASE(1, 1) ADAM_CASE(1, 2) ADAM_CASE(1, 4)

@@ -102,8 +102,9 @@ __global__ void gather_concat_bwd_kernel(
 // Entry o in [0, N): o < M -> gout row o at column offset off0;
 // o >= M -> gout row (o - M) at offset off1 (start/end entries of the
 // terminal table share one sort; the path table passes N == M).
-template <int NP>  // ceil(S/128) column-pair iterations per lane
-__global__ __launch_bounds__(256) void embed_scatter_sorted_kernel(
+template <int NP, int MINW = 1>  // NP: ceil(S/128) column-pair iterations
+                                  // per lane; MINW: forced waves/SIMD
+__global__ __launch_bounds__(256, MINW) void embed_scatter_sorted_kernel(
     const int* __restrict__ sorted_idx, const long* __restrict__ perm,
     const bf16* __restrict__ gout, float* __restrict__ dtable,
     bf16* __restrict__ out_bf16, unsigned char* __restrict__ flags, long N,
@@ -462,11 +463,23 @@ void launch_embed_scatter_sorted(const int* sorted_idx, const long* perm,
   const int wpb = 4;
   const int grid = (int)((waves + wpb - 1) / wpb);
   const int np = (S + 127) / 128;
+  // C2V_ESS_OCC=5/6 forces a waves/SIMD register cap (occupancy A/B)
+  static const char* occ_env = getenv("C2V_ESS_OCC");
+  const int occ = occ_env ? atoi(occ_env) : 0;
 #define SCASE(n)                                                              \
   case n:                                                                     \
-    embed_scatter_sorted_kernel<n><<<grid, 256, 0, stream>>>(                 \
-        sorted_idx, perm, (const bf16*)gout, dtable, (bf16*)out_bf16, flags,  \
-        N, M, KP, S, off0, off1, R);                                          \
+    if (occ == 6)                                                             \
+      embed_scatter_sorted_kernel<n, 6><<<grid, 256, 0, stream>>>(            \
+          sorted_idx, perm, (const bf16*)gout, dtable, (bf16*)out_bf16,       \
+          flags, N, M, KP, S, off0, off1, R);                                 \
+    else if (occ == 5)                                                        \
+      embed_scatter_sorted_kernel<n, 5><<<grid, 256, 0, stream>>>(            \
+          sorted_idx, perm, (const bf16*)gout, dtable, (bf16*)out_bf16,       \
+          flags, N, M, KP, S, off0, off1, R);                                 \
+    else                                                                      \
+      embed_scatter_sorted_kernel<n><<<grid, 256, 0, stream>>>(               \
+          sorted_idx, perm, (const bf16*)gout, dtable, (bf16*)out_bf16,       \
+          flags, N, M, KP, S, off0, off1, R);                                 \
     break;
   switch (np) {
     SCASE(1) SCASE(2) SCASE(3) SCASE(4)
