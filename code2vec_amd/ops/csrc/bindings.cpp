@@ -83,6 +83,9 @@ void launch_norm_project(const void*, const void*, const float*, void*,
                          long, hipStream_t);
 void launch_head_bwd_dw(const void*, const void*, const float*, void*,
                         float*, long, long, hipStream_t);
+void launch_head_bwd_dcv_rc(const void*, const void*, const void*,
+                            const float*, float*, long, long, int,
+                            hipStream_t);
 void launch_head_bwd_dcv(const void*, const void*, const float*, float*,
                          long, long, int, hipStream_t);
 void launch_head_fwd(const void*, const void*, const float*, void*, float*,
@@ -570,6 +573,33 @@ void head_bwd_dcv(torch::Tensor logits, torch::Tensor wimg,
                       cur_stream());
 }
 
+// Streaming dcv: logits never read — recomputed per wave from the
+// swizzle_a fragment images of cv ([B,128]) and W ([L,128]).
+void head_bwd_dcv_rc(torch::Tensor cvimg_a, torch::Tensor wimg_a,
+                     torch::Tensor wimg, torch::Tensor coef_lse,
+                     torch::Tensor partials, long B, long L, long chunk) {
+  CHK_CUDA(wimg); CHK_CONTIG(wimg); CHK_DT(wimg, torch::kBFloat16);
+  CHK_CONTIG(cvimg_a); CHK_DT(cvimg_a, torch::kBFloat16);
+  CHK_CONTIG(wimg_a); CHK_DT(wimg_a, torch::kBFloat16);
+  CHK_CONTIG(coef_lse); CHK_DT(coef_lse, torch::kFloat32);
+  CHK_DT(partials, torch::kFloat32); CHK_CONTIG(partials);
+  TORCH_CHECK(wimg.numel() == (L + 127) / 128 * 4 * 4096,
+              "wimg must be [ceil(L/128)*4, 8, 64, 8]");
+  TORCH_CHECK(cvimg_a.numel() >= (B + 127) / 128 * 8 * 2048,
+              "cvimg_a too small for B");
+  TORCH_CHECK(wimg_a.numel() >= (L + 255) / 256 * 16 * 2048,
+              "wimg_a too small for L");
+  TORCH_CHECK(L % 8 == 0 && chunk % 128 == 0, "head_bwd_dcv_rc shape gates");
+  TORCH_CHECK(L < (1L << 24), "label index carried as f32 needs L < 2^24");
+  TORCH_CHECK(coef_lse.numel() == 4 * B, "coef_lse must be [B, 4]");
+  TORCH_CHECK(partials.numel() == (L + chunk - 1) / chunk * B * 128,
+              "head_bwd_dcv_rc partials shape");
+  launch_head_bwd_dcv_rc(cvimg_a.data_ptr(), wimg_a.data_ptr(),
+                         wimg.data_ptr(), coef_lse.data_ptr<float>(),
+                         partials.data_ptr<float>(), B, L, (int)chunk,
+                         cur_stream());
+}
+
 // K11 angular-margin building blocks (all [N, 128] bf16 row spaces)
 void inv_rownorm(torch::Tensor x, torch::Tensor inv) {
   CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kBFloat16);
@@ -770,6 +800,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_project", &norm_project);
   m.def("head_bwd_dw", &head_bwd_dw);
   m.def("head_bwd_dcv", &head_bwd_dcv);
+  m.def("head_bwd_dcv_rc", &head_bwd_dcv_rc);
   m.def("transpose_w", &transpose_w);
   m.def("dgrad2", &dgrad2);
   m.def("slab_sum_bf16", &slab_sum_bf16);

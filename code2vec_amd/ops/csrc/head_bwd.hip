@@ -507,6 +507,131 @@ __global__ __launch_bounds__(512) void head_bwd_dcv_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// STREAMING variant of the dcv kernel: the logits tensor is never READ —
+// each wave recomputes its [16 batch x 128 label] logits tile with a first
+// MFMA pass (cv and W arrive as swizzle_a fragment images, so both operand
+// streams are contiguous 1-KB wave reads), applies the softmax-gradient
+// transform in registers, transposes G through a wave-PRIVATE LDS
+// A-fragment image (producer wave == consumer wave, so no extra barriers),
+// and runs the usual split-K accumulate against the staged W B-image.
+// Removes the 2-B/elem logits read (534 MB at java-large) at the price of
+// duplicating the forward GEMM's FLOPs (~68 GFLOP, ~3% of one MFMA-second)
+// — the trade the MI355X's 2.5-PFLOP/s : 5.5-TB/s ratio is built for.
+//
+// The recomputed logits are the f32 MFMA accumulators — one bf16 rounding
+// CLOSER to the forward's online-softmax stats (which were computed from
+// the same f32 tiles) than the stored-logits kernel above.
+__global__ __launch_bounds__(512) void head_bwd_dcv_rc_kernel(
+    const bf16* __restrict__ cvimg_a, const bf16* __restrict__ wimg_a,
+    const bf16* __restrict__ wimg, const float* __restrict__ coef_lse,
+    float* __restrict__ partials, long B, long L, int chunk, int GYB) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int total = gridDim.x;
+  const int lin = (total % 8 == 0)
+      ? (int)(blockIdx.x % 8) * (total / 8) + (int)blockIdx.x / 8
+      : (int)blockIdx.x;
+  const int by = lin % GYB;
+  const int sc = lin / GYB;
+  const long bt0 = (long)by * 128;
+
+  // per-lane meta for its 4 MFMA1 C rows (batch = bt0 + wave*16 + g4*4+r)
+  float coef4[4], lse4[4], yf4[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long row = bt0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (row < B) {
+      const f32x4 meta = *(const f32x4*)(coef_lse + 4 * row);
+      coef4[r] = meta[0];
+      lse4[r] = meta[1];
+      yf4[r] = meta[2];
+    } else {
+      coef4[r] = 0.f;  // coef 0 => g == 0 for out-of-range rows
+      lse4[r] = 0.f;
+      yf4[r] = -1.f;
+    }
+  }
+  // MFMA1 A operand: this wave's 16 cv rows, all 128 k — loaded ONCE
+  // (cvimg_a rows past B are zero-padded by swizzle_a)
+  const bf16* ca = cvimg_a + ((bt0 >> 4) + wave) * 4 * 512 + lane * 8;
+  const bf16x8 acv0 = *(const bf16x8*)(ca);
+  const bf16x8 acv1 = *(const bf16x8*)(ca + 512);
+  const bf16x8 acv2 = *(const bf16x8*)(ca + 1024);
+  const bf16x8 acv3 = *(const bf16x8*)(ca + 1536);
+
+  __shared__ bf16 wst[4 * 8 * 512];       // MFMA2 W B-fragments (32 KB)
+  __shared__ bf16 gst[8][4][512];         // per-wave G A-fragment image
+
+  f32x4 acc[8];
+#pragma unroll
+  for (int nt = 0; nt < 8; ++nt) acc[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nsub = chunk / 128;
+  for (int s = 0; s < nsub; ++s) {
+    const long l0s = (long)sc * chunk + (long)s * 128;
+    if (l0s >= L) break;
+    // stage the MFMA2 W image (contiguous 32-KB copy, zero-padded past L)
+    const bf16* src = wimg + (l0s >> 5) * 4096;
+    for (int t = threadIdx.x; t < 2048; t += 512)
+      *(bf16x8*)&wst[t * 8] = *(const bf16x8*)(src + t * 8);
+    __syncthreads();
+
+    // MFMA1: recompute the wave's [16 x 128] logits tile and emit G
+    const bf16* wa = wimg_a + (l0s >> 4) * 4 * 512 + lane * 8;
+#pragma unroll
+    for (int nt2 = 0; nt2 < 8; ++nt2) {
+      f32x4 c = {0.f, 0.f, 0.f, 0.f};
+      const bf16* wp = wa + (long)nt2 * 4 * 512;
+      c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          acv0, *(const bf16x8*)(wp), c, 0, 0, 0);
+      c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          acv1, *(const bf16x8*)(wp + 512), c, 0, 0, 0);
+      c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          acv2, *(const bf16x8*)(wp + 1024), c, 0, 0, 0);
+      c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          acv3, *(const bf16x8*)(wp + 1536), c, 0, 0, 0);
+      const long lab = l0s + nt2 * 16 + (lane & 15);
+      const int lab32 = (nt2 & 1) * 16 + (lane & 15);
+      const int ebase = ((lab32 >> 3) * 16) * 8 + (lab32 & 7);
+      bf16* gout = gst[wave][nt2 >> 1];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float g = 0.f;
+        if (lab < L) {
+          g = coef4[r] * __expf(c[r] - lse4[r]);
+          if ((float)lab == yf4[r]) g -= coef4[r];
+        }
+        gout[ebase + (((lane >> 4) * 4 + r)) * 8] = f2bf(g);
+      }
+    }
+    // MFMA2: split-K accumulate (A = own-wave G image, B = staged W)
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      const bf16x8 a = *(const bf16x8*)&gst[wave][kk][lane * 8];
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt) {
+        const bf16x8 b =
+            *(const bf16x8*)&wst[(kk * 8 + nt) * 512 + lane * 8];
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt],
+                                                          0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  float* slab = partials + (long)sc * B * 128;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long grow = bt0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (grow < B) {
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt)
+        slab[grow * 128 + nt * 16 + (lane & 15)] = acc[nt][r];
+    }
+  }
+}
+
 extern "C" {
 
 void launch_head_bwd_prep(const long* label, const float* weight,
@@ -563,6 +688,17 @@ void launch_head_bwd_dcv(const void* logits, const void* wimg,
   head_bwd_dcv_kernel<<<GYB * split, 512, 0, stream>>>(
       (const bf16*)logits, (const bf16*)wimg, coef_lse, partials, B, L,
       chunk, GYB);
+}
+
+void launch_head_bwd_dcv_rc(const void* cvimg_a, const void* wimg_a,
+                            const void* wimg, const float* coef_lse,
+                            float* partials, long B, long L, int chunk,
+                            hipStream_t stream) {
+  const int GYB = (int)((B + 127) / 128);
+  const int split = (int)((L + chunk - 1) / chunk);
+  head_bwd_dcv_rc_kernel<<<GYB * split, 512, 0, stream>>>(
+      (const bf16*)cvimg_a, (const bf16*)wimg_a, (const bf16*)wimg,
+      coef_lse, partials, B, L, chunk, GYB);
 }
 
 }  // extern "C"

@@ -40,6 +40,10 @@ _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "98304"))
 # (contiguous A reads; re-enables the fused stats epilogue at java-large
 # scale).  C2V_HF_AIMG=0 falls back to hipBLASLt + lsm_partial there.
 _HF_AIMG = os.environ.get("C2V_HF_AIMG", "1") == "1"
+# streaming head backward: recompute logits by MFMA inside dcv instead of
+# reading the [B, L] tensor (C2V_HB_RC=0 reverts to the logits-reading
+# kernel)
+_HB_RC = os.environ.get("C2V_HB_RC", "0") == "1"
 # custom split-K dcv in the head backward (C2V_HEAD_DGRAD=0 -> rocBLAS)
 _HEAD_DGRAD = os.environ.get("C2V_HEAD_DGRAD", "1") == "1"
 # combiner dgrad through dgrad2.hip (C2V_DGRAD2=0 -> rocBLAS)
@@ -692,6 +696,7 @@ def head_logits_with_stats(cv_bf16, w, bias):
             ext().head_fwd_img(cv_bf16, wimg, bias.float(), logits, pm, ps,
                                L)
             logits._c2v_lsm_partials = (pm, ps)
+            logits._c2v_wimg_a = wimg  # reusable by the streaming backward
         elif _HEAD_FWD and EP % 32 == 0 and L <= _HEAD_FWD_MAXL:
             # direct-load kernel: the EP != 128 path (C2V_HF_AIMG=0 also
             # lands here below C2V_HEAD_FWD_MAXL)
@@ -737,6 +742,9 @@ class FusedHeadLoss(torch.autograd.Function):
             acc = _lsm_stats(logits, label, weight, lse)
         loss = acc[0] / acc[1]
         ctx.save_for_backward(logits, cv_bf16, w, label, weight, lse, acc)
+        # saved_tensors re-wraps the tensor object, so python attributes
+        # do not survive — carry the forward's W fragment image on ctx
+        ctx.wimg_a = getattr(logits, "_c2v_wimg_a", None)
         return loss
 
     @staticmethod
@@ -771,7 +779,23 @@ class FusedHeadLoss(torch.autograd.Function):
                              dev)
         ext().swizzle_cv(w, wimg)
         partials = _scratch_f32("head_fused_dcv", (split, B, 128), dev)
-        ext().head_bwd_dcv(logits, wimg, coef_lse, partials, chunk)
+        if _HB_RC:
+            # STREAMING dcv: the [B, L] logits are never read — each wave
+            # recomputes its logits tile by MFMA from the swizzle_a
+            # fragment images of cv and W (the W image is reused from the
+            # forward when head_fwd_img ran)
+            cvimg_a = _scratch_bf16(
+                "head_cvimg_a", ((B + 255) // 256 * 16, 4, 64, 8), dev)
+            ext().swizzle_a(cv.contiguous(), cvimg_a)
+            wimg_a = ctx.wimg_a
+            if wimg_a is None:
+                wimg_a = _scratch_bf16(
+                    "hf_aimg", ((L + 255) // 256 * 16, 4, 64, 8), dev)
+                ext().swizzle_a(w, wimg_a)
+            ext().head_bwd_dcv_rc(cvimg_a, wimg_a, wimg, coef_lse,
+                                  partials, B, L, chunk)
+        else:
+            ext().head_bwd_dcv(logits, wimg, coef_lse, partials, chunk)
         dcv = torch.empty(B, 128, dtype=torch.bfloat16, device=dev)
         ext().slab_sum_bf16(partials, dcv)
         return None, dcv, dw, dbias, None, None

@@ -590,6 +590,38 @@ class TestFusedHeadLoss:
         assert relerr(dw.float(), wr.grad) < 3e-2
         assert relerr(dbias.float(), br.grad) < 3e-2
 
+    @pytest.mark.parametrize("B,L", [(96, 7320), (64, 1992), (1024, 4096)])
+    def test_streaming_dcv_matches(self, dev, B, L):
+        """C2V_HB_RC path: dcv recomputed by MFMA from the cv/W fragment
+        images (logits never read) must match both the logits-reading
+        kernel (bitwise-near: same math, one rounding closer to f32) and
+        the fp32 oracle."""
+        from code2vec_amd.ops import functional as Fn
+
+        cv, w, bias, label, weight = self._setup(dev, B, L, seed=57)
+        loss_a, dcv_a, dw_a, db_a = self._run_fused(cv, w, bias, label,
+                                                    weight)
+        old = Fn._HB_RC
+        Fn._HB_RC = True
+        try:
+            loss_b, dcv_b, dw_b, db_b = self._run_fused(cv, w, bias, label,
+                                                        weight)
+        finally:
+            Fn._HB_RC = old
+        assert abs(float(loss_a.detach()) - float(loss_b.detach())) <= (
+            1e-6 + 1e-5 * abs(float(loss_a.detach())))
+        # the RC path recomputes logits in f32 where the reading path
+        # consumes their bf16 rounding — per-element g shifts up to a few
+        # tenths of a percent, summed into dcv
+        assert relerr(dcv_b.float(), dcv_a.float()) < 5e-2
+        assert torch.equal(dw_a, dw_b)  # dw path identical in both modes
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        ref = R.logsoftmax_nll(cvr @ wr.t() + br, label, weight)
+        ref.backward()
+        assert relerr(dcv_b.float(), cvr.grad) < 3e-2
+
     def test_matches_unfused_chain(self, dev):
         """Fused backward vs the unfused OutputHead+FusedLogSoftmaxNLL
         chain on identical inputs (both bf16 paths -> tight tolerance);

@@ -211,6 +211,17 @@ def main():
         results["head_bwd_dcv"] = timeit(
             lambda: ext().head_bwd_dcv(logits, wimg, coef_lse, partials,
                                        chunk))
+        wimg_a = torch.empty((L + 255) // 256 * 16, 4, 64, 8,
+                             dtype=torch.bfloat16, device=dev)
+        ext().swizzle_a(wout, wimg_a)
+        cvimg_a = torch.empty((B + 255) // 256 * 16, 4, 64, 8,
+                              dtype=torch.bfloat16, device=dev)
+        ext().swizzle_a(cvb, cvimg_a)
+        results["swizzle_a(cv)"] = timeit(
+            lambda: ext().swizzle_a(cvb, cvimg_a))
+        results["head_bwd_dcv_rc"] = timeit(
+            lambda: ext().head_bwd_dcv_rc(cvimg_a, wimg_a, wimg, coef_lse,
+                                          partials, B, L, chunk))
 
     for k, vv in results.items():
         print(f"{k:24s} {vv:10.1f} us")
