@@ -81,6 +81,12 @@ pm = torch.empty(gx, B, dtype=torch.float32, device=dev)
 ps = torch.empty_like(pm)
 for _ in range(REPS):
     ext().head_fwd(cv.to(torch.bfloat16), wout, bias, logits, pm, ps)
+wimg_a = torch.empty((L + 255) // 256 * 16, 4, 64, 8, dtype=torch.bfloat16,
+                     device=dev)
+ext().swizzle_a(wout, wimg_a)
+for _ in range(REPS):
+    ext().head_fwd_img(cv.to(torch.bfloat16), wimg_a, bias, logits, pm, ps,
+                       L)
 lse = torch.empty(B, dtype=torch.float32, device=dev)
 acc_f = torch.empty((B + 15) // 16, 2, dtype=torch.float32, device=dev)
 ext().logsoftmax_nll_finalize(logits, pm, ps, label, weight, lse, acc_f)
