@@ -2,6 +2,7 @@
 config 1 shape) + artifact format round-trips."""
 
 import os
+import sys
 
 import numpy as np
 import pytest
@@ -323,3 +324,29 @@ def test_floyd_metric_line_format(tmp_path, tiny_corpus, capsys):
                  "recall", "f1", "best_f1"):
         assert f'{{"metric": "{name}", "value": ' in out, name
     assert "epoch 0" in out
+
+
+def test_bench_driver_contract(tmp_path):
+    """bench.py is the driver's measurement contract: with no GPU it must
+    still run (torch reference model), finish quickly at the tiny config,
+    and print ONE final JSON line carrying every field the driver and the
+    judge consume."""
+    import json
+    import subprocess
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--config", "tiny",
+         "--steps", "2", "--warmup", "1", "--pool", "2"],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, key
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["warmup"] == 1
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert isinstance(d["config"], dict) and "global_batch" in d["config"]
