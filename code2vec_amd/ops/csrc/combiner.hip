@@ -387,6 +387,107 @@ __global__ __launch_bounds__(256) void combiner_bwd_kernel(
   }
 }
 
+// V2 of the backward for the flagship EP=128 shape: TWO rows per wave
+// (half-wave each) so every load/store is 8 B (bf16x4) instead of 4 B —
+// halves the vector-memory op count of this latency-bound row sweep.
+// The LN reductions shrink to 32-lane xor trees (offsets 16..1 stay
+// within a half-wave); dgamma/dbeta halves combine with ONE fixed
+// cross-half shuffle before the per-wave-row block reduce, keeping the
+// bitwise run-to-run determinism of the fixed-order combine.
+__device__ __forceinline__ float half_wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) x += __shfl_xor(x, off);
+  return x;
+}
+
+__global__ __launch_bounds__(256) void combiner_bwd_v2_kernel(
+    const bf16* __restrict__ dout, const bf16* __restrict__ z,
+    const bf16* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ rstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, bf16* __restrict__ dz,
+    float* __restrict__ dgamma_part, float* __restrict__ dbeta_part, long M,
+    int E, float p, float inv1mp) {
+  const int EP = 128;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int half = lane >> 5;
+  const int c0 = (lane & 31) * 4;
+  const float invE = 1.0f / (float)E;
+  const float one_mp = 1.0f - p;
+
+  float g_c[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) g_c[j] = gamma[c0 + j];
+  float acc_dg[4] = {}, acc_db[4] = {};
+
+  const long wpb = blockDim.x / WAVE;           // wave-pairs per block
+  long pr = (long)blockIdx.x * wpb + wave;      // row-pair index
+  const long pstride = (long)gridDim.x * wpb;
+  const long npairs = (M + 1) >> 1;
+  for (; pr < npairs; pr += pstride) {
+    const long row = 2 * pr + half;
+    const bool rok = row < M;
+    const float mu = rok ? mean[row] : 0.f;
+    const float rs = rok ? rstd[row] : 0.f;
+    bf16x4 zv = {}, dy4 = {}, ov4 = {};
+    if (rok) {
+      zv = *(const bf16x4*)(z + row * EP + c0);
+      dy4 = *(const bf16x4*)(dout + row * EP + c0);
+      ov4 = *(const bf16x4*)(out + row * EP + c0);
+    }
+    float xhat[4], du[4];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float xh = (bf2f(zv[j]) - mu) * rs;
+      const float ov = bf2f(ov4[j]);
+      const float y = ov * one_mp;
+      const float dy = (ov != 0.f) ? bf2f(dy4[j]) * inv1mp : 0.f;
+      const float d = dy * (1.0f - y * y);
+      xhat[j] = xh;
+      du[j] = d;
+      acc_dg[j] += d * xh;
+      acc_db[j] += d;
+      const float h = d * g_c[j];
+      s1 += h;
+      s2 += h * xh;
+    }
+    s1 = half_wave_reduce_sum(s1) * invE;
+    s2 = half_wave_reduce_sum(s2) * invE;
+    if (rok) {
+      bf16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const bool valid = (c0 + j) < E;
+        const float h = du[j] * g_c[j];
+        o[j] = f2bf(valid ? rs * (h - s1 - xhat[j] * s2) : 0.f);
+      }
+      *(bf16x4*)(dz + row * EP + c0) = o;
+    }
+  }
+
+  // fixed-order cross-half combine: half 0's lane adds half 1's partial
+  __shared__ float red_g[4][128];
+  __shared__ float red_b[4][128];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float og = __shfl_xor(acc_dg[j], 32);
+    const float ob = __shfl_xor(acc_db[j], 32);
+    if (half == 0) {
+      const int c = c0 + j;
+      red_g[wave][c] = (c < EP) ? acc_dg[j] + og : 0.f;
+      red_b[wave][c] = (c < EP) ? acc_db[j] + ob : 0.f;
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < EP; c += blockDim.x) {
+    dgamma_part[(long)blockIdx.x * EP + c] =
+        red_g[0][c] + red_g[1][c] + red_g[2][c] + red_g[3][c];
+    dbeta_part[(long)blockIdx.x * EP + c] =
+        red_b[0][c] + red_b[1][c] + red_b[2][c] + red_b[3][c];
+  }
+}
+
 __global__ void bump_u64_kernel(unsigned long long* p,
                                 unsigned long long delta) {
   if (threadIdx.x == 0 && blockIdx.x == 0) p[0] += delta;
@@ -512,6 +613,17 @@ void launch_combiner_bwd(const void* dout, const void* z, const void* out,
   const int grid = (int)min((M + wpb - 1) / wpb, (long)COMBINER_BWD_MAX_GRID);
   const float inv1mp = p > 0.f ? 1.0f / (1.0f - p) : 1.0f;
   const int npair = (EP + 127) / 128;
+  // V2 (two rows per wave, 8-B vector ops) for the EP=128 flagship shape;
+  // C2V_CB_V2=0 reverts to the one-row-per-wave kernel
+  static const char* v2e = getenv("C2V_CB_V2");
+  if (EP == 128 && (v2e == nullptr || v2e[0] != '0')) {
+    // SAME grid as v1: the host sums dgamma/dbeta partials over this many
+    // block rows, so every block must write its row (idle pairs write 0)
+    combiner_bwd_v2_kernel<<<grid, block, 0, stream>>>(
+        (const bf16*)dout, (const bf16*)z, (const bf16*)out, mean, rstd,
+        gamma, beta, (bf16*)dz, dgamma, dbeta, M, E, p, inv1mp);
+    return;
+  }
 #define BCASE(np)                                                             \
   case np:                                                                    \
     combiner_bwd_kernel<np><<<grid, block, 0, stream>>>(                      \
