@@ -58,12 +58,30 @@ class FusedAdam:
         path) — the step counter is bumped here, once."""
         self.step_count += 1
         Fn.ext().adam_tick(self.bc_pow, self.beta1, self.beta2)
+        # fp32 params (LN gamma/beta, attention vector, head bias) batch
+        # into ONE multi-tensor launch; elementwise Adam makes the result
+        # bitwise-identical to per-tensor calls
+        f32_batch = ([], [], [], [])
         for p in self.params:
             if p.grad is None or (exclude_ids and id(p) in exclude_ids):
                 continue
             st = self.state[p]
+            if (st["master"] is None and p.dtype == torch.float32
+                    and p.is_cuda and p.grad.dtype == torch.float32
+                    and p.grad.is_contiguous() and len(f32_batch[0]) < 8):
+                f32_batch[0].append(p.data.view(-1))
+                f32_batch[1].append(p.grad.view(-1))
+                f32_batch[2].append(st["m"])
+                f32_batch[3].append(st["v"])
+                continue
             Fn.adam_step(
                 p.data, p.grad, st["master"], st["m"], st["v"],
+                self.bc_pow, self.lr, self.beta1, self.beta2,
+                self.eps, self.weight_decay,
+            )
+        if f32_batch[0]:
+            Fn.ext().adam_step_f32_multi(
+                f32_batch[0], f32_batch[1], f32_batch[2], f32_batch[3],
                 self.bc_pow, self.lr, self.beta1, self.beta2,
                 self.eps, self.weight_decay,
             )

@@ -116,7 +116,68 @@ __global__ void adam_f32_kernel(float* __restrict__ p,
   }
 }
 
+// Multi-tensor f32 Adam: the model's small fp32 params (LN gamma/beta,
+// attention vector, head bias) update in ONE launch instead of four —
+// grid-strides the concatenated element space, resolving the owning
+// tensor with an unrolled prefix-offset scan (<= 8 tensors).
+#define ADAM_MT_MAX 8
+struct AdamF32Args {
+  float* p[ADAM_MT_MAX];
+  const float* g[ADAM_MT_MAX];
+  float* m[ADAM_MT_MAX];
+  float* v[ADAM_MT_MAX];
+  long off[ADAM_MT_MAX + 1];
+};
+
+__global__ void adam_f32_multi_kernel(AdamF32Args args, int ntens,
+                                      long total, float lr, float b1,
+                                      float b2, float eps, float wd,
+                                      const double* __restrict__ bc_pow) {
+  const float inv_bc1 = (float)(1.0 / (1.0 - bc_pow[0]));
+  const float inv_bc2 = (float)(1.0 / (1.0 - bc_pow[1]));
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < total; j += stride) {
+    int t = 0;
+#pragma unroll
+    for (int k = 1; k < ADAM_MT_MAX; ++k)
+      if (k < ntens && j >= args.off[k]) t = k;
+    const long e = j - args.off[t];
+    float* p = args.p[t];
+    const float* g = args.g[t];
+    float* m = args.m[t];
+    float* v = args.v[t];
+    float grad = g[e] + wd * p[e];
+    m[e] = b1 * m[e] + (1.f - b1) * grad;
+    v[e] = b2 * v[e] + (1.f - b2) * grad * grad;
+    p[e] -= lr * (m[e] * inv_bc1) / (sqrtf(v[e] * inv_bc2) + eps);
+  }
+}
+
 extern "C" {
+
+void launch_adam_f32_multi(float** ps, const float** gs, float** ms,
+                           float** vs, const long* ns, int ntens,
+                           const double* bc_pow, float lr, float b1,
+                           float b2, float eps, float wd,
+                           hipStream_t stream) {
+  AdamF32Args a;
+  long total = 0;
+  for (int t = 0; t < ntens; ++t) {
+    a.p[t] = ps[t]; a.g[t] = gs[t]; a.m[t] = ms[t]; a.v[t] = vs[t];
+    a.off[t] = total;
+    total += ns[t];
+  }
+  a.off[ntens] = total;
+  for (int t = ntens; t < ADAM_MT_MAX; ++t) {
+    a.p[t] = nullptr; a.g[t] = nullptr; a.m[t] = nullptr; a.v[t] = nullptr;
+    if (t > ntens) a.off[t] = total;
+  }
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)4096);
+  adam_f32_multi_kernel<<<grid, block, 0, stream>>>(
+      a, ntens, total, lr, b1, b2, eps, wd, bc_pow);
+}
 
 void launch_adam_tick(double* bc_pow, float b1, float b2,
                       hipStream_t stream) {

@@ -96,6 +96,9 @@ void launch_lsm_finalize(const void*, const float*, const float*, const long*,
                          hipStream_t);
 void launch_lsm_partial(const void*, float*, float*, int, long, hipStream_t);
 void launch_adam_tick(double*, float, float, hipStream_t);
+void launch_adam_f32_multi(float**, const float**, float**, float**,
+                           const long*, int, const double*, float, float,
+                           float, float, float, hipStream_t);
 void launch_adam_bf16(void*, const void*, float*, float*, float*, long,
                       const double*, float, float, float, float, float,
                       hipStream_t);
@@ -766,6 +769,39 @@ void adam_step_f32(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                   (float)b2, (float)eps, (float)wd, cur_stream());
 }
 
+// One launch for up to 8 small fp32 params (grads must be fp32 and
+// contiguous); element-order within each tensor identical to the
+// per-tensor kernel, so the update is bitwise-equal to 4 separate calls.
+void adam_step_f32_multi(std::vector<torch::Tensor> ps,
+                         std::vector<torch::Tensor> gs,
+                         std::vector<torch::Tensor> ms,
+                         std::vector<torch::Tensor> vs,
+                         torch::Tensor bc_pow, double lr, double b1,
+                         double b2, double eps, double wd) {
+  const int n = (int)ps.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "adam_step_f32_multi: 1..8 tensors");
+  TORCH_CHECK((int)gs.size() == n && (int)ms.size() == n &&
+              (int)vs.size() == n, "adam_step_f32_multi: list sizes");
+  CHK_DT(bc_pow, torch::kFloat64);
+  float* pp[8]; const float* gp[8]; float* mp[8]; float* vp[8]; long ns[8];
+  for (int t = 0; t < n; ++t) {
+    CHK_CUDA(ps[t]); CHK_CONTIG(ps[t]); CHK_DT(ps[t], torch::kFloat32);
+    CHK_CONTIG(gs[t]); CHK_DT(gs[t], torch::kFloat32);
+    TORCH_CHECK(ps[t].numel() == gs[t].numel() &&
+                ps[t].numel() == ms[t].numel() &&
+                ps[t].numel() == vs[t].numel(), "sizes");
+    pp[t] = ps[t].data_ptr<float>();
+    gp[t] = gs[t].data_ptr<float>();
+    mp[t] = ms[t].data_ptr<float>();
+    vp[t] = vs[t].data_ptr<float>();
+    ns[t] = ps[t].numel();
+  }
+  launch_adam_f32_multi(pp, gp, mp, vp, ns, n,
+                        (const double*)bc_pow.data_ptr(), (float)lr,
+                        (float)b1, (float)b2, (float)eps, (float)wd,
+                        cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_concat_fwd", &gather_concat_fwd);
   m.def("gather_concat_bwd", &gather_concat_bwd);
@@ -809,4 +845,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_tick", &adam_tick);
   m.def("adam_step_bf16", &adam_step_bf16);
   m.def("adam_step_f32", &adam_step_f32);
+  m.def("adam_step_f32_multi", &adam_step_f32_multi);
 }
