@@ -94,6 +94,7 @@ void launch_slab_sum_f32(const float* p, float* out, int S, int N,
   slab_sum_f32_kernel<<<N, 256, 0, stream>>>(p, out, S, N);
 }
 
+
 void launch_colsum_bf16(const void* x, float* out, long B, long L,
                         hipStream_t stream) {
   const int rows_per_block = 32;  // wider grid: latency-bound otherwise
