@@ -62,3 +62,64 @@ def test_read_vocab_file_no_extras(tmp_path):
     p.write_text("0\t<PAD/>\n1\tp1\n", encoding="utf-8")
     v = read_vocab_file(str(p))
     assert v.stoi["p1"] == 1
+
+
+# ---------------------------------------------------------------------------
+# Property-based parity: our vocab helpers vs a direct transcription of the
+# reference's regex semantics (model/dataset.py:55-56,86-92), over arbitrary
+# method-name-shaped inputs.  These are the two functions every metric and
+# label depends on; a divergence on ANY input breaks F1 parity silently.
+import re
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+_REF_NORM = re.compile(r"[_0-9]+")
+_REF_CAMEL = re.compile(r"([a-z]+)([A-Z][a-z]+)|([A-Z][a-z]+)")
+
+
+def _ref_normalize(name):
+    return _REF_NORM.sub("", name)
+
+
+def _ref_subtokens(name):
+    # verbatim reference semantics (model/dataset.py:90-92): re.split with
+    # capture groups KEEPS unmatched runs (digits, '$', ALLCAPS tails) as
+    # tokens — '0Aa' -> ['0', 'aa'], not ['aa']
+    return [x.lower() for x in _REF_CAMEL.split(name)
+            if x is not None and x != ""]
+
+
+_name_st = st.text(
+    alphabet=st.sampled_from(
+        "abcdefghijklmnopqrstuvwxyzABCDEFGHIJKLMNOPQRSTUVWXYZ0123456789_$"
+    ),
+    min_size=1, max_size=40,
+)
+
+
+@settings(max_examples=300, deadline=None)
+@given(_name_st)
+def test_normalize_matches_reference_regex(name):
+    assert normalize_method_name(name) == _ref_normalize(name)
+
+
+@settings(max_examples=300, deadline=None)
+@given(_name_st)
+def test_subtokens_match_reference_regex(name):
+    assert get_method_subtokens(name) == _ref_subtokens(name)
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(_name_st, min_size=1, max_size=30))
+def test_vocab_append_freq_invariant(names):
+    """freq stays 1 per unique name and indices are stable/dense,
+    regardless of append order or repetition (reference dataset.py:64-74)."""
+    v = Vocab()
+    for n in names:
+        v.append(n)
+    uniq = list(dict.fromkeys(names))
+    assert len(v.stoi) == len(uniq)
+    assert v.get_freq_list() == [1] * len(uniq)
+    for n in names:
+        assert v.itos[v.stoi[n]] == n
