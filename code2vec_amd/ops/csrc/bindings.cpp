@@ -27,6 +27,9 @@ void launch_cast_clear_rows(float*, const int*, unsigned char*, void*, long,
                             int, hipStream_t);
 void launch_scatter_group(const int*, int*, int*, long*, long, hipStream_t);
 void launch_exclusive_scan(const int*, int*, int*, long, hipStream_t);
+size_t cub_exclusive_scan_temp_bytes(long);
+void launch_cub_exclusive_scan(const int*, int*, void*, size_t, long,
+                               hipStream_t);
 void launch_combiner_fwd(const void*, const void*, const float*, const float*,
                          void*, void*, float*, float*, long, int, int, int,
                          float, unsigned long long,
@@ -501,6 +504,25 @@ void exclusive_scan(torch::Tensor x, torch::Tensor partial,
                         out.data_ptr<int>(), n, cur_stream());
 }
 
+long cub_scan_temp_bytes(long n) {
+  return (long)cub_exclusive_scan_temp_bytes(n);
+}
+
+// hipCUB decoupled-lookback exclusive scan; temp is a uint8 scratch of
+// at least cub_scan_temp_bytes(n)
+void cub_exclusive_scan(torch::Tensor x, torch::Tensor temp,
+                        torch::Tensor out) {
+  CHK_CUDA(x); CHK_CONTIG(x); CHK_DT(x, torch::kInt32);
+  CHK_CONTIG(out); CHK_DT(out, torch::kInt32);
+  const long n = x.numel();
+  TORCH_CHECK(out.numel() >= n, "scan out too small");
+  TORCH_CHECK(temp.numel() >= (long)cub_exclusive_scan_temp_bytes(n),
+              "cub scan temp too small");
+  launch_cub_exclusive_scan(x.data_ptr<int>(), out.data_ptr<int>(),
+                            temp.data_ptr(), (size_t)temp.numel(), n,
+                            cur_stream());
+}
+
 void head_bwd_prep(torch::Tensor label, torch::Tensor weight,
                    torch::Tensor acc, torch::Tensor gscale, torch::Tensor lse,
                    torch::Tensor coef_lse) {
@@ -826,6 +848,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_wgrad", &head_wgrad);
   m.def("head_dgrad", &head_dgrad);
   m.def("exclusive_scan", &exclusive_scan);
+  m.def("cub_exclusive_scan", &cub_exclusive_scan);
+  m.def("cub_scan_temp_bytes", &cub_scan_temp_bytes);
   m.def("head_bwd_prep", &head_bwd_prep);
   m.def("swizzle_cv", &swizzle_cv);
   m.def("row_max_argmax", &row_max_argmax);

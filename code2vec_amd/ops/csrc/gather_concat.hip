@@ -15,6 +15,8 @@
 
 #include "common.h"
 
+#include <hipcub/hipcub.hpp>
+
 // One wave per output row (context); 4 waves per block; grid-stride over rows.
 // Each lane copies 16-byte chunks; chunk -> segment resolved per lane.
 __global__ void gather_concat_fwd_kernel(
@@ -510,6 +512,28 @@ void launch_gather_concat_bwd(const int* starts, const int* paths,
   int grid = (int)min((M + waves_per_block - 1) / waves_per_block, (long)16384);
   gather_concat_bwd_kernel<<<grid, block, 0, stream>>>(
       starts, paths, ends, (const bf16*)gout, dterm, dpath, M, TS, PS);
+}
+
+}  // extern "C"
+
+extern "C" {
+
+// rocPRIM/hipCUB single-pass (decoupled-lookback) exclusive scan — one
+// kernel pair instead of the 3-launch partials/spine/apply chain above.
+// Used for the counting-sort cursor over the [T]/[P] histogram.
+size_t cub_exclusive_scan_temp_bytes(long n) {
+  size_t bytes = 0;
+  (void)hipcub::DeviceScan::ExclusiveSum(
+      nullptr, bytes, (const int*)nullptr, (int*)nullptr, (int)n,
+      (hipStream_t)0);
+  return bytes;
+}
+
+void launch_cub_exclusive_scan(const int* in, int* out, void* temp,
+                               size_t temp_bytes, long n,
+                               hipStream_t stream) {
+  (void)hipcub::DeviceScan::ExclusiveSum(temp, temp_bytes, in, out, (int)n,
+                                         stream);
 }
 
 }  // extern "C"

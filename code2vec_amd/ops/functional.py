@@ -40,6 +40,10 @@ _HEAD_FWD_MAXL = int(os.environ.get("C2V_HEAD_FWD_MAXL", "98304"))
 # (contiguous A reads; re-enables the fused stats epilogue at java-large
 # scale).  C2V_HF_AIMG=0 falls back to hipBLASLt + lsm_partial there.
 _HF_AIMG = os.environ.get("C2V_HF_AIMG", "1") == "1"
+# hipCUB decoupled-lookback scan for the counting-sort cursor (one launch
+# vs the 3-kernel partials/spine/apply chain); C2V_SCAN_CUB=0 reverts
+_SCAN_CUB = os.environ.get("C2V_SCAN_CUB", "1") == "1"
+_CUB_TEMP_BYTES: dict = {}
 # streaming head backward: recompute logits by MFMA inside dcv instead of
 # reading the [B, L] tensor (C2V_HB_RC=0 reverts to the logits-reading
 # kernel)
@@ -207,6 +211,15 @@ def _scratch_i32(tag: str, n: int, device, zero: bool = True) -> torch.Tensor:
     return buf
 
 
+def _scratch_u8(tag: str, n: int, device) -> torch.Tensor:
+    key = ("u8", tag, n, str(device))
+    buf = _scratch_cache.get(key)
+    if buf is None:
+        buf = torch.empty(n, dtype=torch.uint8, device=device)
+        _scratch_cache[key] = buf
+    return buf
+
+
 def _scratch_flags(tag: str, rows: int, device) -> torch.Tensor:
     key = ("flags", tag, rows, str(device))
     buf = _scratch_cache.get(key)
@@ -272,7 +285,15 @@ def _group_by_index(idx: torch.Tensor, table_rows: int,
     empty_l = torch.empty(0, dtype=torch.int64, device=dev)
     ext().group_by_index(idx, counts, empty_i, empty_i, empty_l, True)
     if pool_tag is not None:
-        ext().exclusive_scan(counts, spart, cursor)
+        if _SCAN_CUB:
+            tb = _CUB_TEMP_BYTES.get(table_rows + 1)
+            if tb is None:
+                tb = int(ext().cub_scan_temp_bytes(table_rows + 1))
+                _CUB_TEMP_BYTES[table_rows + 1] = tb
+            temp = _scratch_u8(f"cubscan_{pool_tag}", max(tb, 16), dev)
+            ext().cub_exclusive_scan(counts, temp, cursor)
+        else:
+            ext().exclusive_scan(counts, spart, cursor)
     else:
         cursor = torch.zeros_like(counts)
         cursor[1:] = torch.cumsum(counts[:-1], 0)
