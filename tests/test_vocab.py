@@ -123,3 +123,63 @@ def test_vocab_append_freq_invariant(names):
     assert v.get_freq_list() == [1] * len(uniq)
     for n in names:
         assert v.itos[v.stoi[n]] == n
+
+
+# ---------------------------------------------------------------------------
+# read_vocab_file index-shift semantics over generated vocab files
+# (reference model/dataset_reader.py:22-41).
+#
+# NOTE a latent reference bug, deliberately NOT replicated: the reference's
+# extra-token loop assigns `index = 1` and never increments it, so TWO OR
+# MORE extra tokens would all collide at index 1 (overwriting itos[1] and
+# double-counting freq).  That path is unreachable in the reference — the
+# only call sites pass [] or ["@question"] (dataset_reader.py:48,51) — so
+# this implementation uses the evident intent (consecutive indexes from 1),
+# which is identical to the reference on every reachable input.
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    st.lists(
+        st.text(alphabet=st.sampled_from("abcdefgXYZ_09"),
+                min_size=1, max_size=10),
+        min_size=0, max_size=50, unique=True),
+    st.integers(min_value=0, max_value=1),
+)
+def test_read_vocab_file_shift_property(names, n_extra):
+    import os
+    import tempfile
+
+    extra = ["@question"][:n_extra]
+    with tempfile.NamedTemporaryFile(
+            "w", suffix=".txt", delete=False, encoding="utf-8") as f:
+        f.write("0\t<PAD/>\n")
+        for i, n in enumerate(names):
+            f.write(f"{i + 1}\t{n}\n")
+        path = f.name
+    try:
+        v = read_vocab_file(path, extra_tokens=extra)
+        shift = len(extra)
+        assert v.itos[0] == "<PAD/>"
+        for k, n in enumerate(extra):
+            assert v.stoi[n] == 1 + k
+        for i, n in enumerate(names):
+            # duplicate names keep their FIRST index (Vocab.append no-ops)
+            if v.stoi[n] == i + 1 + shift:
+                assert v.itos[i + 1 + shift] == n
+        assert len(v.stoi) == len(set(names) | set(extra)) + 1
+    finally:
+        os.unlink(path)
+
+
+def test_read_vocab_file_question_shift_exact(tmp_path):
+    """The exact reachable case: terminal vocab with ["@question"] —
+    every file index > 0 shifts by one, PAD stays 0
+    (dataset_reader.py:51, QUESTION_TOKEN_INDEX == 1)."""
+    p = tmp_path / "term.txt"
+    p.write_text("0\t<PAD/>\n1\tfoo\n2\tbar\n")
+    v = read_vocab_file(str(p), extra_tokens=["@question"])
+    assert v.stoi["<PAD/>"] == 0
+    assert v.stoi["@question"] == QUESTION_TOKEN_INDEX == 1
+    assert v.stoi["foo"] == 2
+    assert v.stoi["bar"] == 3
