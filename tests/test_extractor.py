@@ -227,3 +227,117 @@ def test_extractor_torture_java():
     m = recs["math"]
     mterms = {x for ctx in m["contexts"] for x in (ctx[0], ctx[2])}
     assert "@double_literal" in mterms and "@char_literal" in mterms
+
+
+EXTRA = """
+public enum Color { RED, GREEN, BLUE;
+  public String tag() { return name().toLowerCase() + ordinal(); }
+}
+
+interface Shape {
+  double area();
+  default String describe(double scale) {
+    double a = area() * scale;
+    return "area=" + a;
+  }
+}
+
+@Deprecated
+public class Outer<T extends Comparable<T>> {
+  static int counter;
+  static { counter = 1; }
+  { counter += 1; }
+
+  public int[] histogram(int... values) {
+    int[] bins = new int[10];
+    for (int v : values) { bins[v % 10]++; }
+    return bins;
+  }
+
+  public String pick(boolean flag, String a, String b) {
+    String r = flag ? a : b;
+    do { r = r.trim(); } while (r.length() > 80);
+    return r;
+  }
+
+  class Inner {
+    public T min(java.util.List<T> xs) {
+      T best = xs.get(0);
+      for (T x : xs) { if (x.compareTo(best) < 0) best = x; }
+      return best;
+    }
+  }
+
+  public Runnable task(final int n) {
+    return new Runnable() {
+      public void run() { System.out.println(n + counter); }
+    };
+  }
+}
+"""
+
+
+def test_extractor_enums_interfaces_inner_classes():
+    """Enums, default interface methods, static/instance initializers,
+    varargs + arrays, ternary, do-while, generic inner classes and
+    anonymous classes must all extract without error."""
+    recs = by_name(X.extract_source(EXTRA))
+    for name in ("tag", "describe", "histogram", "pick", "min"):
+        assert name in recs, sorted(recs)
+        assert "error" not in recs[name], (name, recs[name].get("error"))
+        assert len(recs[name]["contexts"]) > 0, name
+    # varargs parameter is anonymized like any var
+    h = recs["histogram"]
+    assert "values" in dict(h["aliases"])
+
+
+def test_extractor_never_crashes_on_garbage():
+    """The native parser must fail SOFT (error records / empty output) on
+    arbitrary malformed input — never abort the process.  A segfault here
+    kills pytest, which is the assertion."""
+    import random
+
+    rnd = random.Random(7)
+    fragments = [
+        "class", "interface", "enum", "{", "}", "(", ")", ";", ",",
+        "public", "static", "int", "void", "String", "x", "foo", "bar",
+        "=", "+", "->", "::", "<", ">", "[", "]", "@", '"unterminated',
+        "/* open comment", "'c", "0x", "1.2.3", "\\u00zz", "if", "for",
+        "return", "new", "super", "this", "...", "?", ":", "!",
+    ]
+    for trial in range(60):
+        n = rnd.randint(1, 120)
+        src = " ".join(rnd.choice(fragments) for _ in range(n))
+        out = X.extract_source(src)
+        assert isinstance(out, list)
+    # truncations of valid source at every 37th byte
+    for cut in range(1, len(EXTRA), 37):
+        out = X.extract_source(EXTRA[:cut])
+        assert isinstance(out, list)
+    # pathological shapes: unicode identifiers, kilobyte tokens, nested
+    # generic closers, annotation args, comment bombs, empty input
+    nasty = [
+        "",
+        "class \u00e9\u4e2d { void f\u00fc() { int \u03b1 = 1; } }",
+        "class A { void f() { int " + "x" * 4096 + " = 1; } }",
+        "class B { java.util.Map<String, java.util.List<int[]>> m() "
+        "{ return null; } }",
+        "@SuppressWarnings({\"a\", \"b\"}) class C { @Override void g() {} }",
+        "/*" + "*" * 20000 + "/ class D { void h() {} }",
+        "class E { void i() { for(;;){} } }",
+        "\ufeffclass F { void j() {} }",  # BOM
+        "class G { String s = \"" + "\\n" * 2000 + "\"; void k() {} }",
+    ]
+    for src in nasty:
+        out = X.extract_source(src)
+        assert isinstance(out, list)
+
+
+def test_extractor_deep_nesting_bounded():
+    """Deep expression/block nesting must not blow the native stack."""
+    deep_expr = "int v = " + "(" * 180 + "1" + ")" * 180 + ";"
+    deep_block = "{" * 120 + "int q = 1;" + "}" * 120
+    src = ("public class D { public int f(int a) { " + deep_expr +
+           " " + deep_block + " return v + a; } }")
+    out = X.extract_source(src)
+    assert isinstance(out, list)
