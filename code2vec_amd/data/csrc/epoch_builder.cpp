@@ -130,6 +130,20 @@ inline long parse_int(const char* p, const char* e, const char** out) {
   *out = p;
   return neg ? -v : v;
 }
+
+// strict: must consume at least one digit; advances *pp past the number
+inline bool parse_int_strict(const char** pp, const char* e, long* out) {
+  const char* p = *pp;
+  bool neg = false;
+  if (p < e && *p == '-') { neg = true; ++p; }
+  const char* d0 = p;
+  long v = 0;
+  while (p < e && *p >= '0' && *p <= '9') v = v * 10 + (*p++ - '0');
+  if (p == d0) return false;
+  *pp = p;
+  *out = neg ? -v : v;
+  return true;
+}
 }  // namespace
 
 pybind11::dict parse_corpus(const std::string& path) {
@@ -151,6 +165,7 @@ pybind11::dict parse_corpus(const std::string& path) {
 
     bool in_record = false;
     int mode = 0;
+    size_t lineno = 0;
     auto flush = [&]() {
       if (in_record) {
         rec_off.push_back((int64_t)(triples.size() / 3));
@@ -162,6 +177,7 @@ pybind11::dict parse_corpus(const std::string& path) {
     const char* end = base + buf.size();
     const char* line = base;
     while (line <= end) {
+      ++lineno;
       const char* nl = (const char*)memchr(line, '\n', (size_t)(end - line));
       const char* le = nl ? nl : end;
       size_t len;
@@ -190,23 +206,37 @@ pybind11::dict parse_corpus(const std::string& path) {
         } else if (len >= 4 && memcmp(lb, "doc:", 4) == 0) {
           // parsed and discarded
         } else if (mode == 1) {
+          // STRICT start\tpath\tend triple — a malformed line must fail
+          // loudly (the reference's int()/unpack would crash too); silent
+          // tolerance here fabricated PAD-ish triples from corrupt input
           const char* q = lb;
-          long s = parse_int(q, lb + len, &q);
-          if (q < lb + len && *q == '\t') ++q;
-          long pth = parse_int(q, lb + len, &q);
-          if (q < lb + len && *q == '\t') ++q;
-          long e2 = parse_int(q, lb + len, &q);
+          const char* le2 = lb + len;
+          long s, pth, e2;
+          bool ok = parse_int_strict(&q, le2, &s);
+          ok = ok && q < le2 && *q == '\t';
+          if (ok) ++q;
+          ok = ok && parse_int_strict(&q, le2, &pth);
+          ok = ok && q < le2 && *q == '\t';
+          if (ok) ++q;
+          ok = ok && parse_int_strict(&q, le2, &e2);
+          ok = ok && q == le2;  // no trailing fields (split() parity)
+          if (!ok)
+            throw std::runtime_error(
+                "malformed path-context line " + std::to_string(lineno) +
+                " in " + path);
           triples.push_back((int32_t)(s + 1));  // +@question shift
           triples.push_back((int32_t)pth);
           triples.push_back((int32_t)(e2 + 1));
         } else if (mode == 2) {
           const char* tab =
               (const char*)memchr(lb, '\t', len);
-          if (tab) {
-            var_orig.push_back({(size_t)(lb - base), (size_t)(tab - lb)});
-            var_alias.push_back({(size_t)(tab + 1 - base),
-                                 len - (size_t)(tab + 1 - lb)});
-          }
+          if (!tab)
+            throw std::runtime_error(
+                "malformed vars line " + std::to_string(lineno) + " in " +
+                path);
+          var_orig.push_back({(size_t)(lb - base), (size_t)(tab - lb)});
+          var_alias.push_back({(size_t)(tab + 1 - base),
+                               len - (size_t)(tab + 1 - lb)});
         }
       }
       if (!nl) break;

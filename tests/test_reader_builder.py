@@ -414,3 +414,49 @@ def test_synthetic_reader_builder_integration():
     d0 = d.paths.copy()
     d1 = b.refresh_train_dataset(epoch=1)
     assert not np.array_equal(d0, d1.paths)
+
+
+class TestMalformedCorpus:
+    """Corrupt corpora must fail LOUDLY with a line number — the native
+    parser used to fabricate PAD-ish triples from malformed lines
+    silently (the reference's int()/tuple-unpack crashes on them)."""
+
+    def _mk(self, tmp_path, corpus_text):
+        c = tmp_path / "c.txt"
+        c.write_text(corpus_text)
+        t = tmp_path / "t.txt"
+        t.write_text("0\t<PAD/>\n1\tfoo\n2\tbar\n")
+        p = tmp_path / "p.txt"
+        p.write_text("0\t<PAD/>\n1\tpp\n")
+        return str(c), str(p), str(t)
+
+    @pytest.mark.parametrize("bad", [
+        "1\tx\t2",        # non-integer field
+        "1\t1",           # missing field
+        "1\t1\t2\t3",     # extra field (reference unpack would crash too)
+        "",               # (blank ends the record; covered elsewhere)
+    ])
+    def test_malformed_context_line_raises(self, tmp_path, bad):
+        if bad == "":
+            return  # blank line = record separator, legal
+        c, p, t = self._mk(tmp_path,
+                           f"#1\nlabel:getFoo\npaths:\n{bad}\n\n")
+        with pytest.raises((RuntimeError, ValueError),
+                           match="malformed|invalid literal|not enough|"
+                                 "too many"):
+            CorpusReader(c, p, t)
+
+    def test_malformed_vars_line_raises(self, tmp_path):
+        c, p, t = self._mk(tmp_path,
+                           "#1\nlabel:a\nvars:\nnameonly\npaths:\n1\t1\t2\n\n")
+        with pytest.raises((RuntimeError, ValueError), match="malformed|"):
+            CorpusReader(c, p, t)
+
+    def test_error_carries_line_number(self, tmp_path):
+        c, p, t = self._mk(tmp_path,
+                           "#1\nlabel:getFoo\npaths:\n1\t1\t2\n1\tBAD\t2\n\n")
+        try:
+            CorpusReader(c, p, t)
+            assert False, "expected a parse error"
+        except (RuntimeError, ValueError) as e:
+            assert "5" in str(e) or "literal" in str(e)
