@@ -126,6 +126,10 @@ def averaged_subtoken_match(expected_labels, actual_labels, label_vocab):
         precs.append(prec)
         recs.append(rec)
         f1s.append(f1)
+    if not accs:
+        # empty eval set: the reference would emit NaN (np.average of
+        # empty), which is not valid JSON in the metric lines — report 0s
+        return (0.0, 0.0, 0.0, 0.0)
     return (
         float(np.average(accs)),
         float(np.average(precs)),

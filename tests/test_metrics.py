@@ -71,3 +71,15 @@ def test_exact_match_against_sklearn():
 def test_exact_match_perfect():
     acc, prec, rec, f1 = M.exact_match([1, 2, 3], [1, 2, 3])
     assert (acc, prec, rec, f1) == (1.0, 1.0, 1.0, 1.0)
+
+
+def test_metrics_empty_inputs_return_zeros():
+    """Empty eval sets must yield 0-tuples (never NaN — metric lines are
+    JSON; the reference's np.average would produce NaN here)."""
+    from code2vec_amd.data.vocab import Vocab
+
+    v = Vocab()
+    v.append("getfoo", subtokens=["get", "foo"])
+    assert M.exact_match([], []) == (0.0, 0.0, 0.0, 0.0)
+    assert M.subtoken_match([], [], v) == (0.0, 0.0, 0.0, 0.0)
+    assert M.averaged_subtoken_match([], [], v) == (0.0, 0.0, 0.0, 0.0)
