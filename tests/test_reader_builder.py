@@ -460,3 +460,28 @@ class TestMalformedCorpus:
             assert False, "expected a parse error"
         except (RuntimeError, ValueError) as e:
             assert "5" in str(e) or "literal" in str(e)
+
+
+def test_batch_iterator_edge_sizes():
+    """batch > dataset, empty dataset, ragged tail — all legal."""
+    from types import SimpleNamespace
+
+    import torch
+
+    from code2vec_amd.engine.loader import BatchIterator
+
+    def mk(n):
+        return SimpleNamespace(
+            ids=list(range(n)),
+            starts=np.zeros((n, 4), dtype=np.int64),
+            paths=np.zeros((n, 4), dtype=np.int64),
+            ends=np.zeros((n, 4), dtype=np.int64),
+            labels=np.zeros(n, dtype=np.int64))
+
+    cpu = torch.device("cpu")
+    it = BatchIterator(mk(3), 8, shuffle=True, seed=1, device=cpu)
+    assert [b["label"].shape[0] for b in it] == [3] and len(it) == 1
+    it = BatchIterator(mk(0), 4, shuffle=False, seed=1, device=cpu)
+    assert list(it) == [] and len(it) == 0
+    it = BatchIterator(mk(9), 4, shuffle=True, seed=1, device=cpu)
+    assert [b["label"].shape[0] for b in it] == [4, 4, 1]
