@@ -350,3 +350,32 @@ def test_bench_driver_contract(tmp_path):
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["higher_is_better"] is True and d["scaling"] == "weak"
     assert isinstance(d["config"], dict) and "global_batch" in d["config"]
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_world2_cpu():
+    """Rehearse the driver's multi-GPU invocation shape on CPU/gloo:
+    torchrun --nproc-per-node 2 bench.py --gpus 2 must produce exactly one
+    JSON metric line (rank 0), with n_gpus/global_batch scaled."""
+    import json
+    import socket
+    import subprocess
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(repo, "bench.py"), "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--config", "tiny", "--pool", "2"],
+        capture_output=True, text=True, timeout=540, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    json_lines = [ln for ln in out.stdout.splitlines()
+                  if ln.startswith("{") and "path_contexts_per_sec" in ln]
+    assert len(json_lines) == 1, out.stdout[-1500:]
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 64  # 2 ranks x tiny batch 32
