@@ -333,7 +333,7 @@ def run_infer(args, cfg, ctx, model, pool) -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if device.type == "cuda" else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"code2vec-{args.config}", "global_batch": B,

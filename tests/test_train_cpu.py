@@ -350,6 +350,17 @@ def test_bench_driver_contract(tmp_path):
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["higher_is_better"] is True and d["scaling"] == "weak"
     assert isinstance(d["config"], dict) and "global_batch" in d["config"]
+    assert d["dtype"] == "fp32"  # CPU fallback states its real dtype
+
+    # --mode infer carries the same contract shape
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--config", "tiny",
+         "--mode", "infer", "--steps", "2", "--warmup", "1", "--pool", "2"],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["metric"] == "infer_path_contexts_per_sec"
+    assert d["dtype"] == "fp32" and d["value"] > 0
 
 
 @pytest.mark.timeout(600)
