@@ -390,3 +390,51 @@ def test_bench_torchrun_world2_cpu():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 64  # 2 ranks x tiny batch 32
+
+
+def test_cli_flag_parity_with_reference():
+    """Every reference flag (reference main.py:37-79) must
+    exist with the reference's exact default; our additions are
+    new-flag-only (never a changed default)."""
+    p = cli.build_parser()
+    defaults = {a.dest: a.default for a in p._actions if a.dest != "help"}
+    reference = {
+        "random_seed": 123,
+        "corpus_path": "./dataset/corpus.txt",
+        "path_idx_path": "./dataset/path_idxs.txt",
+        "terminal_idx_path": "./dataset/terminal_idxs.txt",
+        "batch_size": 32,
+        "terminal_embed_size": 100,
+        "path_embed_size": 100,
+        "encode_size": 300,
+        "max_path_length": 200,
+        "model_path": "./output",
+        "vectors_path": "./output/code.vec",
+        "test_result_path": None,
+        "max_epoch": 40,
+        "lr": 0.01,
+        "beta_min": 0.9,
+        "beta_max": 0.999,
+        "weight_decay": 0.0,
+        "dropout_prob": 0.25,
+        "no_cuda": False,
+        "gpu": "cuda:0",
+        "num_workers": 4,
+        "env": None,
+        "print_sample_cycle": 10,
+        "eval_method": "subtoken",
+        "find_hyperparams": False,
+        "num_trials": 100,
+        "angular_margin_loss": False,
+        "angular_margin": 0.5,
+        "inverse_temp": 30.0,
+        "infer_method_name": True,
+        "infer_variable_name": False,
+        "shuffle_variable_indexes": False,
+    }
+    for flag, want in reference.items():
+        assert flag in defaults, f"missing reference flag --{flag}"
+        assert defaults[flag] == want, (flag, defaults[flag], want)
+    # eval_method choices parity (main.py:68)
+    ev = next(a for a in p._actions if a.dest == "eval_method")
+    assert set(ev.choices) == {"subtoken", "exact", "ave_subtoken"}
