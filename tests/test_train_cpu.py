@@ -438,3 +438,34 @@ def test_cli_flag_parity_with_reference():
     # eval_method choices parity (main.py:68)
     ev = next(a for a in p._actions if a.dest == "eval_method")
     assert set(ev.choices) == {"subtoken", "exact", "ave_subtoken"}
+
+
+def test_checkpoint_key_set_matches_reference_format():
+    """EXACT state-dict key set of the reference model (model/model.py:
+    21-42) — a checkpoint written here must load into the reference's
+    nn.Module and vice versa."""
+    from code2vec_amd.models.code2vec import (
+        Code2VecTorch, init_logical_params, reference_state_dict_torch)
+    from code2vec_amd.utils.options import Option
+
+    opt = Option(terminal_count=50, path_count=40, label_count=10,
+                 max_path_length=8, terminal_embed_size=6,
+                 path_embed_size=6, encode_size=12, dropout_prob=0.0,
+                 batch_size=4, device=torch.device("cpu"))
+    g = torch.Generator().manual_seed(3)
+    m = Code2VecTorch(opt, init_logical_params(opt, g))
+    sd = reference_state_dict_torch(m)
+    assert set(sd.keys()) == {
+        "terminal_embedding.weight",      # model.py:21
+        "path_embedding.weight",          # model.py:22
+        "input_linear.weight",            # model.py:23 (bias=False)
+        "input_layer_norm.weight",        # model.py:24
+        "input_layer_norm.bias",
+        "attention_parameter",            # model.py:31
+        "output_linear.weight",           # model.py:41
+        "output_linear.bias",             # model.py:42
+    }
+    assert sd["terminal_embedding.weight"].shape == (50, 6)
+    assert sd["output_linear.weight"].shape == (10, 12)
+    assert sd["input_linear.weight"].shape == (12, 18)
+    assert sd["attention_parameter"].shape == (12,)
