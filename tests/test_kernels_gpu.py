@@ -625,6 +625,30 @@ class TestFusedHeadLoss:
         # 3e-2 bound the reading path meets on some shapes
         assert relerr(dcv_b.float(), cvr.grad) < 6e-2
 
+    def test_direct_headfwd_path_matches(self, dev):
+        """C2V_HF_AIMG=0 route: the direct-load head_fwd kernel (the
+        pre-image default) must still match the oracle now that the
+        W-image kernel is the EP=128 default."""
+        from code2vec_amd.ops import functional as Fn
+
+        cv, w, bias, label, weight = self._setup(dev, 96, 7320, seed=61)
+        old = Fn._HF_AIMG
+        Fn._HF_AIMG = False
+        try:
+            loss, dcv, dw, dbias = self._run_fused(cv, w, bias, label,
+                                                   weight)
+        finally:
+            Fn._HF_AIMG = old
+        cvr = cv.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        br = bias.clone().requires_grad_(True)
+        ref = R.logsoftmax_nll(cvr @ wr.t() + br, label, weight)
+        ref.backward()
+        assert abs(float(loss.detach()) - float(ref.detach())) / abs(
+            float(ref.detach())) < 1e-2
+        assert relerr(dcv.float(), cvr.grad) < 3e-2
+        assert relerr(dw.float(), wr.grad) < 3e-2
+
     def test_matches_unfused_chain(self, dev):
         """Fused backward vs the unfused OutputHead+FusedLogSoftmaxNLL
         chain on identical inputs (both bf16 paths -> tight tolerance);
