@@ -392,6 +392,32 @@ def test_bench_torchrun_world2_cpu():
     assert d["config"]["global_batch"] == 64  # 2 ranks x tiny batch 32
 
 
+@pytest.mark.timeout(600)
+def test_bench_torchrun_world4_cpu():
+    """4 gloo ranks through the same launcher shape — more ranks than
+    owned-chunk splits and a deeper bucket partition than world 2."""
+    import json
+    import socket
+    import subprocess
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(repo, "bench.py"), "--gpus", "4", "--steps", "2",
+         "--warmup", "1", "--config", "tiny", "--pool", "2"],
+        capture_output=True, text=True, timeout=540, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    json_lines = [ln for ln in out.stdout.splitlines()
+                  if ln.startswith("{") and "path_contexts_per_sec" in ln]
+    assert len(json_lines) == 1, out.stdout[-1500:]
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 4 and d["config"]["parallelism"] == "dp4"
+
+
 def test_cli_flag_parity_with_reference():
     """Every reference flag (reference main.py:37-79) must
     exist with the reference's exact default; our additions are
