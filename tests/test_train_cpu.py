@@ -495,3 +495,20 @@ def test_checkpoint_key_set_matches_reference_format():
     assert sd["output_linear.weight"].shape == (10, 12)
     assert sd["input_linear.weight"].shape == (12, 18)
     assert sd["attention_parameter"].shape == (12,)
+
+
+def test_all_tool_clis_have_working_help():
+    """Every tools/ CLI must answer --help (or print a graceful GPU-needed
+    message) with exit code 0 — no raw tracebacks for a user probing the
+    toolbox on a CPU box."""
+    import subprocess
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tools = ["gen_corpus.py", "extract_paths.py", "derive_label_vocab.py",
+             "visualize_code_vec.py", "kbench.py", "prof_summary.py",
+             "pmc_summary.py"]
+    for t in tools:
+        out = subprocess.run(
+            [sys.executable, os.path.join(repo, "tools", t), "--help"],
+            capture_output=True, text=True, timeout=90, cwd=repo)
+        assert out.returncode == 0, (t, out.stderr[-400:])

@@ -34,6 +34,12 @@ def timeit(fn, iters=30, warmup=5):
 
 
 def main():
+    if "--help" in sys.argv or "-h" in sys.argv:
+        print(__doc__.strip())
+        return
+    if not torch.cuda.is_available():
+        print("kbench: needs a GPU (per-kernel timing); run under gpurun")
+        return
     dev = torch.device("cuda:0")
     B, C = 1024, 200
     T, P, L = 360632, 342846, 30000

@@ -11,6 +11,9 @@ import sys
 
 
 def summarize(db_path: str):
+    if not db_path.endswith(".db"):
+        print(__doc__.strip())
+        raise SystemExit(0)
     db = sqlite3.connect(db_path)
     cur = db.cursor()
     tables = [r[0] for r in cur.execute(
