@@ -188,6 +188,12 @@ def test_trainer_world2_runs(tmp_path):
         assert pr.exitcode == 0
     obj = float(open(result_file).read())
     assert 0.0 <= obj <= 1.0
+    # artifact export is a rank-0-only side effect (trainer.py guards on
+    # ctx.is_rank0): rank 0's dir holds checkpoint+vectors, rank 1's none
+    assert os.path.exists(str(tmp_path / "out0" / "code2vec.model"))
+    assert os.path.exists(str(tmp_path / "out0" / "code.vec"))
+    assert not os.path.exists(str(tmp_path / "out1" / "code2vec.model"))
+    assert not os.path.exists(str(tmp_path / "out1" / "code.vec"))
 
 
 @pytest.mark.timeout(300)
