@@ -512,3 +512,17 @@ def test_all_tool_clis_have_working_help():
             [sys.executable, os.path.join(repo, "tools", t), "--help"],
             capture_output=True, text=True, timeout=90, cwd=repo)
         assert out.returncode == 0, (t, out.stderr[-400:])
+
+
+def test_option_field_parity_with_reference():
+    """Option carries every reference Option field (main.py:93-115)."""
+    from code2vec_amd.utils.options import Option
+
+    opt = Option(terminal_count=5, path_count=4, label_count=3,
+                 device=torch.device("cpu"))
+    for field in ("max_path_length", "terminal_count", "path_count",
+                  "label_count", "terminal_embed_size", "path_embed_size",
+                  "encode_size", "dropout_prob", "batch_size",
+                  "eval_method", "angular_margin_loss", "angular_margin",
+                  "inverse_temp", "device"):
+        assert hasattr(opt, field), field
