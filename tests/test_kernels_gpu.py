@@ -17,8 +17,8 @@ from code2vec_amd.ops import round_up
 
 
 def relerr(a, b):
-    a = a.float()
-    b = b.float()
+    a = a.detach().float()
+    b = b.detach().float()
     d = (a - b).norm()
     n = b.norm()
     return float(d / n) if float(n) > 0 else float(d)
