@@ -226,6 +226,16 @@ class Trainer:
         finally:
             if self.summary_writer is not None:
                 self.summary_writer.close()
+            # drain any orphaned background epoch build (early stop can
+            # leave one in flight) and retire the worker thread
+            pending = self._next_train
+            if pending is not None:
+                try:
+                    pending[1].result()
+                except Exception:  # noqa: BLE001 - build died with train
+                    pass
+                self._next_train = None
+            self._build_pool.shutdown(wait=True)
             # deregister the owned-grad callbacks: a later model in the
             # same process could reuse this model's table addresses and
             # silently hand its grads to a stale callback otherwise
