@@ -5,10 +5,12 @@ initialized from the same logical fp32 tensors (same seeds => identical
 starting points), and both return ``(outputs, code_vector, attention)``.
 
 The HIP backend stores parameters in MI355X-native padded layouts:
-- embedding tables bf16 [T, TS] / [P, PS] (TS/PS = round_up(dt/dp, 32),
-  rows 64-B aligned for vectorized gathers),
-- combiner weight bf16 [KP, EP] (K-major B-operand layout for the MFMA
-  GEMM; KP = 2*TS+PS, EP = round_up(E, 32)),
+- embedding tables bf16 [T, TS] / [P, PS] (TS/PS = seg_round(dt/dp):
+  16-B granules — dt=100 stores as 104, keeping rows 16-B aligned for
+  bf16x8 gathers without the 32-element MFMA padding waste),
+- combiner weight bf16 [EP, KP] stored TRANSPOSED (so the MFMA B
+  fragment is memory-contiguous; KP = round_up(2*TS+PS, 32),
+  EP = round_up(E, 32)),
 - output weight bf16 [L, EP]; LN/attention/bias params fp32.
 Pad regions are zero and stay zero through training (padding checks in
 tests/test_model_gpu.py).
