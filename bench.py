@@ -10,7 +10,9 @@ weak scaling (per-GPU batch fixed at 1024).
 
 A timed step = forward + fused loss + backward + bucketed gradient
 all-reduce + fused Adam — the full training step.  label_count for top11 is
-not published by the reference; 30,000 is assumed and stated in config.
+not published by the reference; 72,416 is DERIVED from its real data via a
+Heaps-law fit to methods.txt's unique-name curve (tools/derive_label_vocab.py)
+and stated in config.
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--config top11]
 For N>1 the driver launches via torch.distributed.run.
