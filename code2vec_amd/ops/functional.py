@@ -1,9 +1,15 @@
 """Autograd wrappers around the HIP kernels.
 
 Each ``torch.autograd.Function`` here pairs a hand-written CDNA4 forward
-kernel with its backward kernels; plain dgrad/wgrad GEMMs go through
-rocBLAS (``torch.matmul``) by design — hand-written MFMA kernels cover the
-*fused* ops that no library provides.
+kernel with its backward kernels.  Nearly every
+hot GEMM ended up custom after A/B against the tuned libraries: the
+combiner forward (fused LN/tanh/dropout epilogue), combiner dgrad
+(dgrad2) and wgrad (split-K — the skinny big-K shape where hipBLASLt is
+~3.5x off), the head forward (+online-softmax stats epilogue) and the
+fused recompute-G head backward (dW/dbias/dcv).  rocBLAS/hipBLASLt
+remain only where they measured faster or equal: the head forward above
+the vocab gate with C2V_HF_AIMG=0, and small reductions torch already
+does at bandwidth (the wgrad partial-slab sum).
 
 Numerical contracts are defined by ops/reference.py; tests compare against
 it in fp32.
