@@ -485,3 +485,57 @@ def test_batch_iterator_edge_sizes():
     assert list(it) == [] and len(it) == 0
     it = BatchIterator(mk(9), 4, shuffle=True, seed=1, device=cpu)
     assert [b["label"].shape[0] for b in it] == [4, 4, 1]
+
+
+def test_builder_epoch_invariants_random_corpora():
+    """Invariants of every built epoch, over randomized tiny corpora:
+    fixed [N, C] shape, zero-padding after each method's true length,
+    indices in vocab range, @method_0 never surviving in starts/ends
+    (replaced by @question = 1), labels in label-vocab range."""
+    from code2vec_amd.data.builder import DatasetBuilder
+    from code2vec_amd.data.synthetic import SyntheticSpec, write_synthetic_corpus
+    from code2vec_amd.data.reader import CorpusReader
+    from code2vec_amd.utils.options import Option
+    import tempfile
+
+    import torch
+
+    for seed in (11, 23, 47):
+        with tempfile.TemporaryDirectory() as d:
+            files = write_synthetic_corpus(
+                d, SyntheticSpec(n_methods=30, n_terminals=50, n_paths=40,
+                                 max_contexts=12, seed=seed))
+            reader = CorpusReader(files["corpus_path"],
+                                  files["path_idx_path"],
+                                  files["terminal_idx_path"])
+            C = 8
+            opt = Option(terminal_count=len(reader.terminal_vocab),
+                         path_count=len(reader.path_vocab),
+                         label_count=len(reader.label_vocab),
+                         max_path_length=C, terminal_embed_size=8,
+                         path_embed_size=8, encode_size=8,
+                         dropout_prob=0.0, batch_size=4,
+                         device=torch.device("cpu"))
+            builder = DatasetBuilder(reader, opt, seed=seed)
+            mzero = reader.terminal_vocab.stoi.get("@method_0", -1)
+            for epoch in (0, 1):
+                ds = builder.refresh_train_dataset(epoch)
+                for arr in (ds.starts, ds.paths, ds.ends):
+                    assert arr.shape[1] == C
+                    assert arr.min() >= 0
+                assert ds.starts.max() < len(reader.terminal_vocab)
+                assert ds.ends.max() < len(reader.terminal_vocab)
+                assert ds.paths.max() < len(reader.path_vocab)
+                assert ds.labels.min() >= 0
+                assert ds.labels.max() < len(reader.label_vocab)
+                if mzero > 0:
+                    # the method's own anonymized name must be masked
+                    # (stoi already carries the +@question shift, and the
+                    # corpus start/end values get the same shift at read)
+                    assert not (ds.starts == mzero).any()
+                    assert not (ds.ends == mzero).any()
+                # pad runs are suffixes: once a column is 0 for a row in
+                # starts, paths and ends are 0 there too
+                pad = ds.starts == 0
+                assert (ds.paths[pad] == 0).all()
+                assert (ds.ends[pad] == 0).all()
