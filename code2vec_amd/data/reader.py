@@ -210,7 +210,7 @@ class CorpusReader:
             triples = []
 
         with open(corpus_path, mode="r", encoding="utf-8") as f:
-            for raw in f:
+            for lineno, raw in enumerate(f, start=1):
                 line = raw.strip(" \r\n\t")
                 if line == "":
                     flush()
@@ -238,10 +238,21 @@ class CorpusReader:
                 elif line.startswith("doc:"):
                     pass  # parsed and discarded (reference :109-110)
                 elif parse_mode == 1:
-                    s, p, e = line.split("\t")
-                    triples.extend((int(s) + q, int(p), int(e) + q))
+                    try:
+                        s, p, e = line.split("\t")
+                        triples.extend((int(s) + q, int(p), int(e) + q))
+                    except ValueError as exc:
+                        raise ValueError(
+                            f"malformed path-context line {lineno} in "
+                            f"{corpus_path}"
+                        ) from exc
                 elif parse_mode == 2:
-                    original_name, alias_name = line.split("\t")[:2]
+                    fields = line.split("\t")
+                    if len(fields) < 2:
+                        raise ValueError(
+                            f"malformed vars line {lineno} in {corpus_path}"
+                        )
+                    original_name, alias_name = fields[:2]
                     normalized = normalize_method_name(original_name)
                     subtokens = get_method_subtokens(normalized)
                     lower = normalized.lower()
