@@ -461,6 +461,25 @@ class TestMalformedCorpus:
         except (RuntimeError, ValueError) as e:
             assert "5" in str(e) or "literal" in str(e)
 
+    def test_python_fallback_parser_same_contract(self, tmp_path,
+                                                  monkeypatch):
+        """With the native extension disabled, the pure-Python parser
+        must produce the SAME items on good corpora and the same
+        line-numbered errors on malformed ones."""
+        import code2vec_amd.data.reader as rd
+
+        good = "#1\nlabel:getFoo\npaths:\n1\t1\t2\n\n"
+        c, p, t = self._mk(tmp_path, good)
+        ref = CorpusReader(c, p, t)
+        monkeypatch.setattr(rd, "_native", None)
+        alt = rd.CorpusReader(c, p, t)
+        assert len(alt.items) == len(ref.items) == 1
+        assert (alt.items[0].path_contexts == ref.items[0].path_contexts).all()
+        assert alt.items[0].normalized_label == ref.items[0].normalized_label
+        c2, p2, t2 = self._mk(tmp_path, "#1\nlabel:a\npaths:\n1\tX\t2\n\n")
+        with pytest.raises(ValueError, match="malformed.*line 4"):
+            rd.CorpusReader(c2, p2, t2)
+
 
 def test_batch_iterator_edge_sizes():
     """batch > dataset, empty dataset, ragged tail — all legal."""
